@@ -1,0 +1,207 @@
+"""Component-function -> AppDef materialization.
+
+A *component* is a typed Python function returning an AppDef.  We build an
+argparse parser from its signature (types + docstring help), parse the user's
+string args, and call it (parity: torchx/specs/builders.py:30-270).
+"""
+
+from __future__ import annotations
+
+import argparse
+import inspect
+import re
+import typing
+from typing import Any, Callable, Dict, List, Optional
+
+from .api import AppDef, BindMount, DeviceMount, Mount, MountType, VolumeMount
+
+
+class ComponentError(Exception):
+    pass
+
+
+def _decode_string(value: str, annotation: Any) -> Any:
+    """Decode a CLI string into the annotated type."""
+    origin = typing.get_origin(annotation)
+    args = typing.get_args(annotation)
+    if annotation in (inspect.Parameter.empty, str, Any, None):
+        return value
+    if annotation is bool:
+        return value.strip().lower() in ("1", "true", "yes", "on")
+    if annotation in (int, float):
+        return annotation(value)
+    if origin is typing.Union:  # Optional[T]
+        non_none = [a for a in args if a is not type(None)]
+        if value.lower() == "none":
+            return None
+        return _decode_string(value, non_none[0]) if non_none else value
+    if origin in (list, List):
+        elem = args[0] if args else str
+        return [_decode_string(v, elem) for v in value.split(",") if v != ""]
+    if origin in (dict, Dict):
+        kt = args[0] if args else str
+        vt = args[1] if len(args) > 1 else str
+        out = {}
+        for pair in value.split(","):
+            if not pair:
+                continue
+            k, _, v = pair.partition("=")
+            if not _:
+                k, _, v = pair.partition(":")
+            out[_decode_string(k, kt)] = _decode_string(v, vt)
+        return out
+    return value
+
+
+def _docstring_param_help(fn: Callable[..., Any]) -> Dict[str, str]:
+    """Extract ``Args:``-section parameter help from a Google/Sphinx docstring."""
+    doc = inspect.getdoc(fn) or ""
+    helps: Dict[str, str] = {}
+    # Google style: "    name: help text"
+    in_args = False
+    current: Optional[str] = None
+    for line in doc.splitlines():
+        stripped = line.strip()
+        if stripped.lower() in ("args:", "arguments:", "parameters:"):
+            in_args = True
+            continue
+        if in_args:
+            if not stripped or (stripped.endswith(":") and " " not in stripped):
+                in_args = False
+                current = None
+                continue
+            m = re.match(r"^([*\w]+)\s*(?:\([^)]*\))?\s*:\s*(.*)$", stripped)
+            if m:
+                current = m.group(1).lstrip("*")
+                helps[current] = m.group(2)
+            elif current:
+                helps[current] += " " + stripped
+    # Sphinx style  :param name: help
+    for m in re.finditer(r":param\s+([*\w]+)\s*:\s*(.+)", doc):
+        helps.setdefault(m.group(1).lstrip("*"), m.group(2).strip())
+    return helps
+
+
+def _fn_summary(fn: Callable[..., Any]) -> str:
+    doc = inspect.getdoc(fn) or ""
+    lines = []
+    for line in doc.splitlines():
+        if not line.strip():
+            if lines:
+                break
+            continue
+        lines.append(line.strip())
+    return " ".join(lines)
+
+
+def create_args_parser(fn: Callable[..., Any]) -> argparse.ArgumentParser:
+    sig = inspect.signature(fn)
+    helps = _docstring_param_help(fn)
+    parser = argparse.ArgumentParser(
+        prog=fn.__name__,
+        description=_fn_summary(fn),
+        formatter_class=argparse.ArgumentDefaultsHelpFormatter,
+    )
+    for name, p in sig.parameters.items():
+        help_txt = helps.get(name, "")
+        if p.kind == inspect.Parameter.VAR_POSITIONAL:
+            parser.add_argument(name, nargs="*", default=[], help=help_txt)
+        elif p.default is inspect.Parameter.empty:
+            parser.add_argument(f"--{name}", required=True, help=help_txt)
+        else:
+            parser.add_argument(f"--{name}", default=p.default, help=help_txt)
+    return parser
+
+
+def materialize_appdef(
+    fn: Callable[..., AppDef],
+    args: List[str],
+    defaults: Optional[Dict[str, str]] = None,
+) -> AppDef:
+    """Parse ``args`` against ``fn``'s signature and call it.
+
+    Precedence: CLI args > ``defaults`` (e.g. from .torchxconfig) > function
+    defaults (parity: torchx/specs/builders.py:137-179).
+    """
+    sig = inspect.signature(fn)
+    parser = create_args_parser(fn)
+    if defaults:
+        known = {
+            f"--{k}": v
+            for k, v in defaults.items()
+            if k in sig.parameters
+        }
+        # inject config defaults for args not given on the CLI
+        given = {a.split("=")[0] for a in args if a.startswith("--")}
+        for flag, v in known.items():
+            if flag not in given:
+                args = [flag, v] + args
+    has_varargs = any(
+        p.kind == inspect.Parameter.VAR_POSITIONAL for p in sig.parameters.values()
+    )
+    if has_varargs:
+        ns, extras = parser.parse_known_args(args)
+    else:
+        ns = parser.parse_args(args)
+        extras = []
+
+    call_args: List[Any] = []
+    call_kwargs: Dict[str, Any] = {}
+    for name, p in sig.parameters.items():
+        raw = getattr(ns, name)
+        if p.kind == inspect.Parameter.VAR_POSITIONAL:
+            call_args.extend((raw or []) + extras)
+            continue
+        if isinstance(raw, str) and p.annotation is not str:
+            raw = _decode_string(raw, p.annotation)
+        call_kwargs[name] = raw
+
+    appdef = fn(*call_args, **call_kwargs)
+    if not isinstance(appdef, AppDef):
+        raise ComponentError(
+            f"component {fn.__name__} returned {type(appdef)}, expected AppDef"
+        )
+    return appdef
+
+
+_MOUNT_TYPES = {t.value: t for t in MountType}
+
+
+def parse_mounts(opts: List[str]) -> List[Mount]:
+    """Parse ``type=bind,src=/x,dst=/y[,readonly]`` mount strings
+    (parity: torchx/specs/builders.py:336)."""
+    mounts: List[Mount] = []
+    # each mount is one comma-joined group starting with type=
+    group: Dict[str, str] = {}
+    groups: List[Dict[str, str]] = []
+    for opt in opts:
+        for part in opt.split(","):
+            if not part:
+                continue
+            k, _, v = part.partition("=")
+            k = k.strip().lower()
+            if k == "type" and group:
+                groups.append(group)
+                group = {}
+            group[k] = v.strip() if _ else "true"
+    if group:
+        groups.append(group)
+
+    for g in groups:
+        mtype = _MOUNT_TYPES.get(g.get("type", ""))
+        if mtype is None:
+            raise ValueError(f"mount needs type=bind|volume|device, got {g}")
+        ro = g.get("readonly", "false").lower() in ("true", "1", "")
+        if mtype == MountType.BIND:
+            mounts.append(BindMount(src_path=g["src"], dst_path=g["dst"], read_only=ro))
+        elif mtype == MountType.VOLUME:
+            mounts.append(VolumeMount(src=g["src"], dst_path=g["dst"], read_only=ro))
+        else:
+            mounts.append(
+                DeviceMount(
+                    src_path=g["src"],
+                    dst_path=g.get("dst", g["src"]),
+                    permissions=g.get("perm", "rwm"),
+                )
+            )
+    return mounts
