@@ -145,18 +145,24 @@ def materialize_appdef(
         ns = parser.parse_args(args)
         extras = []
 
-    call_args: List[Any] = []
-    call_kwargs: Dict[str, Any] = {}
+    # params before *args must be passed positionally; keyword-only after it
+    pos_vals: List[Any] = []
+    kw_vals: Dict[str, Any] = {}
+    seen_varargs = False
     for name, p in sig.parameters.items():
-        raw = getattr(ns, name)
         if p.kind == inspect.Parameter.VAR_POSITIONAL:
-            call_args.extend((raw or []) + extras)
+            pos_vals.extend((getattr(ns, name) or []) + extras)
+            seen_varargs = True
             continue
+        raw = getattr(ns, name)
         if isinstance(raw, str) and p.annotation is not str:
             raw = _decode_string(raw, p.annotation)
-        call_kwargs[name] = raw
+        if seen_varargs or p.kind == inspect.Parameter.KEYWORD_ONLY or not has_varargs:
+            kw_vals[name] = raw
+        else:
+            pos_vals.append(raw)
 
-    appdef = fn(*call_args, **call_kwargs)
+    appdef = fn(*pos_vals, **kw_vals)
     if not isinstance(appdef, AppDef):
         raise ComponentError(
             f"component {fn.__name__} returned {type(appdef)}, expected AppDef"
