@@ -139,11 +139,34 @@ def materialize_appdef(
     has_varargs = any(
         p.kind == inspect.Parameter.VAR_POSITIONAL for p in sig.parameters.values()
     )
+    extras: List[str] = []
     if has_varargs:
-        ns, extras = parser.parse_known_args(args)
+        # everything from the first token that is not a known `--param value`
+        # pair belongs to *args, preserving order (REMAINDER semantics that
+        # also allow `--known` flags before the tail)
+        known_flags = {
+            f"--{n}"
+            for n, p in sig.parameters.items()
+            if p.kind != inspect.Parameter.VAR_POSITIONAL
+        }
+        head: List[str] = []
+        i = 0
+        while i < len(args):
+            tok = args[i]
+            name, eq, _ = tok.partition("=")
+            if name in known_flags:
+                if eq:
+                    head.append(tok)
+                    i += 1
+                else:
+                    head.extend(args[i : i + 2])
+                    i += 2
+            else:
+                extras = args[i:]
+                break
+        ns = parser.parse_args(head)
     else:
         ns = parser.parse_args(args)
-        extras = []
 
     # params before *args must be passed positionally; keyword-only after it
     pos_vals: List[Any] = []
