@@ -1,0 +1,161 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Llama-3-8B bf16 training step (the BASELINE.json
+headline config: dist.ddp Llama-3-8B, RCCL over xGMI, synthetic data,
+random-init weights).
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+launched for N>1 as torch.distributed.run with one rank per GPU; reads
+RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from env; rank 0 prints ONE JSON line
+with the whole-job aggregate tokens/s.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--micro-batch", type=int, default=4)
+    p.add_argument("--seq-len", type=int, default=4096)
+    p.add_argument("--model", type=str, default="llama3_8b",
+                   choices=["llama3_8b", "llama3_8b_small", "tiny"])
+    p.add_argument("--device", type=str, default=None)
+    args = p.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+
+    use_gpu = torch.cuda.is_available() and args.device != "cpu"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+        backend = "nccl"  # RCCL on ROCm
+    else:
+        device = torch.device("cpu")
+        backend = "gloo"
+
+    distributed = world > 1
+    if distributed:
+        import torch.distributed as dist
+
+        dist.init_process_group(backend=backend)
+
+    from torchx_amd.models.llama import (
+        LlamaConfig, LlamaModel, llama3_8b, llama_tiny,
+    )
+    from torchx_amd.parallel import FlatAdamW, FlatDDP, FlatParams
+
+    if args.model == "llama3_8b":
+        cfg = llama3_8b()
+    elif args.model == "llama3_8b_small":
+        cfg = LlamaConfig(num_layers=4)  # 8B shape, 4 layers (debug)
+    else:
+        cfg = llama_tiny()
+        args.seq_len = min(args.seq_len, cfg.max_seq_len)
+        args.micro_batch = 2
+
+    cfg.max_seq_len = max(cfg.max_seq_len, args.seq_len)
+    torch.manual_seed(1234 + rank)
+
+    if use_gpu:
+        from torchx_amd import ops
+
+        ops.hip_ops(required=True)  # fail loudly if the extension is missing
+
+    model = LlamaModel(cfg, device=device)
+    flat = FlatParams(model, device)
+    ddp = FlatDDP(flat)
+    opt = FlatAdamW(flat, lr=3e-4)
+
+    B, S = args.micro_batch, args.seq_len
+    tokens = torch.randint(0, cfg.vocab_size, (B, S), device=device)
+    targets = torch.roll(tokens, shifts=-1, dims=1)
+
+    def one_step() -> float:
+        opt.zero_grad()
+        loss = model(tokens, targets)
+        loss.backward()
+        ddp.finish()
+        opt.step()
+        return float(loss.detach())
+
+    # warmup
+    for _ in range(args.warmup):
+        one_step()
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    loss = 0.0
+    for _ in range(args.steps):
+        loss = one_step()
+    if use_gpu:
+        torch.cuda.synchronize()
+    if distributed:
+        import torch.distributed as dist
+
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if distributed:
+        import torch.distributed as dist
+
+        t = torch.tensor([elapsed], device=device if use_gpu else None,
+                         dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    tokens_per_step = B * S * world
+    tok_s = tokens_per_step / (elapsed / args.steps)
+
+    if rank == 0:
+        result = {
+            "metric": "tokens_per_second",
+            "value": tok_s,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": B * world,
+                "seq_len": S,
+                "parallelism": f"dp{world}",
+                "final_loss": loss,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if distributed:
+        import torch.distributed as dist
+
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,190 @@
+"""GPU numerics tests: every CDNA4 HIP kernel vs the plain PyTorch fp32
+reference of the same op (tests/conftest.py registers the gpu marker)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def dev():
+    assert torch.cuda.is_available()
+    torch.manual_seed(0)
+    return torch.device("cuda:0")
+
+
+def _hip():
+    from torchx_amd import ops
+
+    assert ops.extension_available(), "HIP extension must be built in-tree"
+    return ops
+
+
+def rel_err(a, b):
+    a, b = a.float(), b.float()
+    return (a - b).abs().max().item() / (b.abs().max().item() + 1e-6)
+
+
+def test_mfma_layout_probe(dev):
+    """Validates the assumed A/B/C fragment lane maps for
+    v_mfma_f32_32x32x16_bf16 against torch.matmul (asymmetric inputs)."""
+    ops = _hip()
+    a = torch.randn(32, 16, device=dev).to(torch.bfloat16)
+    b = torch.randn(16, 32, device=dev).to(torch.bfloat16)
+    c = ops.mfma_probe(a, b)
+    ref = a.float() @ b.float()
+    assert torch.allclose(c, ref, atol=2e-2, rtol=2e-2), (
+        f"MFMA layout mismatch: max err {(c - ref).abs().max().item()}"
+    )
+
+
+def test_rmsnorm_fwd_bwd(dev):
+    ops = _hip()
+    x = torch.randn(512, 4096, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(4096, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    y = ops.rmsnorm(x, w)
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    yr = (xr * torch.rsqrt(xr.pow(2).mean(-1, keepdim=True) + 1e-5)) * wr
+    assert rel_err(y, yr) < 2e-2
+
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    assert rel_err(x.grad, xr.grad) < 3e-2
+    assert rel_err(w.grad, wr.grad) < 3e-2
+
+
+def test_rope_fwd_bwd(dev):
+    from torchx_amd.ops import reference, rope_tables
+
+    ops = _hip()
+    B, S, H, D = 2, 256, 4, 128
+    cos, sin = rope_tables(S, D, device=dev)
+    x = torch.randn(B, S, H, D, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    y = ops.rope(x, cos, sin)
+    yr = reference.rope(x.detach(), cos, sin)
+    assert rel_err(y, yr) < 2e-2
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    # inverse-rotation backward equals reference autograd backward
+    xr = x.detach().float().requires_grad_(True)
+    c = cos[:S].view(1, S, 1, D // 2)
+    s = sin[:S].view(1, S, 1, D // 2)
+    x1, x2 = xr[..., : D // 2], xr[..., D // 2:]
+    yref = torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], -1)
+    yref.backward(dy.float())
+    assert rel_err(x.grad, xr.grad) < 2e-2
+
+
+def test_swiglu_fwd_bwd(dev):
+    ops = _hip()
+    g = torch.randn(4096, 1024, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    u = torch.randn_like(g).requires_grad_(True)
+    out = ops.swiglu(g, u)
+    gr = g.detach().float().requires_grad_(True)
+    ur = u.detach().float().requires_grad_(True)
+    outr = torch.nn.functional.silu(gr) * ur
+    assert rel_err(out, outr) < 2e-2
+    do = torch.randn_like(out)
+    out.backward(do)
+    outr.backward(do.float())
+    assert rel_err(g.grad, gr.grad) < 2e-2
+    assert rel_err(u.grad, ur.grad) < 2e-2
+
+
+def test_cross_entropy(dev):
+    ops = _hip()
+    T, V = 128, 128256
+    logits = torch.randn(T, V, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    targets = torch.randint(0, V, (T,), device=dev)
+    loss = ops.cross_entropy(logits, targets)
+    lr_ = logits.detach().float().requires_grad_(True)
+    loss_ref = torch.nn.functional.cross_entropy(lr_, targets)
+    assert abs(loss.item() - loss_ref.item()) < 2e-2
+    loss.backward()
+    loss_ref.backward()
+    assert rel_err(logits.grad, lr_.grad) < 3e-2
+
+
+@pytest.mark.parametrize("S,causal", [(128, True), (256, True), (512, True),
+                                      (192, True), (256, False)])
+def test_attention_fwd(dev, S, causal):
+    ops = _hip()
+    B, Hq, Hkv, D = 2, 8, 2, 128
+    q = torch.randn(B, S, Hq, D, device=dev, dtype=torch.bfloat16)
+    k = torch.randn(B, S, Hkv, D, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(B, S, Hkv, D, device=dev, dtype=torch.bfloat16)
+    from torchx_amd.ops import reference
+
+    o = ops.flash_attention(q, k, v, causal=causal)
+    o_ref = reference.attention(q, k, v, causal=causal)
+    assert rel_err(o, o_ref) < 3e-2, f"attention fwd err {rel_err(o, o_ref)}"
+
+
+def test_attention_lse(dev):
+    ops = _hip()
+    from torchx_amd.ops import reference
+
+    B, S, Hq, Hkv, D = 1, 256, 4, 4, 128
+    q = torch.randn(B, S, Hq, D, device=dev, dtype=torch.bfloat16)
+    k = torch.randn(B, S, Hkv, D, device=dev, dtype=torch.bfloat16)
+    v = torch.randn(B, S, Hkv, D, device=dev, dtype=torch.bfloat16)
+    o, lse = ops.hip_ops().attn_fwd(q, k, v, 1.0 / math.sqrt(D), True)
+    lse_ref = reference.attention_lse(q, k, v, causal=True)
+    assert rel_err(lse, lse_ref) < 2e-2
+
+
+@pytest.mark.parametrize("S,causal", [(256, True), (192, True), (128, False)])
+def test_attention_bwd(dev, S, causal):
+    ops = _hip()
+    B, Hq, Hkv, D = 2, 8, 2, 128
+    q = torch.randn(B, S, Hq, D, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(B, S, Hkv, D, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(B, S, Hkv, D, device=dev, dtype=torch.bfloat16, requires_grad=True)
+    o = ops.flash_attention(q, k, v, causal=causal)
+    do = torch.randn_like(o)
+    o.backward(do)
+
+    from torchx_amd.ops import reference
+
+    qr = q.detach().float().requires_grad_(True)
+    kr = k.detach().float().requires_grad_(True)
+    vr = v.detach().float().requires_grad_(True)
+    g = Hq // Hkv
+    qf = qr.permute(0, 2, 1, 3)
+    kf = kr.permute(0, 2, 1, 3).repeat_interleave(g, dim=1)
+    vf = vr.permute(0, 2, 1, 3).repeat_interleave(g, dim=1)
+    s = qf @ kf.transpose(-1, -2) / math.sqrt(D)
+    if causal:
+        mask = torch.ones(S, S, dtype=torch.bool, device=dev).triu(1)
+        s = s.masked_fill(mask, float("-inf"))
+    o_ref = (torch.softmax(s, -1) @ vf).permute(0, 2, 1, 3)
+    o_ref.backward(do.float())
+    assert rel_err(q.grad, qr.grad) < 4e-2, f"dq err {rel_err(q.grad, qr.grad)}"
+    assert rel_err(k.grad, kr.grad) < 4e-2, f"dk err {rel_err(k.grad, kr.grad)}"
+    assert rel_err(v.grad, vr.grad) < 4e-2, f"dv err {rel_err(v.grad, vr.grad)}"
+
+
+def test_adamw(dev):
+    ops = _hip()
+    from torchx_amd.ops import reference
+
+    n = 8192
+    p32 = torch.randn(n, device=dev)
+    p16 = p32.to(torch.bfloat16)
+    g = torch.randn(n, device=dev, dtype=torch.bfloat16)
+    m = torch.zeros(n, device=dev)
+    v = torch.zeros(n, device=dev)
+    p32r, p16r, mr, vr = p32.clone(), p16.clone(), m.clone(), v.clone()
+    for step in (1, 2, 3):
+        ops.adamw_step(p32, p16, g, m, v, lr=1e-3, step=step)
+        reference.adamw_step(p32r, p16r, g, mr, vr, lr=1e-3, beta1=0.9,
+                             beta2=0.95, eps=1e-8, weight_decay=0.1, step=step)
+    assert rel_err(p32, p32r) < 1e-4
+    assert rel_err(m, mr) < 1e-4
+    assert rel_err(v, vr) < 1e-4
+    assert rel_err(p16, p16r) < 1e-2

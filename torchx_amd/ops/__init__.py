@@ -1,0 +1,246 @@
+"""torchx_amd.ops — CDNA4 HIP hot-op library with autograd bindings.
+
+On a GPU (MI355X) these ops REQUIRE the in-tree `_hip_ops.so` extension and
+fail loudly if it is missing — there is no silent eager fallback on the GPU
+path.  On CPU (CI containers have no GPU) they fall back to the fp32 torch
+reference implementations in `torchx_amd.ops.reference`, which are also the
+ground truth the GPU numerics tests compare against.
+"""
+
+from __future__ import annotations
+
+import importlib.util
+import math
+from pathlib import Path
+from typing import Optional, Tuple
+
+import torch
+
+from . import reference
+
+_SO = Path(__file__).resolve().parent / "_hip_ops.so"
+_hip = None
+_load_err: Optional[str] = None
+
+
+def _load():
+    global _hip, _load_err
+    if _hip is not None:
+        return _hip
+    if not _SO.exists():
+        _load_err = f"{_SO} not built (run torchx_amd/ops/build.py)"
+        return None
+    try:
+        spec = importlib.util.spec_from_file_location("torchx_amd_hip_ops", _SO)
+        mod = importlib.util.module_from_spec(spec)
+        spec.loader.exec_module(mod)  # type: ignore[union-attr]
+        _hip = mod
+    except Exception as e:  # noqa: BLE001
+        _load_err = str(e)
+    return _hip
+
+
+def hip_ops(required: bool = True):
+    """The raw extension module.  On a GPU box this must exist."""
+    mod = _load()
+    if mod is None and required:
+        raise RuntimeError(
+            f"torchx_amd HIP extension not available: {_load_err}. "
+            "GPU execution requires the native kernels; refusing to fall "
+            "back to eager."
+        )
+    return mod
+
+
+def extension_available() -> bool:
+    return _load() is not None
+
+
+def _on_gpu(*ts: torch.Tensor) -> bool:
+    return any(t.is_cuda for t in ts)
+
+
+# ---------------------------------------------------------------------------
+# RMSNorm
+# ---------------------------------------------------------------------------
+
+
+class _RMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, eps: float):
+        y, invrms = hip_ops().rmsnorm_fwd(x.contiguous(), w.contiguous(), eps)
+        ctx.save_for_backward(x, w, invrms)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        x, w, invrms = ctx.saved_tensors
+        dx, dw = hip_ops().rmsnorm_bwd(dy.contiguous(), x, w, invrms)
+        return dx, dw.to(w.dtype), None
+
+
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if _on_gpu(x):
+        return _RMSNorm.apply(x, w, eps)
+    return reference.rmsnorm(x, w, eps)
+
+
+# ---------------------------------------------------------------------------
+# RoPE
+# ---------------------------------------------------------------------------
+
+
+def rope_tables(
+    seq_len: int, head_dim: int, theta: float = 500000.0,
+    device: Optional[torch.device] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Host-precomputed f32 cos/sin tables [S, D/2] (guide: on-device trig
+    turns a memory-bound elementwise op VALU-bound)."""
+    inv_freq = 1.0 / (
+        theta ** (torch.arange(0, head_dim, 2, dtype=torch.float32) / head_dim)
+    )
+    pos = torch.arange(seq_len, dtype=torch.float32)
+    ang = torch.outer(pos, inv_freq)  # [S, D/2]
+    cos, sin = torch.cos(ang), torch.sin(ang)
+    if device is not None:
+        cos, sin = cos.to(device), sin.to(device)
+    return cos, sin
+
+
+class _RoPE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                heads: int):
+        ctx.save_for_backward(cos, sin)
+        ctx.heads = heads
+        return hip_ops().rope(x.contiguous(), cos, sin, heads, 1.0)
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        cos, sin = ctx.saved_tensors
+        dx = hip_ops().rope(dy.contiguous(), cos, sin, ctx.heads, -1.0)
+        return dx, None, None, None
+
+
+def rope(x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+    """x: [B, S, H, D] bf16; rotate-half (Llama) convention."""
+    heads = x.shape[2]
+    if _on_gpu(x):
+        return _RoPE.apply(x, cos, sin, heads)
+    return reference.rope(x, cos, sin)
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU
+# ---------------------------------------------------------------------------
+
+
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, g: torch.Tensor, u: torch.Tensor):
+        ctx.save_for_backward(g, u)
+        return hip_ops().swiglu_fwd(g.contiguous(), u.contiguous())
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        g, u = ctx.saved_tensors
+        dg, du = hip_ops().swiglu_bwd(dout.contiguous(), g, u)
+        return dg, du
+
+
+def swiglu(g: torch.Tensor, u: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(g):
+        return _SwiGLU.apply(g, u)
+    return reference.swiglu(g, u)
+
+
+# ---------------------------------------------------------------------------
+# Fused cross entropy (mean reduction over T tokens)
+# ---------------------------------------------------------------------------
+
+
+class _CrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, targets: torch.Tensor):
+        V = logits.shape[-1]
+        logits2d = logits.contiguous().view(-1, V)
+        loss, lse = hip_ops().ce_fwd(logits2d, targets.contiguous().view(-1))
+        ctx.save_for_backward(logits2d, targets, lse)
+        ctx.orig_shape = logits.shape
+        return loss.mean()
+
+    @staticmethod
+    def backward(ctx, gout: torch.Tensor):
+        logits2d, targets, lse = ctx.saved_tensors
+        T = logits2d.shape[0]
+        gscale = (gout.float() / T).expand(T).contiguous()
+        dlogits = hip_ops().ce_bwd(logits2d, targets.view(-1), lse, gscale)
+        return dlogits.view(ctx.orig_shape), None
+
+
+def cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
+    if _on_gpu(logits):
+        return _CrossEntropy.apply(logits, targets)
+    return reference.cross_entropy(logits, targets)
+
+
+# ---------------------------------------------------------------------------
+# Flash attention (causal, GQA), BSHD layout
+# ---------------------------------------------------------------------------
+
+
+class _FlashAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                causal: bool, scale: float):
+        o, lse = hip_ops().attn_fwd(
+            q.contiguous(), k.contiguous(), v.contiguous(), scale, causal
+        )
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = hip_ops().attn_bwd(
+            q, k, v, o, dout.contiguous(), lse, ctx.scale, ctx.causal
+        )
+        return dq, dk, dv, None, None
+
+
+def flash_attention(
+    q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+    causal: bool = True, scale: Optional[float] = None,
+) -> torch.Tensor:
+    """q [B,S,Hq,D], k/v [B,S,Hkv,D] bf16 -> o [B,S,Hq,D]."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        return _FlashAttention.apply(q, k, v, causal, scale)
+    return reference.attention(q, k, v, causal=causal, scale=scale)
+
+
+# ---------------------------------------------------------------------------
+# Fused AdamW (flat-buffer optimizer step; see torchx_amd.parallel.optim)
+# ---------------------------------------------------------------------------
+
+
+def adamw_step(
+    p32: torch.Tensor, p16: torch.Tensor, grad: torch.Tensor,
+    m: torch.Tensor, v: torch.Tensor, *, lr: float, beta1: float = 0.9,
+    beta2: float = 0.95, eps: float = 1e-8, weight_decay: float = 0.1,
+    step: int = 1,
+) -> None:
+    if p32.is_cuda:
+        hip_ops().adamw_step(p32, p16, grad, m, v, lr, beta1, beta2, eps,
+                             weight_decay, step)
+    else:
+        reference.adamw_step(p32, p16, grad, m, v, lr=lr, beta1=beta1,
+                             beta2=beta2, eps=eps, weight_decay=weight_decay,
+                             step=step)
+
+
+def mfma_probe(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    return hip_ops().mfma_probe(a, b)

@@ -1,0 +1,658 @@
+// Flash-style fused attention (forward + backward) for CDNA4 / gfx950.
+//
+// MFMA tiling with v_mfma_f32_32x32x16_bf16 (2xK gfx950 form), online
+// softmax, causal masking, GQA.  Head dim fixed at 128 (Llama-3/Mixtral).
+// Layouts are BSHD: q [B,S,Hq,128], k/v [B,S,Hkv,128], all bf16; lse/delta
+// [B,Hq,S] f32.
+//
+// Design notes (see /opt/skills/guides/cdna_hip_programming.md Appendix B):
+//  * "swapped QK^T": S^T = mfma(A=K, B=Q) so each lane owns ONE q row's
+//    scores (C-layout col = lane&31) -> softmax is in-register (16 regs +
+//    one shfl_xor(32) partner exchange), no cross-lane LDS reduction.
+//  * P (f32, C-layout) -> bf16 MFMA fragments via pack + lane-half exchange
+//    (the guide's cvt_pk + permlane32_swap idiom; v1 uses shfl_xor).
+//  * PV uses A=V^T (from a transposed LDS image) so the O accumulator stays
+//    in the lane-owns-q-column layout: per-row rescale/divide are per-lane
+//    scalars.
+//  * Backward is FA2-style: preprocess delta = rowsum(dO*O); dQ kernel
+//    (grid over q tiles, loop kv); dK/dV kernel (grid over kv tiles of 128
+//    rows, loop q tiles x GQA group) -> no atomics anywhere.
+//
+// MFMA fragment maps used (verified against rocm CK xdlops_gemm.hpp and the
+// CDNA4 guide; C/D map from the guide):
+//   A[m=32, k=16]: lane l holds A[l&31][(l>>5)*8 + j], j=0..7  (bf16x8)
+//   B[k=16, n=32]: lane l holds B[(l>>5)*8 + j][l&31]
+//   C[m=32, n=32]: lane l holds C[(r&3) + 8*(r>>2) + 4*(l>>5)][l&31], r=0..15
+#include "common.h"
+
+typedef __bf16 mbf16x8 __attribute__((ext_vector_type(8)));
+
+#define HD 128           // head dim
+#define QBLK 32          // q rows per wave
+#define KBLK 32          // kv rows per tile
+#define NWAVES 4         // waves per block
+#define BLOCK_Q (QBLK * NWAVES)  // q rows per block (fwd / dq)
+#define BLOCK_K (KBLK * NWAVES)  // kv rows per block (dkv)
+
+// padded LDS strides (bank-conflict-free ds_read_b128, see guide §6 G4)
+#define NAT_STRIDE 136   // [32][136] natural tile (row-major, 8-col pad)
+#define TR_STRIDE 40     // [128][40] transposed tile (d-major, 8-col pad)
+
+__device__ __forceinline__ f32x16 mfma32(mbf16x8 a, mbf16x8 b, f32x16 c) {
+  return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+__device__ __forceinline__ unsigned int pack_bf16x2(float a, float b) {
+  return (unsigned int)f32_to_bf16(a) |
+         ((unsigned int)f32_to_bf16(b) << 16);
+}
+
+// C-layout row index of accumulator register r for this lane-half.
+__device__ __forceinline__ int c_row(int r, int half) {
+  return (r & 3) + 8 * (r >> 2) + 4 * half;
+}
+
+// Build the two k-slice MFMA fragments (slice s covers k in [16s, 16s+16))
+// from 16 f32 values held in C-layout.  Lane-half exchange via shfl_xor(32).
+__device__ __forceinline__ void cvals_to_frags(const float* p, bool hi,
+                                               mbf16x8* f0, mbf16x8* f1) {
+  unsigned int w[2][4];
+  #pragma unroll
+  for (int s = 0; s < 2; ++s) {
+    unsigned int a01 = pack_bf16x2(p[8 * s + 0], p[8 * s + 1]);
+    unsigned int a23 = pack_bf16x2(p[8 * s + 2], p[8 * s + 3]);
+    unsigned int a45 = pack_bf16x2(p[8 * s + 4], p[8 * s + 5]);
+    unsigned int a67 = pack_bf16x2(p[8 * s + 6], p[8 * s + 7]);
+    unsigned int x01 = __shfl_xor((int)a01, 32, 64);
+    unsigned int x23 = __shfl_xor((int)a23, 32, 64);
+    unsigned int x45 = __shfl_xor((int)a45, 32, 64);
+    unsigned int x67 = __shfl_xor((int)a67, 32, 64);
+    w[s][0] = hi ? x45 : a01;
+    w[s][1] = hi ? x67 : a23;
+    w[s][2] = hi ? a45 : x01;
+    w[s][3] = hi ? a67 : x23;
+  }
+  uint4_v u0 = {w[0][0], w[0][1], w[0][2], w[0][3]};
+  uint4_v u1 = {w[1][0], w[1][1], w[1][2], w[1][3]};
+  *f0 = __builtin_bit_cast(mbf16x8, u0);
+  *f1 = __builtin_bit_cast(mbf16x8, u1);
+}
+
+// A/B-style fragment read from an LDS tile: lane l -> row (l&31),
+// cols [col0 + (l>>5)*8, +8).  One ds_read_b128 when 16-B aligned.
+__device__ __forceinline__ mbf16x8 lds_frag(const unsigned short* tile,
+                                            int row_stride, int col0) {
+  const int lane = threadIdx.x & 63;
+  const unsigned short* p =
+      tile + (lane & 31) * row_stride + col0 + (lane >> 5) * 8;
+  return __builtin_bit_cast(mbf16x8, *(const ushort8*)p);
+}
+
+// Natural-layout cooperative staging of a [rows<=32][128] bf16 tile from a
+// BSHD tensor into LDS (guarded, zero-filled beyond seq).  256 threads.
+__device__ __forceinline__ void stage_nat(
+    const unsigned short* __restrict__ src,  // tensor base
+    long row0,                                // first seq row
+    long seq_stride,                          // H*128 (elements)
+    int rows, int S, unsigned short* dst) {
+  const int id0 = threadIdx.x;               // 512 chunks of 8, 2 per thread
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int id = id0 + p * 256;
+    int r = id / 16, c8 = (id % 16) * 8;
+    ushort8 v;
+    if (r < rows && row0 + r < S) {
+      v = *(const ushort8*)(src + (row0 + r) * seq_stride + c8);
+    } else {
+      v = (ushort8)0;
+    }
+    *(ushort8*)(dst + r * NAT_STRIDE + c8) = v;
+  }
+}
+
+// Transposed staging: dst[d][k] = src[k][d], dst is [128][TR_STRIDE].
+__device__ __forceinline__ void stage_tr(
+    const unsigned short* __restrict__ src, long row0, long seq_stride,
+    int rows, int S, unsigned short* dst, float scale) {
+  const int id0 = threadIdx.x;
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    int id = id0 + p * 256;
+    int r = id / 16, c8 = (id % 16) * 8;
+    ushort8 v;
+    if (r < rows && row0 + r < S) {
+      v = *(const ushort8*)(src + (row0 + r) * seq_stride + c8);
+    } else {
+      v = (ushort8)0;
+    }
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      unsigned short u = v[j];
+      if (scale != 1.0f) u = f32_to_bf16(bf16_to_f32(u) * scale);
+      dst[(c8 + j) * TR_STRIDE + r] = u;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Forward
+// ---------------------------------------------------------------------------
+
+__global__ void __launch_bounds__(256, 2)
+attn_fwd_kernel(const unsigned short* __restrict__ q,
+                const unsigned short* __restrict__ k,
+                const unsigned short* __restrict__ v,
+                unsigned short* __restrict__ o,
+                float* __restrict__ lse,  // [B,Hq,S]
+                int B, int S, int Hq, int Hkv, float scale, int causal) {
+  __shared__ __align__(16) unsigned short smem[KBLK * NAT_STRIDE + HD * TR_STRIDE];
+  unsigned short* k_nat = smem;
+  unsigned short* v_tr = smem + KBLK * NAT_STRIDE;
+
+  const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
+  int bid = blockIdx.x;
+  const int b = bid / (Hq * nqt);
+  bid -= b * Hq * nqt;
+  const int hq = bid / nqt;
+  const int qt = bid % nqt;
+  const int hkv = hq / (Hq / Hkv);
+
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const bool hi = lane >= 32;
+  const int qr = lane & 31;
+
+  const long q_seq_stride = (long)Hq * HD;
+  const long kv_seq_stride = (long)Hkv * HD;
+  const unsigned short* qb = q + (long)b * S * q_seq_stride + (long)hq * HD;
+  const unsigned short* kb = k + (long)b * S * kv_seq_stride + (long)hkv * HD;
+  const unsigned short* vb = v + (long)b * S * kv_seq_stride + (long)hkv * HD;
+
+  const int q0_blk = qt * BLOCK_Q;
+  const int qw0 = q0_blk + wave * QBLK;       // wave's first q row
+  const long q_row = qw0 + qr;                 // this lane's q row
+  const bool wave_active = qw0 < S;
+
+  // Q fragments in registers: chunk c covers d in [16c, 16c+16)
+  mbf16x8 qfrag[8];
+  {
+    const long row = (q_row < S) ? q_row : (S - 1);
+    const unsigned short* qp = qb + row * q_seq_stride + (hi ? 8 : 0);
+    #pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      qfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(qp + c * 16));
+    }
+  }
+
+  f32x16 acc_o[4] = {};
+  float m_run = -3.0e38f, l_run = 0.f;
+
+  const int kv_limit = causal ? min(S, q0_blk + BLOCK_Q) : S;
+  const int ntiles = (kv_limit + KBLK - 1) / KBLK;
+  const int qw_max = min(qw0 + QBLK - 1, S - 1);
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int kv0 = t * KBLK;
+    __syncthreads();
+    stage_nat(kb, kv0, kv_seq_stride, KBLK, S, k_nat);
+    stage_tr(vb, kv0, kv_seq_stride, KBLK, S, v_tr, 1.0f);
+    __syncthreads();
+
+    const bool needed = wave_active && (!causal || kv0 <= qw_max);
+    if (!needed) continue;
+
+    // S^T[k, q] = K . Q^T
+    f32x16 acc = {};
+    #pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      acc = mfma32(lds_frag(k_nat, NAT_STRIDE, c * 16), qfrag[c], acc);
+    }
+
+    float sv[16];
+    const bool mask_tile =
+        (causal && kv0 + KBLK - 1 > qw0) || (kv0 + KBLK > S) || (q_row >= S);
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      float s = acc[r] * scale;
+      if (mask_tile) {
+        const long kg = kv0 + c_row(r, hi);
+        if (kg >= S || q_row >= S || (causal && kg > q_row)) s = -3.0e38f;
+      }
+      sv[r] = s;
+    }
+
+    // online softmax (lane owns q row; partner holds the other 16 k's)
+    float m_tile = sv[0];
+    #pragma unroll
+    for (int r = 1; r < 16; ++r) m_tile = fmaxf(m_tile, sv[r]);
+    m_tile = fmaxf(m_tile, __shfl_xor(m_tile, 32, 64));
+    const float m_new = fmaxf(m_run, m_tile);
+    const float alpha = __expf(m_run - m_new);
+
+    float p[16], row_sum = 0.f;
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      p[r] = __expf(sv[r] - m_new);
+      row_sum += p[r];
+    }
+    row_sum += __shfl_xor(row_sum, 32, 64);
+    l_run = l_run * alpha + row_sum;
+    m_run = m_new;
+    #pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) acc_o[dt][r] *= alpha;
+    }
+
+    mbf16x8 pf0, pf1;
+    cvals_to_frags(p, hi, &pf0, &pf1);
+
+    // O^T[d, q] += V^T . P  (A = V^T from transposed LDS image)
+    #pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+      acc_o[dt] = mfma32(lds_frag(v_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
+                         pf0, acc_o[dt]);
+      acc_o[dt] = mfma32(lds_frag(v_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
+                         pf1, acc_o[dt]);
+    }
+  }
+
+  if (!wave_active || q_row >= S) return;
+
+  const float inv_l = (l_run > 0.f) ? 1.0f / l_run : 0.f;
+  unsigned short* ob = o + ((long)b * S + q_row) * q_seq_stride + (long)hq * HD;
+  #pragma unroll
+  for (int dt = 0; dt < 4; ++dt) {
+    #pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      // regs 4g..4g+3 are consecutive d: d = dt*32 + 8g + 4*hi + (0..3)
+      bf16x4_raw w;
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        w[j] = (short)f32_to_bf16(acc_o[dt][4 * g + j] * inv_l);
+      }
+      *(bf16x4_raw*)(ob + dt * 32 + 8 * g + 4 * (hi ? 1 : 0)) = w;
+    }
+  }
+  if (!hi) {
+    lse[((long)b * Hq + hq) * S + q_row] = m_run + __logf(l_run);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward preprocess: delta[b,h,s] = rowsum(dO * O)
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256)
+attn_bwd_pre_kernel(const unsigned short* __restrict__ dout,
+                    const unsigned short* __restrict__ o,
+                    float* __restrict__ delta,  // [B,Hq,S]
+                    long rows,  // B*S*Hq
+                    int S, int Hq) {
+  // one wave per row; lane covers 2 elements of HD=128
+  const long row = (long)blockIdx.x * 4 + threadIdx.x / 64;
+  if (row >= rows) return;
+  const int lane = threadIdx.x & 63;
+  const unsigned short* dp = dout + row * HD + lane * 2;
+  const unsigned short* op = o + row * HD + lane * 2;
+  float acc = bf16_to_f32(dp[0]) * bf16_to_f32(op[0]) +
+              bf16_to_f32(dp[1]) * bf16_to_f32(op[1]);
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) {
+    const long b = row / ((long)S * Hq);
+    const long s = (row / Hq) % S;
+    const long h = row % Hq;
+    delta[(b * Hq + h) * S + s] = acc;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward dQ: grid over q tiles; inner loop over kv tiles.
+// dQ[q,d] = scale * sum_k (P*(dP - delta))[q,k] * K[k,d]
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256, 2)
+attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
+                   const unsigned short* __restrict__ k,
+                   const unsigned short* __restrict__ v,
+                   const unsigned short* __restrict__ dout,
+                   const float* __restrict__ lse,
+                   const float* __restrict__ delta,
+                   unsigned short* __restrict__ dq,
+                   int B, int S, int Hq, int Hkv, float scale, int causal) {
+  __shared__ __align__(16) unsigned short smem[KBLK * NAT_STRIDE + HD * TR_STRIDE +
+                                 KBLK * NAT_STRIDE];
+  unsigned short* k_nat = smem;
+  unsigned short* k_tr = smem + KBLK * NAT_STRIDE;                // [128][40]
+  unsigned short* v_nat = k_tr + HD * TR_STRIDE;
+
+  const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
+  int bid = blockIdx.x;
+  const int b = bid / (Hq * nqt);
+  bid -= b * Hq * nqt;
+  const int hq = bid / nqt;
+  const int qt = bid % nqt;
+  const int hkv = hq / (Hq / Hkv);
+
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const bool hi = lane >= 32;
+  const int qr = lane & 31;
+
+  const long q_seq_stride = (long)Hq * HD;
+  const long kv_seq_stride = (long)Hkv * HD;
+  const unsigned short* qb = q + (long)b * S * q_seq_stride + (long)hq * HD;
+  const unsigned short* kb = k + (long)b * S * kv_seq_stride + (long)hkv * HD;
+  const unsigned short* vb = v + (long)b * S * kv_seq_stride + (long)hkv * HD;
+  const unsigned short* dob = dout + (long)b * S * q_seq_stride + (long)hq * HD;
+
+  const int q0_blk = qt * BLOCK_Q;
+  const int qw0 = q0_blk + wave * QBLK;
+  const long q_row = qw0 + qr;
+  const bool wave_active = qw0 < S;
+  const bool row_valid = q_row < S;
+
+  mbf16x8 qfrag[8], dofrag[8];
+  float my_lse = 0.f, my_delta = 0.f;
+  {
+    const long row = row_valid ? q_row : (S - 1);
+    const unsigned short* qp = qb + row * q_seq_stride + (hi ? 8 : 0);
+    const unsigned short* dp = dob + row * q_seq_stride + (hi ? 8 : 0);
+    #pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      qfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(qp + c * 16));
+      dofrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(dp + c * 16));
+    }
+    if (row_valid) {
+      my_lse = lse[((long)b * Hq + hq) * S + q_row];
+      my_delta = delta[((long)b * Hq + hq) * S + q_row];
+    }
+  }
+
+  f32x16 acc_dq[4] = {};
+
+  const int kv_limit = causal ? min(S, q0_blk + BLOCK_Q) : S;
+  const int ntiles = (kv_limit + KBLK - 1) / KBLK;
+  const int qw_max = min(qw0 + QBLK - 1, S - 1);
+
+  for (int t = 0; t < ntiles; ++t) {
+    const int kv0 = t * KBLK;
+    __syncthreads();
+    stage_nat(kb, kv0, kv_seq_stride, KBLK, S, k_nat);
+    stage_tr(kb, kv0, kv_seq_stride, KBLK, S, k_tr, 1.0f);
+    stage_nat(vb, kv0, kv_seq_stride, KBLK, S, v_nat);
+    __syncthreads();
+
+    const bool needed = wave_active && (!causal || kv0 <= qw_max);
+    if (!needed) continue;
+
+    f32x16 acc_s = {}, acc_dp = {};
+    #pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      acc_s = mfma32(lds_frag(k_nat, NAT_STRIDE, c * 16), qfrag[c], acc_s);
+      acc_dp = mfma32(lds_frag(v_nat, NAT_STRIDE, c * 16), dofrag[c], acc_dp);
+    }
+
+    const bool mask_tile =
+        (causal && kv0 + KBLK - 1 > qw0) || (kv0 + KBLK > S) || !row_valid;
+    float ds[16];
+    #pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      float s = acc_s[r] * scale;
+      if (mask_tile) {
+        const long kg = kv0 + c_row(r, hi);
+        if (kg >= S || !row_valid || (causal && kg > q_row)) s = -3.0e38f;
+      }
+      const float pr = __expf(s - my_lse);
+      ds[r] = scale * pr * (acc_dp[r] - my_delta);
+    }
+
+    mbf16x8 df0, df1;
+    cvals_to_frags(ds, hi, &df0, &df1);
+
+    // dQ^T[d, q] += K^T . dS^T
+    #pragma unroll
+    for (int dt = 0; dt < 4; ++dt) {
+      acc_dq[dt] = mfma32(lds_frag(k_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
+                          df0, acc_dq[dt]);
+      acc_dq[dt] = mfma32(lds_frag(k_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
+                          df1, acc_dq[dt]);
+    }
+  }
+
+  if (!wave_active || !row_valid) return;
+  unsigned short* dqb =
+      dq + ((long)b * S + q_row) * q_seq_stride + (long)hq * HD;
+  #pragma unroll
+  for (int dt = 0; dt < 4; ++dt) {
+    #pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      bf16x4_raw w;
+      #pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        w[j] = (short)f32_to_bf16(acc_dq[dt][4 * g + j]);
+      }
+      *(bf16x4_raw*)(dqb + dt * 32 + 8 * g + 4 * (hi ? 1 : 0)) = w;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward dK/dV: grid over kv blocks of 128 rows (wave owns 32); inner loop
+// over GQA group heads x q tiles.  No atomics: each (b, hkv, kv-row) is
+// owned by exactly one wave.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(256, 1)
+attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
+                    const unsigned short* __restrict__ k,
+                    const unsigned short* __restrict__ v,
+                    const unsigned short* __restrict__ dout,
+                    const float* __restrict__ lse,
+                    const float* __restrict__ delta,
+                    unsigned short* __restrict__ dk,
+                    unsigned short* __restrict__ dv,
+                    int B, int S, int Hq, int Hkv, float scale, int causal) {
+  // staging area: Q nat + dO nat + Q^T + dO^T + lse/delta rows
+  __shared__ __align__(16) unsigned short smem[2 * QBLK * NAT_STRIDE + 2 * HD * TR_STRIDE];
+  __shared__ float lse_s[QBLK], del_s[QBLK];
+  unsigned short* q_nat = smem;
+  unsigned short* do_nat = smem + QBLK * NAT_STRIDE;
+  unsigned short* q_tr = do_nat + QBLK * NAT_STRIDE;
+  unsigned short* do_tr = q_tr + HD * TR_STRIDE;
+
+  const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
+  int bid = blockIdx.x;
+  const int b = bid / (Hkv * nkt);
+  bid -= b * Hkv * nkt;
+  const int hkv = bid / nkt;
+  const int kt = bid % nkt;
+  const int G = Hq / Hkv;
+
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const bool hi = lane >= 32;
+  const int kr = lane & 31;
+
+  const long q_seq_stride = (long)Hq * HD;
+  const long kv_seq_stride = (long)Hkv * HD;
+  const unsigned short* kb = k + (long)b * S * kv_seq_stride + (long)hkv * HD;
+  const unsigned short* vb = v + (long)b * S * kv_seq_stride + (long)hkv * HD;
+
+  const int kv0_blk = kt * BLOCK_K;
+  const int kw0 = kv0_blk + wave * KBLK;       // wave's first kv row
+  const long k_row = kw0 + kr;                  // lane's kv row
+  const bool wave_active = kw0 < S;
+  const bool krow_valid = k_row < S;
+
+  // K, V fragments resident in registers (lane = k row)
+  mbf16x8 kfrag[8], vfrag[8];
+  {
+    const long row = krow_valid ? k_row : (S - 1);
+    const unsigned short* kp = kb + row * kv_seq_stride + (hi ? 8 : 0);
+    const unsigned short* vp = vb + row * kv_seq_stride + (hi ? 8 : 0);
+    #pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      kfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(kp + c * 16));
+      vfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(vp + c * 16));
+    }
+  }
+
+  f32x16 acc_dk[4] = {};
+  f32x16 acc_dv[4] = {};
+
+  const int t0 = causal ? (kv0_blk / QBLK) : 0;
+  const int nt = (S + QBLK - 1) / QBLK;
+
+  for (int gh = 0; gh < G; ++gh) {
+    const int hq = hkv * G + gh;
+    const unsigned short* qb = q + (long)b * S * q_seq_stride + (long)hq * HD;
+    const unsigned short* dob =
+        dout + (long)b * S * q_seq_stride + (long)hq * HD;
+    const float* lse_b = lse + ((long)b * Hq + hq) * S;
+    const float* del_b = delta + ((long)b * Hq + hq) * S;
+
+    for (int t = t0; t < nt; ++t) {
+      const int q0 = t * QBLK;
+      __syncthreads();
+      stage_nat(qb, q0, q_seq_stride, QBLK, S, q_nat);
+      stage_nat(dob, q0, q_seq_stride, QBLK, S, do_nat);
+      stage_tr(qb, q0, q_seq_stride, QBLK, S, q_tr, 1.0f);
+      stage_tr(dob, q0, q_seq_stride, QBLK, S, do_tr, 1.0f);
+      if (threadIdx.x < QBLK) {
+        const int qg = q0 + threadIdx.x;
+        lse_s[threadIdx.x] = (qg < S) ? lse_b[qg] : 0.f;
+        del_s[threadIdx.x] = (qg < S) ? del_b[qg] : 0.f;
+      }
+      __syncthreads();
+
+      // wave needs this q tile iff its last q row >= wave's first kv row
+      const bool needed =
+          wave_active && (!causal || q0 + QBLK - 1 >= kw0);
+      if (!needed) continue;
+
+      // S[q, k] (lane = k col), dP[q, k]
+      f32x16 acc_s = {}, acc_dp = {};
+      #pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        acc_s = mfma32(lds_frag(q_nat, NAT_STRIDE, c * 16), kfrag[c], acc_s);
+        acc_dp =
+            mfma32(lds_frag(do_nat, NAT_STRIDE, c * 16), vfrag[c], acc_dp);
+      }
+
+      const bool mask_tile = causal || (q0 + QBLK > S) || !krow_valid;
+      float pv[16], dsv[16];
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qg = q0 + c_row(r, hi);
+        float s = acc_s[r] * scale;
+        if (mask_tile) {
+          if (qg >= S || !krow_valid || (causal && k_row > qg)) s = -3.0e38f;
+        }
+        const float pr = __expf(s - lse_s[qg - q0]);
+        pv[r] = pr;
+        dsv[r] = scale * pr * (acc_dp[r] - del_s[qg - q0]);
+      }
+
+      mbf16x8 pf0, pf1, df0, df1;
+      cvals_to_frags(pv, hi, &pf0, &pf1);
+      cvals_to_frags(dsv, hi, &df0, &df1);
+
+      // dV[k, d] += P^T . dO ; dK[k, d] += dS^T . Q
+      #pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        acc_dv[dt] =
+            mfma32(pf0, lds_frag(do_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
+                   acc_dv[dt]);
+        acc_dv[dt] =
+            mfma32(pf1, lds_frag(do_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
+                   acc_dv[dt]);
+        acc_dk[dt] =
+            mfma32(df0, lds_frag(q_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
+                   acc_dk[dt]);
+        acc_dk[dt] =
+            mfma32(df1, lds_frag(q_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
+                   acc_dk[dt]);
+      }
+    }
+  }
+
+  // Store via LDS transpose so global writes are 8-B vectors.
+  // Each wave owns a private [32 k][36 d] f32 region (reuses staging smem).
+  __syncthreads();
+  float* tr = (float*)smem + wave * (32 * 36);
+  if (wave_active) {
+    #pragma unroll
+    for (int which = 0; which < 2; ++which) {
+      f32x16* acc = which == 0 ? acc_dv : acc_dk;
+      unsigned short* out = which == 0 ? dv : dk;
+      #pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        __syncthreads();
+        // write C-layout regs: row k = c_row(r, hi), col d in [0,32)
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          tr[c_row(r, hi) * 36 + kr] = acc[dt][r];
+        }
+        __syncthreads();
+        // read back row-per-lane: each half covers 16 of the 32 d columns
+        #pragma unroll
+        for (int rep = 0; rep < 4; ++rep) {
+          const int kk = kr;
+          const int d0 = (hi ? 16 : 0) + rep * 4;
+          if (kw0 + kk < S) {
+            bf16x4_raw w;
+            #pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              w[j] = (short)f32_to_bf16(tr[kk * 36 + d0 + j]);
+            }
+            *(bf16x4_raw*)(out + ((long)b * S + kw0 + kk) * kv_seq_stride +
+                           (long)hkv * HD + dt * 32 + d0) = w;
+          }
+        }
+      }
+    }
+  } else {
+    // keep barrier counts matched across waves
+    #pragma unroll
+    for (int i = 0; i < 16; ++i) __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Launchers
+// ---------------------------------------------------------------------------
+extern "C" void attn_fwd_launch(const void* q, const void* k, const void* v,
+                                void* o, void* lse, int B, int S, int Hq,
+                                int Hkv, float scale, int causal,
+                                hipStream_t stream) {
+  const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
+  hipLaunchKernelGGL(attn_fwd_kernel, dim3(B * Hq * nqt), dim3(256), 0,
+                     stream, (const unsigned short*)q,
+                     (const unsigned short*)k, (const unsigned short*)v,
+                     (unsigned short*)o, (float*)lse, B, S, Hq, Hkv, scale,
+                     causal);
+}
+
+extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
+                                const void* o, const void* dout,
+                                const void* lse, void* delta, void* dq,
+                                void* dk, void* dv, int B, int S, int Hq,
+                                int Hkv, float scale, int causal,
+                                hipStream_t stream) {
+  const long rows = (long)B * S * Hq;
+  hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3((int)((rows + 3) / 4)),
+                     dim3(256), 0, stream, (const unsigned short*)dout,
+                     (const unsigned short*)o, (float*)delta, rows, S, Hq);
+  const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
+  hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3(B * Hq * nqt), dim3(256), 0,
+                     stream, (const unsigned short*)q,
+                     (const unsigned short*)k, (const unsigned short*)v,
+                     (const unsigned short*)dout, (const float*)lse,
+                     (const float*)delta, (unsigned short*)dq, B, S, Hq, Hkv,
+                     scale, causal);
+  const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
+  hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(B * Hkv * nkt), dim3(256), 0,
+                     stream, (const unsigned short*)q,
+                     (const unsigned short*)k, (const unsigned short*)v,
+                     (const unsigned short*)dout, (const float*)lse,
+                     (const float*)delta, (unsigned short*)dk,
+                     (unsigned short*)dv, B, S, Hq, Hkv, scale, causal);
+}

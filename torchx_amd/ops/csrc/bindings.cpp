@@ -1,0 +1,218 @@
+// Torch extension bindings for the torchx_amd CDNA4 kernels.
+// Compiled with hipcc for gfx950; no CUDA paths, no multi-backend dispatch.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void rope_launch(const void*, void*, const void*, const void*, long, int, int,
+                 int, float, hipStream_t);
+void swiglu_fwd_launch(const void*, const void*, void*, long, hipStream_t);
+void swiglu_bwd_launch(const void*, const void*, const void*, void*, void*,
+                       long, hipStream_t);
+void rmsnorm_fwd_launch(const void*, const void*, void*, void*, long, int,
+                        float, hipStream_t);
+void rmsnorm_bwd_launch(const void*, const void*, const void*, const void*,
+                        void*, void*, long, int, hipStream_t);
+void adamw_launch(void*, void*, const void*, void*, void*, long, float, float,
+                  float, float, float, float, float, hipStream_t);
+void ce_fwd_launch(const void*, const void*, void*, void*, long, int,
+                   hipStream_t);
+void ce_bwd_launch(const void*, const void*, const void*, const void*, void*,
+                   long, int, hipStream_t);
+void attn_fwd_launch(const void*, const void*, const void*, void*, void*, int,
+                     int, int, int, float, int, hipStream_t);
+void attn_bwd_launch(const void*, const void*, const void*, const void*,
+                     const void*, const void*, void*, void*, void*, void*,
+                     int, int, int, int, float, int, hipStream_t);
+void mfma_probe_launch(const void*, const void*, void*, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16, name, " must be bf16");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+void check_f32(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == at::kFloat, name, " must be f32");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+}  // namespace
+
+// ---- RMSNorm --------------------------------------------------------------
+std::vector<at::Tensor> rmsnorm_fwd(at::Tensor x, at::Tensor w, double eps) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const long H = x.size(-1);
+  TORCH_CHECK(H % 2048 == 0 && H <= 8192, "rmsnorm: H must be k*2048, <=8192");
+  const long rows = x.numel() / H;
+  auto y = at::empty_like(x);
+  auto invrms = at::empty({rows}, x.options().dtype(at::kFloat));
+  rmsnorm_fwd_launch(x.data_ptr(), w.data_ptr(), y.data_ptr(),
+                     invrms.data_ptr(), rows, (int)H, (float)eps,
+                     cur_stream());
+  return {y, invrms};
+}
+
+std::vector<at::Tensor> rmsnorm_bwd(at::Tensor dy, at::Tensor x, at::Tensor w,
+                                    at::Tensor invrms) {
+  check_bf16(dy, "dy");
+  check_bf16(x, "x");
+  const long H = x.size(-1);
+  const long rows = x.numel() / H;
+  auto dx = at::empty_like(x);
+  auto dw = at::zeros({H}, x.options().dtype(at::kFloat));
+  rmsnorm_bwd_launch(dy.data_ptr(), x.data_ptr(), w.data_ptr(),
+                     invrms.data_ptr(), dx.data_ptr(), dw.data_ptr(), rows,
+                     (int)H, cur_stream());
+  return {dx, dw};
+}
+
+// ---- RoPE -----------------------------------------------------------------
+at::Tensor rope(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t, long heads,
+                double sign) {
+  check_bf16(x, "x");
+  check_f32(cos_t, "cos");
+  check_f32(sin_t, "sin");
+  const long D = x.size(-1);
+  const long rows = x.numel() / D;
+  const long S = cos_t.size(0);
+  TORCH_CHECK(cos_t.size(1) == D / 2, "cos table must be [S, D/2]");
+  auto y = at::empty_like(x);
+  rope_launch(x.data_ptr(), y.data_ptr(), cos_t.data_ptr(), sin_t.data_ptr(),
+              rows, (int)D, (int)heads, (int)S, (float)sign, cur_stream());
+  return y;
+}
+
+// ---- SwiGLU ---------------------------------------------------------------
+at::Tensor swiglu_fwd(at::Tensor g, at::Tensor u) {
+  check_bf16(g, "g");
+  check_bf16(u, "u");
+  auto out = at::empty_like(g);
+  swiglu_fwd_launch(g.data_ptr(), u.data_ptr(), out.data_ptr(), g.numel(),
+                    cur_stream());
+  return out;
+}
+
+std::vector<at::Tensor> swiglu_bwd(at::Tensor dout, at::Tensor g,
+                                   at::Tensor u) {
+  check_bf16(dout, "dout");
+  auto dg = at::empty_like(g);
+  auto du = at::empty_like(u);
+  swiglu_bwd_launch(dout.data_ptr(), g.data_ptr(), u.data_ptr(), dg.data_ptr(),
+                    du.data_ptr(), g.numel(), cur_stream());
+  return {dg, du};
+}
+
+// ---- AdamW ----------------------------------------------------------------
+void adamw_step(at::Tensor p32, at::Tensor p16, at::Tensor g, at::Tensor m,
+                at::Tensor v, double lr, double beta1, double beta2,
+                double eps, double wd, long step) {
+  check_f32(p32, "p32");
+  check_bf16(p16, "p16");
+  check_bf16(g, "g");
+  const long n = p32.numel();
+  TORCH_CHECK(n % 4 == 0, "adamw: n must be a multiple of 4");
+  const float bc1 = 1.0f - powf((float)beta1, (float)step);
+  const float bc2 = 1.0f - powf((float)beta2, (float)step);
+  adamw_launch(p32.data_ptr(), p16.data_ptr(), g.data_ptr(), m.data_ptr(),
+               v.data_ptr(), n, (float)lr, (float)beta1, (float)beta2,
+               (float)eps, (float)wd, bc1, bc2, cur_stream());
+}
+
+// ---- Cross entropy --------------------------------------------------------
+std::vector<at::Tensor> ce_fwd(at::Tensor logits, at::Tensor targets) {
+  check_bf16(logits, "logits");
+  TORCH_CHECK(targets.scalar_type() == at::kLong, "targets must be int64");
+  const long V = logits.size(-1);
+  const long T = logits.numel() / V;
+  auto loss = at::empty({T}, logits.options().dtype(at::kFloat));
+  auto lse = at::empty({T}, logits.options().dtype(at::kFloat));
+  ce_fwd_launch(logits.data_ptr(), targets.data_ptr(), loss.data_ptr(),
+                lse.data_ptr(), T, (int)V, cur_stream());
+  return {loss, lse};
+}
+
+at::Tensor ce_bwd(at::Tensor logits, at::Tensor targets, at::Tensor lse,
+                  at::Tensor gscale) {
+  check_bf16(logits, "logits");
+  check_f32(lse, "lse");
+  check_f32(gscale, "gscale");
+  const long V = logits.size(-1);
+  const long T = logits.numel() / V;
+  auto dlogits = at::empty_like(logits);
+  ce_bwd_launch(logits.data_ptr(), targets.data_ptr(), lse.data_ptr(),
+                gscale.data_ptr(), dlogits.data_ptr(), T, (int)V,
+                cur_stream());
+  return dlogits;
+}
+
+// ---- Attention ------------------------------------------------------------
+std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 double scale, bool causal) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  check_bf16(v, "v");
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hkv = k.size(2);
+  TORCH_CHECK(D == 128, "attention: head_dim must be 128");
+  TORCH_CHECK(Hq % Hkv == 0, "attention: Hq must be a multiple of Hkv");
+  auto o = at::empty_like(q);
+  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  lse.data_ptr(), B, S, Hq, Hkv, (float)scale, causal ? 1 : 0,
+                  cur_stream());
+  return {o, lse};
+}
+
+std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
+                                 at::Tensor o, at::Tensor dout, at::Tensor lse,
+                                 double scale, bool causal) {
+  check_bf16(dout, "dout");
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2);
+  const int Hkv = k.size(2);
+  auto dq = at::empty_like(q);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  attn_bwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                  dout.data_ptr(), lse.data_ptr(), delta.data_ptr(),
+                  dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), B, S, Hq, Hkv,
+                  (float)scale, causal ? 1 : 0, cur_stream());
+  return {dq, dk, dv};
+}
+
+// ---- probe ----------------------------------------------------------------
+at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  TORCH_CHECK(a.size(0) == 32 && a.size(1) == 16, "a must be [32,16]");
+  TORCH_CHECK(b.size(0) == 16 && b.size(1) == 32, "b must be [16,32]");
+  auto c = at::empty({32, 32}, a.options().dtype(at::kFloat));
+  mfma_probe_launch(a.data_ptr(), b.data_ptr(), c.data_ptr(), cur_stream());
+  return c;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd);
+  m.def("rmsnorm_bwd", &rmsnorm_bwd);
+  m.def("rope", &rope);
+  m.def("swiglu_fwd", &swiglu_fwd);
+  m.def("swiglu_bwd", &swiglu_bwd);
+  m.def("adamw_step", &adamw_step);
+  m.def("ce_fwd", &ce_fwd);
+  m.def("ce_bwd", &ce_bwd);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
+  m.def("mfma_probe", &mfma_probe);
+}

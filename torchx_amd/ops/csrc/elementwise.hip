@@ -1,0 +1,149 @@
+// Element-wise hot ops: RoPE (fwd/bwd) and SwiGLU (fwd/bwd), bf16.
+//
+// Memory-bound: everything is vectorized (8 bf16 = 16 B per lane load) per
+// the CDNA4 guide (scalar bf16 loads are ~2-2.5x slower).  RoPE uses
+// host-precomputed f32 cos/sin tables (on-device trig would turn this
+// VALU-bound).
+//
+// Replaces what the reference (meta-pytorch/torchx) delegates to stock
+// PyTorch ops in its launched apps.
+#include "common.h"
+
+// ---------------------------------------------------------------------------
+// RoPE, rotate-half (Llama) pairing: (i, i + D/2), D = head_dim.
+// x: [rows, D] bf16 where rows = B*S*H, layout [B, S, H, D];
+// cos/sin: [S, D/2] f32; pos_of_row = (row / H) % S.
+// sign=+1 forward, -1 backward (inverse rotation).
+// Each thread: 4 pairs (4 bf16 from each half = 8 B loads).
+// ---------------------------------------------------------------------------
+__global__ void rope_fwdbwd_kernel(
+    const unsigned short* __restrict__ x,
+    unsigned short* __restrict__ y,
+    const float* __restrict__ cos_t,   // [S, D/2]
+    const float* __restrict__ sin_t,   // [S, D/2]
+    long np4,      // (rows * D/2) / 4
+    int half_d,    // D/2, multiple of 4
+    int heads,     // H
+    int seq_len,   // S
+    float sign) {
+  typedef unsigned short us4 __attribute__((ext_vector_type(4)));
+  typedef float f4 __attribute__((ext_vector_type(4)));
+  for (long p4 = grid_stride_begin(); p4 < np4; p4 += grid_stride()) {
+    long p = p4 * 4;
+    long row = p / half_d;
+    int i = (int)(p % half_d);
+    int pos = (int)((row / heads) % seq_len);
+    const long base = row * (2L * half_d);
+    us4 u1 = *(const us4*)(x + base + i);
+    us4 u2 = *(const us4*)(x + base + half_d + i);
+    f4 c = *(const f4*)(cos_t + (long)pos * half_d + i);
+    f4 s = *(const f4*)(sin_t + (long)pos * half_d + i);
+    us4 o1, o2;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float x1 = bf16_to_f32(u1[j]);
+      float x2 = bf16_to_f32(u2[j]);
+      float y1 = fmaf(x1, c[j], -sign * s[j] * x2);
+      float y2 = fmaf(x2, c[j], sign * s[j] * x1);
+      o1[j] = f32_to_bf16(y1);
+      o2[j] = f32_to_bf16(y2);
+    }
+    *(us4*)(y + base + i) = o1;
+    *(us4*)(y + base + half_d + i) = o2;
+  }
+}
+
+extern "C" void rope_launch(
+    const void* x, void* y, const void* cos_t, const void* sin_t,
+    long rows, int head_dim, int heads, int seq_len, float sign,
+    hipStream_t stream) {
+  const int half_d = head_dim / 2;
+  const long np4 = rows * half_d / 4;
+  const int block = 256;
+  int grid = (int)((np4 + block - 1) / block);
+  if (grid > 65535 * 8) grid = 65535 * 8;
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL(rope_fwdbwd_kernel, dim3(grid), dim3(block), 0, stream,
+                     (const unsigned short*)x, (unsigned short*)y,
+                     (const float*)cos_t, (const float*)sin_t, np4, half_d,
+                     heads, seq_len, sign);
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU: out = silu(g) * u   (all bf16, flat n elements, n % 8 == 0)
+// ---------------------------------------------------------------------------
+__global__ void swiglu_fwd_kernel(
+    const unsigned short* __restrict__ g,
+    const unsigned short* __restrict__ u,
+    unsigned short* __restrict__ out,
+    long n8) {
+  for (long i8 = grid_stride_begin(); i8 < n8; i8 += grid_stride()) {
+    long i = i8 * 8;
+    ushort8 gv = *(const ushort8*)(g + i);
+    ushort8 uv = *(const ushort8*)(u + i);
+    ushort8 ov;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf16_to_f32(gv[j]);
+      float uf = bf16_to_f32(uv[j]);
+      float sig = 1.0f / (1.0f + __expf(-gf));
+      ov[j] = f32_to_bf16(gf * sig * uf);
+    }
+    *(ushort8*)(out + i) = ov;
+  }
+}
+
+__global__ void swiglu_bwd_kernel(
+    const unsigned short* __restrict__ dout,
+    const unsigned short* __restrict__ g,
+    const unsigned short* __restrict__ u,
+    unsigned short* __restrict__ dg,
+    unsigned short* __restrict__ du,
+    long n8) {
+  for (long i8 = grid_stride_begin(); i8 < n8; i8 += grid_stride()) {
+    long i = i8 * 8;
+    ushort8 dov = *(const ushort8*)(dout + i);
+    ushort8 gv = *(const ushort8*)(g + i);
+    ushort8 uv = *(const ushort8*)(u + i);
+    ushort8 dgv, duv;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float dof = bf16_to_f32(dov[j]);
+      float gf = bf16_to_f32(gv[j]);
+      float uf = bf16_to_f32(uv[j]);
+      float sig = 1.0f / (1.0f + __expf(-gf));
+      float silu = gf * sig;
+      float dsilu = sig * (1.0f + gf * (1.0f - sig));
+      dgv[j] = f32_to_bf16(dof * uf * dsilu);
+      duv[j] = f32_to_bf16(dof * silu);
+    }
+    *(ushort8*)(dg + i) = dgv;
+    *(ushort8*)(du + i) = duv;
+  }
+}
+
+extern "C" void swiglu_fwd_launch(const void* g, const void* u, void* out,
+                                  long n, hipStream_t stream) {
+  const long n8 = n / 8;
+  const int block = 256;
+  long grid = (n8 + block - 1) / block;
+  if (grid > 65535 * 8) grid = 65535 * 8;
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL(swiglu_fwd_kernel, dim3((int)grid), dim3(block), 0,
+                     stream, (const unsigned short*)g,
+                     (const unsigned short*)u, (unsigned short*)out, n8);
+}
+
+extern "C" void swiglu_bwd_launch(const void* dout, const void* g,
+                                  const void* u, void* dg, void* du, long n,
+                                  hipStream_t stream) {
+  const long n8 = n / 8;
+  const int block = 256;
+  long grid = (n8 + block - 1) / block;
+  if (grid > 65535 * 8) grid = 65535 * 8;
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL(swiglu_bwd_kernel, dim3((int)grid), dim3(block), 0,
+                     stream, (const unsigned short*)dout,
+                     (const unsigned short*)g, (const unsigned short*)u,
+                     (unsigned short*)dg, (unsigned short*)du, n8);
+}

@@ -1,0 +1,139 @@
+// RMSNorm forward/backward, bf16 I/O, f32 accumulation.
+//
+// One 256-thread workgroup per row; row data is register-resident between
+// the square-sum pass and the scale pass (H <= 8192).  LDS tree reduction
+// across the 4 waves.  Memory-bound: vectorized 8-bf16 loads throughout
+// (the guide's RMSNorm number: scalar 2.35 TB/s -> bf16x8 4.89 TB/s).
+#include "common.h"
+
+template <int ITERS>  // ITERS = H / (256 * 8)
+__global__ void __launch_bounds__(256)
+rmsnorm_fwd_kernel(
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ w,
+    unsigned short* __restrict__ y,
+    float* __restrict__ invrms,   // [R], saved for backward
+    int H, float eps) {
+  __shared__ float red[4];
+  const long row = blockIdx.x;
+  const unsigned short* xr = x + row * (long)H;
+  unsigned short* yr = y + row * (long)H;
+
+  float xs[ITERS][8];
+  float acc = 0.f;
+  #pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    int i = (it * 256 + threadIdx.x) * 8;
+    ushort8 v = *(const ushort8*)(xr + i);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf16_to_f32(v[j]);
+      xs[it][j] = f;
+      acc = fmaf(f, f, acc);
+    }
+  }
+  float total = block_reduce<4>(acc, red,
+      [] __device__ (float a, float b) { return a + b; });
+  float r = rsqrtf(total / (float)H + eps);
+  if (threadIdx.x == 0) invrms[row] = r;
+
+  #pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    int i = (it * 256 + threadIdx.x) * 8;
+    ushort8 wv = *(const ushort8*)(w + i);
+    ushort8 ov;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      ov[j] = f32_to_bf16(xs[it][j] * r * bf16_to_f32(wv[j]));
+    }
+    *(ushort8*)(yr + i) = ov;
+  }
+}
+
+// dx = r*(dy*w) - x * (r^3/H) * sum(dy*w*x);  dw += dy * x * r  (f32 atomics)
+template <int ITERS>
+__global__ void __launch_bounds__(256)
+rmsnorm_bwd_kernel(
+    const unsigned short* __restrict__ dy,
+    const unsigned short* __restrict__ x,
+    const unsigned short* __restrict__ w,
+    const float* __restrict__ invrms,
+    unsigned short* __restrict__ dx,
+    float* __restrict__ dw,  // [H] f32, pre-zeroed
+    int H) {
+  __shared__ float red[4];
+  const long row = blockIdx.x;
+  const unsigned short* xr = x + row * (long)H;
+  const unsigned short* dyr = dy + row * (long)H;
+  unsigned short* dxr = dx + row * (long)H;
+  const float r = invrms[row];
+
+  float xs[ITERS][8], dyw[ITERS][8], dwv[ITERS][8];
+  float t = 0.f;
+  #pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    int i = (it * 256 + threadIdx.x) * 8;
+    ushort8 xv = *(const ushort8*)(xr + i);
+    ushort8 dv = *(const ushort8*)(dyr + i);
+    ushort8 wv = *(const ushort8*)(w + i);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float xf = bf16_to_f32(xv[j]);
+      float df = bf16_to_f32(dv[j]);
+      float wf = bf16_to_f32(wv[j]);
+      xs[it][j] = xf;
+      dyw[it][j] = df * wf;
+      dwv[it][j] = df * xf * r;
+      t = fmaf(df * wf, xf, t);
+    }
+  }
+  float ts = block_reduce<4>(t, red,
+      [] __device__ (float a, float b) { return a + b; });
+  const float k = ts * r * r * r / (float)H;
+
+  #pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    int i = (it * 256 + threadIdx.x) * 8;
+    ushort8 ov;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      ov[j] = f32_to_bf16(fmaf(dyw[it][j], r, -xs[it][j] * k));
+      atomicAdd(dw + i + j, dwv[it][j]);
+    }
+    *(ushort8*)(dxr + i) = ov;
+  }
+}
+
+#define DISPATCH_ITERS(H, FN)                                                  \
+  do {                                                                         \
+    int iters = (H) / 2048;                                                    \
+    if (iters == 1) FN(1);                                                     \
+    else if (iters == 2) FN(2);                                                \
+    else if (iters == 3) FN(3);                                                \
+    else if (iters == 4) FN(4);                                                \
+  } while (0)
+
+extern "C" void rmsnorm_fwd_launch(const void* x, const void* w, void* y,
+                                   void* invrms, long rows, int H, float eps,
+                                   hipStream_t stream) {
+#define LAUNCH_F(I)                                                            \
+  hipLaunchKernelGGL((rmsnorm_fwd_kernel<I>), dim3((int)rows), dim3(256), 0,   \
+                     stream, (const unsigned short*)x,                         \
+                     (const unsigned short*)w, (unsigned short*)y,             \
+                     (float*)invrms, H, eps)
+  DISPATCH_ITERS(H, LAUNCH_F);
+#undef LAUNCH_F
+}
+
+extern "C" void rmsnorm_bwd_launch(const void* dy, const void* x,
+                                   const void* w, const void* invrms, void* dx,
+                                   void* dw, long rows, int H,
+                                   hipStream_t stream) {
+#define LAUNCH_B(I)                                                            \
+  hipLaunchKernelGGL((rmsnorm_bwd_kernel<I>), dim3((int)rows), dim3(256), 0,   \
+                     stream, (const unsigned short*)dy,                        \
+                     (const unsigned short*)x, (const unsigned short*)w,       \
+                     (const float*)invrms, (unsigned short*)dx, (float*)dw, H)
+  DISPATCH_ITERS(H, LAUNCH_B);
+#undef LAUNCH_B
+}

@@ -1,0 +1,85 @@
+"""Flat-parameter machinery: the MI355X-first optimizer/DDP substrate.
+
+All model parameters are re-pointed into two flat bf16 buffers (weight-decay
+group and no-decay group); gradients accumulate into matching flat bf16
+buffers.  This gives:
+  * the fused-AdamW kernel ONE memory-bound sweep per group (no per-tensor
+    launches) — torchx_amd/ops/csrc/adamw.hip;
+  * DDP gradient all-reduce over contiguous bucket slices of the flat grad
+    buffer, overlapped with backward (torchx_amd.parallel.ddp) and sized for
+    RCCL over xGMI (7 p2p links/GPU -> fewer, larger buckets than the
+    NVSwitch-tuned defaults).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Dict, Iterator, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+
+def _round_up(n: int, align: int) -> int:
+    return (n + align - 1) // align * align
+
+
+@dataclass
+class _ParamSlot:
+    param: nn.Parameter
+    group: str          # "decay" | "no_decay"
+    offset: int         # element offset into the group's flat buffer
+    numel: int
+
+
+class FlatParams:
+    """Flattens a module's parameters in place.
+
+    After construction every parameter's ``.data`` is a view into
+    ``flat_p16[group]`` and ``.grad`` is a view into ``flat_grad[group]``
+    (autograd accumulates into existing .grad views).
+    """
+
+    ALIGN = 64  # element alignment of each param slice (vector kernels)
+
+    def __init__(self, module: nn.Module, device: torch.device):
+        self.module = module
+        self.device = device
+        self.slots: List[_ParamSlot] = []
+        self.flat_p16: Dict[str, torch.Tensor] = {}
+        self.flat_grad: Dict[str, torch.Tensor] = {}
+
+        sizes = {"decay": 0, "no_decay": 0}
+        params: List[Tuple[nn.Parameter, str]] = []
+        seen = set()
+        for p in module.parameters():
+            if id(p) in seen:  # tied weights appear once
+                continue
+            seen.add(id(p))
+            group = "decay" if p.dim() >= 2 else "no_decay"
+            params.append((p, group))
+            sizes[group] += _round_up(p.numel(), self.ALIGN)
+
+        for g, n in sizes.items():
+            n = max(n, self.ALIGN)
+            self.flat_p16[g] = torch.zeros(n, dtype=torch.bfloat16, device=device)
+            self.flat_grad[g] = torch.zeros(n, dtype=torch.bfloat16, device=device)
+
+        offsets = {"decay": 0, "no_decay": 0}
+        for p, g in params:
+            off = offsets[g]
+            n = p.numel()
+            flat = self.flat_p16[g]
+            flat[off:off + n].copy_(p.data.to(device).reshape(-1))
+            p.data = flat[off:off + n].view(p.shape)
+            p.grad = self.flat_grad[g][off:off + n].view(p.shape)
+            self.slots.append(_ParamSlot(p, g, off, n))
+            offsets[g] = off + _round_up(n, self.ALIGN)
+
+    def zero_grad(self) -> None:
+        for g in self.flat_grad.values():
+            g.zero_()
+
+    def groups(self) -> Iterator[Tuple[str, torch.Tensor, torch.Tensor]]:
+        for g in self.flat_p16:
+            yield g, self.flat_p16[g], self.flat_grad[g]
