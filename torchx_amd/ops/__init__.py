@@ -31,7 +31,8 @@ def _load():
         _load_err = f"{_SO} not built (run torchx_amd/ops/build.py)"
         return None
     try:
-        spec = importlib.util.spec_from_file_location("torchx_amd_hip_ops", _SO)
+        # name must match TORCH_EXTENSION_NAME (PyInit__hip_ops)
+        spec = importlib.util.spec_from_file_location("_hip_ops", _SO)
         mod = importlib.util.module_from_spec(spec)
         spec.loader.exec_module(mod)  # type: ignore[union-attr]
         _hip = mod
