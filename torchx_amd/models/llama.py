@@ -117,12 +117,12 @@ class LlamaModel(nn.Module):
                                  dtype=torch.bfloat16)
         if cfg.tie_embeddings:
             self.lm_head.weight = self.embed.weight
-        if device is not None:
-            self.to(device)
         cos, sin = ops.rope_tables(cfg.max_seq_len, cfg.head_dim,
-                                   cfg.rope_theta)
+                                   cfg.rope_theta, device=device)
         self.register_buffer("rope_cos", cos, persistent=False)
         self.register_buffer("rope_sin", sin, persistent=False)
+        if device is not None:
+            self.to(device)
         self.apply(self._init_weights)
 
     def _init_weights(self, m: nn.Module) -> None:
