@@ -438,25 +438,28 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
 // ---------------------------------------------------------------------------
 // Backward dK/dV: grid over kv blocks of 128 rows (wave owns 32); inner loop
 // over GQA group heads x q tiles.  No atomics: each (b, hkv, kv-row) is
-// owned by exactly one wave.
+// owned by exactly one wave.  Split into a dV pass and a dK pass so each
+// kernel fits 2 waves/SIMD (the combined kernel needed 316 regs -> 1
+// wave/SIMD with zero latency hiding; the S^T recompute in the second pass
+// costs 8 extra MFMAs/tile but doubles occupancy).
 // ---------------------------------------------------------------------------
-__global__ void __launch_bounds__(256, 1)
+template <bool DK>
+__global__ void __launch_bounds__(256, 2)
 attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     const unsigned short* __restrict__ k,
                     const unsigned short* __restrict__ v,
                     const unsigned short* __restrict__ dout,
                     const float* __restrict__ lse,
                     const float* __restrict__ delta,
-                    unsigned short* __restrict__ dk,
-                    unsigned short* __restrict__ dv,
+                    unsigned short* __restrict__ out,  // dk or dv
                     int B, int S, int Hq, int Hkv, float scale, int causal) {
-  // staging area: Q nat + dO nat + Q^T + dO^T + lse/delta rows
-  __shared__ __align__(16) unsigned short smem[2 * QBLK * NAT_STRIDE + 2 * HD * TR_STRIDE];
+  // dK pass stages {Q nat, dO nat, Q^T}; dV pass stages {Q nat, dO^T}
+  __shared__ __align__(16) unsigned short smem[2 * QBLK * NAT_STRIDE +
+                                               HD * TR_STRIDE];
   __shared__ float lse_s[QBLK], del_s[QBLK];
   unsigned short* q_nat = smem;
-  unsigned short* do_nat = smem + QBLK * NAT_STRIDE;
-  unsigned short* q_tr = do_nat + QBLK * NAT_STRIDE;
-  unsigned short* do_tr = q_tr + HD * TR_STRIDE;
+  unsigned short* do_nat = smem + QBLK * NAT_STRIDE;         // dK only
+  unsigned short* tr_tile = smem + 2 * QBLK * NAT_STRIDE;    // q_tr / do_tr
 
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
   int bid = blockIdx.x;
@@ -482,7 +485,7 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   const bool wave_active = kw0 < S;
   const bool krow_valid = k_row < S;
 
-  // K, V fragments resident in registers (lane = k row)
+  // K (and V for the dK pass) fragments resident in registers (lane = k row)
   mbf16x8 kfrag[8], vfrag[8];
   {
     const long row = krow_valid ? k_row : (S - 1);
@@ -491,12 +494,14 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
     #pragma unroll
     for (int c = 0; c < 8; ++c) {
       kfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(kp + c * 16));
-      vfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(vp + c * 16));
+      if (DK) {
+        vfrag[c] =
+            __builtin_bit_cast(mbf16x8, *(const ushort8*)(vp + c * 16));
+      }
     }
   }
 
-  f32x16 acc_dk[4] = {};
-  f32x16 acc_dv[4] = {};
+  f32x16 acc[4] = {};
 
   const int t0 = causal ? (kv0_blk / QBLK) : 0;
   const int nt = (S + QBLK - 1) / QBLK;
@@ -513,32 +518,36 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
       const int q0 = t * QBLK;
       __syncthreads();
       stage_nat(qb, q0, q_seq_stride, QBLK, S, q_nat);
-      stage_nat(dob, q0, q_seq_stride, QBLK, S, do_nat);
-      stage_tr(qb, q0, q_seq_stride, QBLK, S, q_tr, 1.0f);
-      stage_tr(dob, q0, q_seq_stride, QBLK, S, do_tr, 1.0f);
+      if (DK) {
+        stage_nat(dob, q0, q_seq_stride, QBLK, S, do_nat);
+        stage_tr(qb, q0, q_seq_stride, QBLK, S, tr_tile, 1.0f);
+      } else {
+        stage_tr(dob, q0, q_seq_stride, QBLK, S, tr_tile, 1.0f);
+      }
       if (threadIdx.x < QBLK) {
         const int qg = q0 + threadIdx.x;
         lse_s[threadIdx.x] = (qg < S) ? lse_b[qg] : 0.f;
-        del_s[threadIdx.x] = (qg < S) ? del_b[qg] : 0.f;
+        if (DK) del_s[threadIdx.x] = (qg < S) ? del_b[qg] : 0.f;
       }
       __syncthreads();
 
-      // wave needs this q tile iff its last q row >= wave's first kv row
       const bool needed =
           wave_active && (!causal || q0 + QBLK - 1 >= kw0);
       if (!needed) continue;
 
-      // S[q, k] (lane = k col), dP[q, k]
+      // S[q, k] (lane = k col); dP[q, k] for the dK pass
       f32x16 acc_s = {}, acc_dp = {};
       #pragma unroll
       for (int c = 0; c < 8; ++c) {
         acc_s = mfma32(lds_frag(q_nat, NAT_STRIDE, c * 16), kfrag[c], acc_s);
-        acc_dp =
-            mfma32(lds_frag(do_nat, NAT_STRIDE, c * 16), vfrag[c], acc_dp);
+        if (DK) {
+          acc_dp =
+              mfma32(lds_frag(do_nat, NAT_STRIDE, c * 16), vfrag[c], acc_dp);
+        }
       }
 
       const bool mask_tile = causal || (q0 + QBLK > S) || !krow_valid;
-      float pv[16], dsv[16];
+      float cv[16];
       #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qg = q0 + c_row(r, hi);
@@ -547,29 +556,21 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
           if (qg >= S || !krow_valid || (causal && k_row > qg)) s = -3.0e38f;
         }
         const float pr = __expf(s - lse_s[qg - q0]);
-        pv[r] = pr;
-        dsv[r] = scale * pr * (acc_dp[r] - del_s[qg - q0]);
+        cv[r] = DK ? scale * pr * (acc_dp[r] - del_s[qg - q0]) : pr;
       }
 
-      mbf16x8 pf0, pf1, df0, df1;
-      cvals_to_frags(pv, hi, &pf0, &pf1);
-      cvals_to_frags(dsv, hi, &df0, &df1);
+      mbf16x8 f0, f1;
+      cvals_to_frags(cv, hi, &f0, &f1);
 
-      // dV[k, d] += P^T . dO ; dK[k, d] += dS^T . Q
+      // dV[k,d] += P^T . dO   |   dK[k,d] += dS^T . Q
       #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
-        acc_dv[dt] =
-            mfma32(pf0, lds_frag(do_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
-                   acc_dv[dt]);
-        acc_dv[dt] =
-            mfma32(pf1, lds_frag(do_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
-                   acc_dv[dt]);
-        acc_dk[dt] =
-            mfma32(df0, lds_frag(q_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
-                   acc_dk[dt]);
-        acc_dk[dt] =
-            mfma32(df1, lds_frag(q_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
-                   acc_dk[dt]);
+        acc[dt] =
+            mfma32(f0, lds_frag(tr_tile + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
+                   acc[dt]);
+        acc[dt] =
+            mfma32(f1, lds_frag(tr_tile + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
+                   acc[dt]);
       }
     }
   }
@@ -580,39 +581,31 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   float* tr = (float*)smem + wave * (32 * 36);
   if (wave_active) {
     #pragma unroll
-    for (int which = 0; which < 2; ++which) {
-      f32x16* acc = which == 0 ? acc_dv : acc_dk;
-      unsigned short* out = which == 0 ? dv : dk;
+    for (int dt = 0; dt < 4; ++dt) {
+      __syncthreads();
       #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        __syncthreads();
-        // write C-layout regs: row k = c_row(r, hi), col d in [0,32)
-        #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          tr[c_row(r, hi) * 36 + kr] = acc[dt][r];
-        }
-        __syncthreads();
-        // read back row-per-lane: each half covers 16 of the 32 d columns
-        #pragma unroll
-        for (int rep = 0; rep < 4; ++rep) {
-          const int kk = kr;
-          const int d0 = (hi ? 16 : 0) + rep * 4;
-          if (kw0 + kk < S) {
-            bf16x4_raw w;
-            #pragma unroll
-            for (int j = 0; j < 4; ++j) {
-              w[j] = (short)f32_to_bf16(tr[kk * 36 + d0 + j]);
-            }
-            *(bf16x4_raw*)(out + ((long)b * S + kw0 + kk) * kv_seq_stride +
-                           (long)hkv * HD + dt * 32 + d0) = w;
+      for (int r = 0; r < 16; ++r) {
+        tr[c_row(r, hi) * 36 + kr] = acc[dt][r];
+      }
+      __syncthreads();
+      #pragma unroll
+      for (int rep = 0; rep < 4; ++rep) {
+        const int kk = kr;
+        const int d0 = (hi ? 16 : 0) + rep * 4;
+        if (kw0 + kk < S) {
+          bf16x4_raw w;
+          #pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            w[j] = (short)f32_to_bf16(tr[kk * 36 + d0 + j]);
           }
+          *(bf16x4_raw*)(out + ((long)b * S + kw0 + kk) * kv_seq_stride +
+                         (long)hkv * HD + dt * 32 + d0) = w;
         }
       }
     }
   } else {
-    // keep barrier counts matched across waves
     #pragma unroll
-    for (int i = 0; i < 16; ++i) __syncthreads();
+    for (int i = 0; i < 8; ++i) __syncthreads();
   }
 }
 
@@ -649,10 +642,16 @@ extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
                      (const float*)delta, (unsigned short*)dq, B, S, Hq, Hkv,
                      scale, causal);
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(B * Hkv * nkt), dim3(256), 0,
-                     stream, (const unsigned short*)q,
+  hipLaunchKernelGGL((attn_bwd_dkv_kernel<false>), dim3(B * Hkv * nkt),
+                     dim3(256), 0, stream, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v,
                      (const unsigned short*)dout, (const float*)lse,
-                     (const float*)delta, (unsigned short*)dk,
-                     (unsigned short*)dv, B, S, Hq, Hkv, scale, causal);
+                     (const float*)delta, (unsigned short*)dv, B, S, Hq, Hkv,
+                     scale, causal);
+  hipLaunchKernelGGL((attn_bwd_dkv_kernel<true>), dim3(B * Hkv * nkt),
+                     dim3(256), 0, stream, (const unsigned short*)q,
+                     (const unsigned short*)k, (const unsigned short*)v,
+                     (const unsigned short*)dout, (const float*)lse,
+                     (const float*)delta, (unsigned short*)dk, B, S, Hq, Hkv,
+                     scale, causal);
 }
