@@ -1,0 +1,290 @@
+"""torchx_amd elastic agent — the native torchrun replacement.
+
+Launched by the dist.ddp component (one agent per node), it:
+  * joins a c10d TCPStore rendezvous (rendezvous.py),
+  * spawns one worker process per local GPU (or --nproc-per-node),
+  * wires the torch.distributed env contract (RANK/LOCAL_RANK/WORLD_SIZE/
+    MASTER_ADDR/MASTER_PORT/...) so workers bring up RCCL over xGMI,
+  * supervises workers: on failure, signals every agent through the store,
+    tears down, re-rendezvouses and restarts (up to --max-restarts),
+  * propagates the first failure's torchelastic error file to the
+    scheduler's TORCHELASTIC_ERROR_FILE reply file.
+
+Usage:
+  python -m torchx_amd.agent --nnodes 1:1 --nproc-per-node 8 \
+      --rdzv-endpoint localhost:29500 --rdzv-id app_1 --max-restarts 0 \
+      [--tee] [--no-python] script.py [script args...]
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import logging
+import os
+import shutil
+import signal
+import subprocess
+import sys
+import threading
+import time
+import traceback
+from typing import Dict, List, Optional
+
+from .rendezvous import C10dRendezvous, RendezvousResult
+
+log = logging.getLogger("torchx_amd.agent")
+
+POLL_INTERVAL = 1.0
+
+
+def parse_nnodes(spec: str):
+    if ":" in spec:
+        lo, hi = spec.split(":")
+        return int(lo), int(hi)
+    n = int(spec)
+    return n, n
+
+
+class Worker:
+    def __init__(self, proc: subprocess.Popen, local_rank: int,
+                 global_rank: int, error_file: str,
+                 pumps: List[threading.Thread]) -> None:
+        self.proc = proc
+        self.local_rank = local_rank
+        self.global_rank = global_rank
+        self.error_file = error_file
+        self.pumps = pumps
+
+
+def _pump(src, dst, prefix: str) -> threading.Thread:
+    def run() -> None:
+        for line in iter(src.readline, b""):
+            try:
+                dst.buffer.write(prefix.encode() + line)
+                dst.flush()
+            except ValueError:
+                break
+        src.close()
+
+    t = threading.Thread(target=run, daemon=True)
+    t.start()
+    return t
+
+
+def _default_nproc() -> int:
+    from torchx_amd.schedulers.devices import hip_device_count
+
+    n = hip_device_count()
+    return n if n > 0 else 1
+
+
+def spawn_workers(args, rdzv: RendezvousResult, restart_count: int,
+                  log_dir: str) -> List[Worker]:
+    nproc = args.nproc_per_node
+    world_size = rdzv.num_nodes * nproc
+    workers: List[Worker] = []
+    cmd_base: List[str]
+    if args.no_python:
+        cmd_base = [args.script]
+    else:
+        cmd_base = [sys.executable, args.script]
+    for lr in range(nproc):
+        grank = rdzv.node_rank * nproc + lr
+        error_file = os.path.join(log_dir, f"worker_{grank}_error.json")
+        if os.path.exists(error_file):
+            os.unlink(error_file)
+        env = {
+            **os.environ,
+            "RANK": str(grank),
+            "LOCAL_RANK": str(lr),
+            "WORLD_SIZE": str(world_size),
+            "LOCAL_WORLD_SIZE": str(nproc),
+            "GROUP_RANK": str(rdzv.node_rank),
+            "ROLE_RANK": str(grank),
+            "ROLE_WORLD_SIZE": str(world_size),
+            "MASTER_ADDR": rdzv.master_addr,
+            "MASTER_PORT": str(rdzv.master_port),
+            "TORCHELASTIC_RESTART_COUNT": str(restart_count),
+            "TORCHELASTIC_MAX_RESTARTS": str(args.max_restarts),
+            "TORCHELASTIC_RUN_ID": args.rdzv_id,
+            "TORCHELASTIC_ERROR_FILE": error_file,
+            "OMP_NUM_THREADS": os.environ.get("OMP_NUM_THREADS", "1"),
+        }
+        stdout = subprocess.PIPE if args.tee else None
+        stderr = subprocess.PIPE if args.tee else None
+        proc = subprocess.Popen(
+            cmd_base + args.script_args,
+            env=env,
+            stdout=stdout,
+            stderr=stderr,
+            start_new_session=True,
+        )
+        pumps = []
+        if args.tee:
+            pumps.append(_pump(proc.stdout, sys.stdout, f"[rank{grank}]: "))
+            pumps.append(_pump(proc.stderr, sys.stderr, f"[rank{grank}]: "))
+        workers.append(Worker(proc, lr, grank, error_file, pumps))
+    return workers
+
+
+def kill_workers(workers: List[Worker]) -> None:
+    for w in workers:
+        if w.proc.poll() is None:
+            try:
+                os.killpg(w.proc.pid, signal.SIGTERM)
+            except ProcessLookupError:
+                pass
+    deadline = time.time() + 10
+    for w in workers:
+        while w.proc.poll() is None and time.time() < deadline:
+            time.sleep(0.1)
+        if w.proc.poll() is None:
+            try:
+                os.killpg(w.proc.pid, signal.SIGKILL)
+            except ProcessLookupError:
+                pass
+            w.proc.wait()
+
+
+def propagate_error(workers: List[Worker]) -> Optional[dict]:
+    """Earliest worker error file wins; copied to the scheduler reply file."""
+    best = None
+    best_mtime = float("inf")
+    for w in workers:
+        if os.path.isfile(w.error_file):
+            mt = os.path.getmtime(w.error_file)
+            if mt < best_mtime:
+                best_mtime = mt
+                best = w.error_file
+    reply = os.environ.get("TORCHELASTIC_ERROR_FILE")
+    if best and reply and best != reply:
+        os.makedirs(os.path.dirname(reply) or ".", exist_ok=True)
+        shutil.copyfile(best, reply)
+    if best:
+        try:
+            with open(best) as f:
+                return json.load(f)
+        except (OSError, json.JSONDecodeError):
+            return None
+    return None
+
+
+def write_agent_error(msg: str) -> None:
+    reply = os.environ.get("TORCHELASTIC_ERROR_FILE")
+    if not reply:
+        return
+    try:
+        os.makedirs(os.path.dirname(reply) or ".", exist_ok=True)
+        with open(reply, "w") as f:
+            json.dump(
+                {"message": {"message": msg,
+                             "extraInfo": {"timestamp": str(int(time.time()))}}},
+                f,
+            )
+    except OSError:
+        pass
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    p = argparse.ArgumentParser(prog="torchx_amd.agent")
+    p.add_argument("--nnodes", type=str, default="1:1")
+    p.add_argument("--nproc-per-node", type=str, default="auto")
+    p.add_argument("--rdzv-endpoint", type=str, default="localhost:29500")
+    p.add_argument("--rdzv-id", type=str, default="default")
+    p.add_argument("--max-restarts", type=int, default=0)
+    p.add_argument("--tee", action="store_true", default=True)
+    p.add_argument("--no-tee", dest="tee", action="store_false")
+    p.add_argument("--no-python", action="store_true")
+    p.add_argument("--log-dir", type=str, default=None)
+    p.add_argument("script", type=str)
+    p.add_argument("script_args", nargs=argparse.REMAINDER)
+    args = p.parse_args(argv)
+
+    logging.basicConfig(
+        level=os.environ.get("LOGLEVEL", "INFO"),
+        format="[torchx_amd.agent] %(levelname)s: %(message)s",
+    )
+
+    if args.nproc_per_node in ("auto", "gpu"):
+        args.nproc_per_node = _default_nproc()
+    else:
+        args.nproc_per_node = int(args.nproc_per_node)
+
+    min_nodes, max_nodes = parse_nnodes(args.nnodes)
+    log_dir = args.log_dir or os.environ.get("PET_LOG_DIR") or "/tmp"
+    os.makedirs(log_dir, exist_ok=True)
+
+    rdzv = C10dRendezvous(args.rdzv_endpoint, args.rdzv_id, min_nodes,
+                          max_nodes)
+
+    restart_count = 0
+    workers: List[Worker] = []
+
+    def handle_term(signum, frame):
+        kill_workers(workers)
+        sys.exit(128 + signum)
+
+    signal.signal(signal.SIGTERM, handle_term)
+    signal.signal(signal.SIGINT, handle_term)
+
+    try:
+        round_ = rdzv.restart_round()
+        while True:
+            log.info(
+                "rendezvous round %d (restart %d/%d), nproc_per_node=%d",
+                round_, restart_count, args.max_restarts, args.nproc_per_node,
+            )
+            result = rdzv.join(round_)
+            log.info(
+                "joined as node %d/%d; master %s:%d",
+                result.node_rank, result.num_nodes, result.master_addr,
+                result.master_port,
+            )
+            workers = spawn_workers(args, result, restart_count, log_dir)
+
+            failed: Optional[Worker] = None
+            while True:
+                time.sleep(POLL_INTERVAL)
+                states = [w.proc.poll() for w in workers]
+                if any(rc is not None and rc != 0 for rc in states):
+                    failed = next(
+                        w for w, rc in zip(workers, states)
+                        if rc is not None and rc != 0
+                    )
+                    break
+                if all(rc == 0 for rc in states):
+                    log.info("all %d local workers succeeded", len(workers))
+                    return 0
+                # another node may have failed: check the restart counter
+                cur = rdzv.restart_round()
+                if cur > round_:
+                    log.warning("restart signalled by another node (round %d)",
+                                cur)
+                    round_ = cur
+                    failed = None
+                    break
+
+            kill_workers(workers)
+            if failed is not None:
+                log.error(
+                    "worker rank %d exited rc=%d", failed.global_rank,
+                    failed.proc.returncode,
+                )
+                round_ = rdzv.signal_restart(round_)
+            restart_count += 1
+            if restart_count > args.max_restarts:
+                propagate_error(workers)
+                log.error("exceeded max restarts (%d); failing",
+                          args.max_restarts)
+                return 1
+    except Exception as e:  # noqa: BLE001
+        kill_workers(workers)
+        write_agent_error(f"agent error: {e}\n{traceback.format_exc()}")
+        raise
+    finally:
+        kill_workers(workers)
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,127 @@
+"""HIP/MI355X device enumeration and per-replica pinning.
+
+The MI355X analog of the reference's `nvidia-smi -L` + CUDA_VISIBLE_DEVICES
+partitioning (torchx/schedulers/local_scheduler.py:837-950): enumerate via
+`amd-smi`/`rocm-smi` (torch.cuda as a fallback probe), partition contiguous
+device-index ranges per replica, and export BOTH `HIP_VISIBLE_DEVICES` and
+`ROCR_VISIBLE_DEVICES`.  Also maps named devices (e.g. RDMA NICs) to
+DeviceMounts for containerized schedulers.
+"""
+
+from __future__ import annotations
+
+import functools
+import json
+import logging
+import shutil
+import subprocess
+from typing import Dict, List, Optional
+
+from torchx_amd.specs import DeviceMount, Resource
+
+log = logging.getLogger(__name__)
+
+HIP_VISIBLE_DEVICES = "HIP_VISIBLE_DEVICES"
+ROCR_VISIBLE_DEVICES = "ROCR_VISIBLE_DEVICES"
+
+# named-device -> container device mounts (KFD + DRI are what ROCm containers
+# need; see the docker scheduler)
+DEVICES: Dict[str, List[DeviceMount]] = {
+    "amd.com/gpu": [
+        DeviceMount(src_path="/dev/kfd", dst_path="/dev/kfd"),
+        DeviceMount(src_path="/dev/dri", dst_path="/dev/dri"),
+    ],
+}
+
+
+def get_device_mounts(devices: Dict[str, int]) -> List[DeviceMount]:
+    mounts: List[DeviceMount] = []
+    for name in devices:
+        mounts.extend(DEVICES.get(name, []))
+    return mounts
+
+
+@functools.lru_cache(maxsize=1)
+def hip_device_count() -> int:
+    """Number of visible AMD GPUs on this host (0 if none)."""
+    # 1) amd-smi (ROCm >= 5.7)
+    if shutil.which("amd-smi"):
+        try:
+            out = subprocess.run(
+                ["amd-smi", "list", "--json"], capture_output=True, timeout=30
+            )
+            if out.returncode == 0:
+                data = json.loads(out.stdout.decode() or "[]")
+                if isinstance(data, list):
+                    return len(data)
+                if isinstance(data, dict):  # some versions nest under a key
+                    for v in data.values():
+                        if isinstance(v, list):
+                            return len(v)
+        except Exception as e:  # noqa: BLE001
+            log.debug("amd-smi probe failed: %s", e)
+    # 2) rocm-smi
+    if shutil.which("rocm-smi"):
+        try:
+            out = subprocess.run(
+                ["rocm-smi", "--showid", "--json"], capture_output=True,
+                timeout=30,
+            )
+            if out.returncode == 0:
+                data = json.loads(out.stdout.decode() or "{}")
+                n = len([k for k in data if k.startswith("card")])
+                if n:
+                    return n
+        except Exception as e:  # noqa: BLE001
+            log.debug("rocm-smi probe failed: %s", e)
+    # 3) torch runtime probe
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            return torch.cuda.device_count()
+    except Exception:  # noqa: BLE001
+        pass
+    return 0
+
+
+def partition_devices(
+    role_replicas: Dict[str, int],
+    gpus_per_replica: Dict[str, int],
+    total_gpus: Optional[int] = None,
+) -> Dict[str, List[Optional[str]]]:
+    """Assign contiguous device-index ranges per replica, role order.
+
+    Returns role -> list (len num_replicas) of comma-joined device strings
+    (None = don't set the env var, e.g. when over-subscribed).
+    """
+    if total_gpus is None:
+        total_gpus = hip_device_count()
+    assignments: Dict[str, List[Optional[str]]] = {}
+    next_dev = 0
+    requested = sum(
+        role_replicas[r] * gpus_per_replica.get(r, 0) for r in role_replicas
+    )
+    oversubscribed = requested > total_gpus
+    for role, n in role_replicas.items():
+        per = gpus_per_replica.get(role, 0)
+        out: List[Optional[str]] = []
+        for _ in range(n):
+            if per <= 0 or oversubscribed:
+                out.append(None)
+            else:
+                out.append(",".join(str(d) for d in range(next_dev, next_dev + per)))
+                next_dev += per
+        assignments[role] = out
+    if oversubscribed and requested > 0:
+        log.warning(
+            "requested %d GPUs but host has %d; not setting %s",
+            requested, total_gpus, HIP_VISIBLE_DEVICES,
+        )
+    return assignments
+
+
+def device_env(devices: Optional[str]) -> Dict[str, str]:
+    if devices is None:
+        return {}
+    return {HIP_VISIBLE_DEVICES: devices, ROCR_VISIBLE_DEVICES: devices}

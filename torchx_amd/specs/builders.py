@@ -124,6 +124,10 @@ def materialize_appdef(
     defaults (parity: torchx/specs/builders.py:137-179).
     """
     sig = inspect.signature(fn)
+    try:
+        hints = typing.get_type_hints(fn)
+    except Exception:  # noqa: BLE001 — unresolvable forward refs
+        hints = {}
     parser = create_args_parser(fn)
     if defaults:
         known = {
@@ -178,8 +182,9 @@ def materialize_appdef(
             seen_varargs = True
             continue
         raw = getattr(ns, name)
-        if isinstance(raw, str) and p.annotation is not str:
-            raw = _decode_string(raw, p.annotation)
+        ann = hints.get(name, p.annotation)
+        if isinstance(raw, str) and ann is not str:
+            raw = _decode_string(raw, ann)
         if seen_varargs or p.kind == inspect.Parameter.KEYWORD_ONLY or not has_varargs:
             kw_vals[name] = raw
         else:
