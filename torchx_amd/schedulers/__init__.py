@@ -35,9 +35,9 @@ def get_scheduler_factories() -> Dict[str, SchedulerFactory]:
     try:
         from torchx_amd.plugins import registry
 
-        plugin_factories = registry().scheduler_factories()
-        if plugin_factories:
-            factories = dict(plugin_factories)
+        # merge over the defaults (plugins add or override by name; they
+        # never silently remove the builtins)
+        factories.update(registry().scheduler_factories())
     except Exception:  # noqa: BLE001 — plugins must never break core
         pass
     return factories
