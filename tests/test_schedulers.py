@@ -1,0 +1,254 @@
+"""Request-generation tests for docker/slurm/kubernetes schedulers against
+mocked data planes (reference test strategy: SURVEY.md §4 — assert the exact
+generated request; no cluster needed)."""
+
+import json
+from unittest.mock import MagicMock, patch
+
+import pytest
+
+from torchx_amd.specs import (
+    AppDef,
+    AppState,
+    BindMount,
+    DeviceMount,
+    Resource,
+    RetryPolicy,
+    Role,
+    macros,
+    resource,
+)
+
+
+def _app(gpu=8, replicas=2, max_retries=0, min_replicas=None):
+    return AppDef(
+        name="trainer",
+        roles=[
+            Role(
+                name="worker",
+                image="rocm/pytorch:latest",
+                entrypoint="python3",
+                args=["-m", "train", "--id", macros.replica_id],
+                env={"FOO": "bar"},
+                num_replicas=replicas,
+                min_replicas=min_replicas,
+                max_retries=max_retries,
+                retry_policy=RetryPolicy.APPLICATION,
+                resource=resource(h="mi355x.8gpu") if gpu == 8
+                else Resource(cpu=4, gpu=gpu, memMB=8192),
+            )
+        ],
+    )
+
+
+# ---------------------------------------------------------------------------
+# docker
+# ---------------------------------------------------------------------------
+
+
+def _docker_sched():
+    from torchx_amd.schedulers.docker_scheduler import DockerScheduler
+
+    return DockerScheduler("test", client=MagicMock())
+
+
+def test_docker_dryrun_request():
+    sched = _docker_sched()
+    info = sched.submit_dryrun(_app(), {})
+    req = info.request
+    assert len(req.containers) == 2
+    c0 = req.containers[0]
+    assert c0.image == "rocm/pytorch:latest"
+    # macro substitution happened per replica
+    assert c0.command == ["python3", "-m", "train", "--id", "0"]
+    assert req.containers[1].command[-1] == "1"
+    # ROCm GPU wiring: kfd + dri devices, no CUDA device requests
+    assert "/dev/kfd:/dev/kfd:rwm" in c0.kwargs["devices"]
+    assert "/dev/dri:/dev/dri:rwm" in c0.kwargs["devices"]
+    assert "video" in c0.kwargs["group_add"]
+    # rank0 DNS name on the bridge network
+    env = c0.kwargs["environment"]
+    assert env["TORCHX_RANK0_HOST"] == c0.kwargs["name"]
+    assert req.containers[1].kwargs["environment"]["TORCHX_RANK0_HOST"] == \
+        c0.kwargs["name"]
+    assert env["FOO"] == "bar"
+    # resources
+    assert c0.kwargs["nano_cpus"] == int(192 * 1e9)
+    assert c0.kwargs["network"] == "torchx"
+
+
+def test_docker_restart_policy():
+    sched = _docker_sched()
+    info = sched.submit_dryrun(_app(max_retries=3), {})
+    c0 = info.request.containers[0]
+    assert c0.kwargs["restart_policy"] == {
+        "Name": "on-failure", "MaximumRetryCount": 3
+    }
+
+
+def test_docker_mounts():
+    from torchx_amd.schedulers.docker_scheduler import DockerScheduler
+
+    app = _app(gpu=0, replicas=1)
+    app.roles[0].mounts = [
+        BindMount("/host", "/cont", read_only=True),
+        DeviceMount("/dev/xyz", "/dev/xyz"),
+    ]
+    sched = _docker_sched()
+    info = sched.submit_dryrun(app, {})
+    kw = info.request.containers[0].kwargs
+    assert kw["volumes"]["/host"] == {"bind": "/cont", "mode": "ro"}
+    assert "/dev/xyz:/dev/xyz:rwm" in kw["devices"]
+
+
+def test_docker_describe_states():
+    sched = _docker_sched()
+    c = MagicMock()
+    c.labels = {"torchx.ai/app-id": "app1", "torchx.ai/role": "worker",
+                "torchx.ai/replica": "0"}
+    c.status = "exited"
+    c.attrs = {"State": {"ExitCode": 1}}
+    c.name = "app1-worker-0"
+    sched._DockerScheduler__client.containers.list.return_value = [c]
+    desc = sched.describe("app1")
+    assert desc.state == AppState.FAILED
+
+
+# ---------------------------------------------------------------------------
+# slurm
+# ---------------------------------------------------------------------------
+
+
+def test_slurm_sbatch_script():
+    from torchx_amd.schedulers.slurm_scheduler import SlurmScheduler
+
+    sched = SlurmScheduler("test")
+    with patch("torchx_amd.schedulers.slurm_scheduler.version",
+               return_value=(24, 11)):
+        info = sched.submit_dryrun(
+            _app(replicas=2, max_retries=2), {"partition": "gpu"}
+        )
+    script = info.request.materialize()
+    assert "#SBATCH hetjob" in script          # het group separator
+    assert "#SBATCH --gpus-per-node=8" in script
+    assert "#SBATCH --partition=gpu" in script
+    assert "#SBATCH --requeue" in script
+    assert "scontrol requeue $SLURM_JOB_ID" in script
+    assert "TORCHX_MAX_RETRIES=2" in script
+    assert " : " in script                     # heterogeneous srun groups
+    assert "--kill-on-bad-exit=1" in script
+    # macros: app_id becomes the shell var
+    assert "$SLURM_JOB_ID" in script or "SLURM_JOB_ID" in script
+
+
+def test_slurm_old_version_gpu_flag():
+    from torchx_amd.schedulers.slurm_scheduler import SlurmScheduler
+
+    sched = SlurmScheduler("test")
+    with patch("torchx_amd.schedulers.slurm_scheduler.version",
+               return_value=(22, 5)):
+        info = sched.submit_dryrun(_app(replicas=1), {})
+    assert "--gpus-per-task=8" in info.request.materialize()
+
+
+def test_slurm_schedule_and_describe(tmp_path):
+    from torchx_amd.schedulers.slurm_scheduler import SlurmScheduler
+
+    sched = SlurmScheduler("test")
+    with patch("torchx_amd.schedulers.slurm_scheduler.version",
+               return_value=(24, 11)):
+        info = sched.submit_dryrun(_app(replicas=1), {"job_dir": str(tmp_path)})
+    with patch("subprocess.check_output", return_value=b"1234\n") as co:
+        app_id = sched.schedule(info)
+    assert app_id == "1234"
+    assert (tmp_path / "torchx-sbatch.sh").exists()
+
+    squeue_json = json.dumps({
+        "jobs": [
+            {"job_id": 1234, "name": "worker-0", "job_state": ["RUNNING"],
+             "nodes": "node1"},
+        ]
+    }).encode()
+    with patch("subprocess.check_output", return_value=squeue_json):
+        desc = sched.describe("1234")
+    assert desc.state == AppState.RUNNING
+    assert desc.roles_statuses[0].replicas[0].hostname == "node1"
+
+    sacct_out = (
+        "JobID|JobName|State\n"
+        "1234|worker-0|COMPLETED\n"
+        "1234.batch|batch|COMPLETED\n"
+    ).encode()
+
+    def fallback(cmd, **kw):
+        if cmd[0] == "squeue":
+            raise FileNotFoundError
+        return sacct_out
+
+    with patch("subprocess.check_output", side_effect=fallback):
+        desc = sched.describe("1234")
+    assert desc.state == AppState.SUCCEEDED
+
+
+# ---------------------------------------------------------------------------
+# kubernetes
+# ---------------------------------------------------------------------------
+
+
+def test_k8s_volcano_resource():
+    from torchx_amd.schedulers.kubernetes_scheduler import KubernetesScheduler
+
+    sched = KubernetesScheduler("test", client=MagicMock())
+    info = sched.submit_dryrun(
+        _app(replicas=2, max_retries=1), {"queue": "default"}
+    )
+    res = info.request.resource
+    assert res["apiVersion"] == "batch.volcano.sh/v1alpha1"
+    spec = res["spec"]
+    assert spec["schedulerName"] == "volcano"
+    assert spec["queue"] == "default"
+    assert spec["minAvailable"] == 2          # gang scheduling
+    tasks = spec["tasks"]
+    assert len(tasks) == 2
+    pod = tasks[0]["template"]
+    limits = pod["spec"]["containers"][0]["resources"]["limits"]
+    assert limits["amd.com/gpu"] == 8          # AMD device plugin resource
+    # /dev/shm for RCCL
+    assert {"name": "dshm", "emptyDir": {"medium": "Memory"}} in \
+        pod["spec"]["volumes"]
+    # retry policy mapping
+    assert {"event": "PodFailed", "action": "RestartJob"} in \
+        tasks[0]["policies"]
+    # per-replica macro substitution
+    cmd0 = tasks[0]["template"]["spec"]["containers"][0]["command"]
+    cmd1 = tasks[1]["template"]["spec"]["containers"][0]["command"]
+    assert cmd0[-1] == "0" and cmd1[-1] == "1"
+
+
+def test_k8s_elastic_min_available():
+    from torchx_amd.schedulers.kubernetes_scheduler import KubernetesScheduler
+
+    sched = KubernetesScheduler("test", client=MagicMock())
+    info = sched.submit_dryrun(
+        _app(replicas=4, min_replicas=2), {"queue": "q"}
+    )
+    spec = info.request.resource["spec"]
+    assert spec["minAvailable"] == 2
+    mins = [t.get("minAvailable") for t in spec["tasks"]]
+    assert mins == [1, 1, 0, 0]
+
+
+def test_k8s_requires_queue():
+    from torchx_amd.schedulers.kubernetes_scheduler import KubernetesScheduler
+    from torchx_amd.specs import InvalidRunConfigException
+
+    sched = KubernetesScheduler("test", client=MagicMock())
+    with pytest.raises(InvalidRunConfigException):
+        sched.submit_dryrun(_app(), {})
+
+
+def test_k8s_pod_name_sanitized():
+    from torchx_amd.schedulers.kubernetes_scheduler import sanitize_for_k8s
+
+    assert sanitize_for_k8s("My_App.Name") == "my-app-name"
+    assert len(sanitize_for_k8s("x" * 100)) == 63
