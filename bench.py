@@ -29,7 +29,8 @@ def main() -> int:
     p.add_argument("--micro-batch", type=int, default=4)
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--model", type=str, default="llama3_8b",
-                   choices=["llama3_8b", "llama3_8b_small", "tiny"])
+                   choices=["llama3_8b", "llama3_8b_small", "tiny",
+                            "mixtral8x7b", "mixtral_small"])
     p.add_argument("--device", type=str, default=None)
     args = p.parse_args()
 
@@ -55,12 +56,18 @@ def main() -> int:
     from torchx_amd.models.llama import (
         LlamaConfig, LlamaModel, llama3_8b, llama_tiny,
     )
+    from torchx_amd.models.mixtral import MixtralConfig, MixtralModel, mixtral_8x7b
     from torchx_amd.parallel import FlatAdamW, FlatDDP, FlatParams
 
+    is_moe = args.model.startswith("mixtral")
     if args.model == "llama3_8b":
         cfg = llama3_8b()
     elif args.model == "llama3_8b_small":
         cfg = LlamaConfig(num_layers=4)  # 8B shape, 4 layers (debug)
+    elif args.model == "mixtral8x7b":
+        cfg = mixtral_8x7b()
+    elif args.model == "mixtral_small":
+        cfg = MixtralConfig(num_layers=4)  # 8x7B shape, 4 layers (debug)
     else:
         cfg = llama_tiny()
         args.seq_len = min(args.seq_len, cfg.max_seq_len)
@@ -74,9 +81,25 @@ def main() -> int:
 
         ops.hip_ops(required=True)  # fail loudly if the extension is missing
 
-    model = LlamaModel(cfg, device=device)
-    flat = FlatParams(model, device)
-    ddp = FlatDDP(flat)
+    if is_moe:
+        # expert parallelism over the whole job when it divides the
+        # expert count (BASELINE config 5: EP all-to-all over xGMI);
+        # expert params are EP-sharded -> excluded from the DP all-reduce
+        ep_size = world if (world > 1 and cfg.num_experts % world == 0) else 1
+        model = MixtralModel(cfg, device=device, ep_group=None,
+                             ep_size=ep_size, ep_rank=rank % ep_size)
+
+        def group_fn(name, param):
+            if ep_size > 1 and "local_experts" in name:
+                return "expert"
+            return "decay" if param.dim() >= 2 else "no_decay"
+
+        flat = FlatParams(model, device, group_fn=group_fn)
+        ddp = FlatDDP(flat, local_groups={"expert"})
+    else:
+        model = LlamaModel(cfg, device=device)
+        flat = FlatParams(model, device)
+        ddp = FlatDDP(flat)
     opt = FlatAdamW(flat, lr=3e-4)
 
     B, S = args.micro_batch, args.seq_len
@@ -144,7 +167,7 @@ def main() -> int:
                 "model": args.model,
                 "global_batch": B * world,
                 "seq_len": S,
-                "parallelism": f"dp{world}",
+                "parallelism": (f"dp{world}_ep{world}" if is_moe and world > 1 else f"dp{world}"),
                 "final_loss": loss,
             },
         }

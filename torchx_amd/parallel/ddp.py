@@ -42,9 +42,15 @@ class FlatDDP:
         flat: FlatParams,
         process_group: Optional[dist.ProcessGroup] = None,
         bucket_bytes: Optional[int] = None,
+        local_groups: Optional[set] = None,
     ):
+        """``local_groups``: flat-param groups that are sharded (e.g. EP
+        experts) — never all-reduced or broadcast, but still divided by
+        world size in finish() so their grads match the dense params'
+        (global objective = mean over ranks of local losses)."""
         self.flat = flat
         self.pg = process_group
+        self.local_groups = local_groups or set()
         self.enabled = dist.is_available() and dist.is_initialized() and (
             dist.get_world_size(process_group) > 1
         )
@@ -64,6 +70,8 @@ class FlatDDP:
         elem_size = 2  # bf16
         by_group: Dict[str, List] = {}
         for slot in self.flat.slots:
+            if slot.group in self.local_groups:
+                continue
             by_group.setdefault(slot.group, []).append(slot)
         for group, slots in by_group.items():
             cur: Optional[_Bucket] = None
@@ -88,7 +96,9 @@ class FlatDDP:
             slot.param.register_post_accumulate_grad_hook(self._hook)
 
     def _broadcast_params(self) -> None:
-        for _, p16, _ in self.flat.groups():
+        for g, p16, _ in self.flat.groups():
+            if g in self.local_groups:
+                continue
             dist.broadcast(p16, src=0, group=self.pg)
 
     # -- runtime ------------------------------------------------------------

@@ -42,30 +42,41 @@ class FlatParams:
 
     ALIGN = 64  # element alignment of each param slice (vector kernels)
 
-    def __init__(self, module: nn.Module, device: torch.device):
+    def __init__(self, module: nn.Module, device: torch.device,
+                 group_fn: Optional[callable] = None):
+        """``group_fn(qualified_name, param) -> group name`` partitions the
+        params (default: "decay" for dim>=2 else "no_decay").  Extra groups
+        (e.g. "expert" for EP-sharded weights) let DDP skip their
+        all-reduce while the optimizer still sweeps them."""
         self.module = module
         self.device = device
         self.slots: List[_ParamSlot] = []
         self.flat_p16: Dict[str, torch.Tensor] = {}
         self.flat_grad: Dict[str, torch.Tensor] = {}
 
-        sizes = {"decay": 0, "no_decay": 0}
+        if group_fn is None:
+            def group_fn(name: str, p: nn.Parameter) -> str:
+                return "decay" if p.dim() >= 2 else "no_decay"
+
+        sizes: Dict[str, int] = {}
         params: List[Tuple[nn.Parameter, str]] = []
         seen = set()
-        for p in module.parameters():
+        for name, p in module.named_parameters():
             if id(p) in seen:  # tied weights appear once
                 continue
             seen.add(id(p))
-            group = "decay" if p.dim() >= 2 else "no_decay"
+            group = group_fn(name, p)
             params.append((p, group))
-            sizes[group] += _round_up(p.numel(), self.ALIGN)
+            sizes[group] = sizes.get(group, 0) + _round_up(p.numel(), self.ALIGN)
+        sizes.setdefault("decay", 0)
+        sizes.setdefault("no_decay", 0)
 
         for g, n in sizes.items():
             n = max(n, self.ALIGN)
             self.flat_p16[g] = torch.zeros(n, dtype=torch.bfloat16, device=device)
             self.flat_grad[g] = torch.zeros(n, dtype=torch.bfloat16, device=device)
 
-        offsets = {"decay": 0, "no_decay": 0}
+        offsets = {g: 0 for g in self.flat_p16}
         for p, g in params:
             off = offsets[g]
             n = p.numel()

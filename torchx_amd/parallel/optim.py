@@ -42,7 +42,7 @@ class FlatAdamW:
         self.step_count += 1
         for g, p16, grad in self.flat.groups():
             st = self.state[g]
-            wd = self.weight_decay if g == "decay" else 0.0
+            wd = 0.0 if g == "no_decay" else self.weight_decay
             ops.adamw_step(
                 st["p32"], p16, grad, st["m"], st["v"],
                 lr=self.lr, beta1=self.beta1, beta2=self.beta2, eps=self.eps,
