@@ -1,0 +1,82 @@
+"""GPU end-to-end model tests: full training steps of tiny Llama and
+tiny Mixtral through the HIP-kernel path (FlatParams/FlatDDP/FlatAdamW),
+plus a launcher e2e smoke: ``dist.ddp -j 1x1`` bringing up an RCCL
+process group on the box.
+
+Reference parity: the reference's integration tests launch
+compute_world_size via dist.ddp (torchx/components/integration_tests/
+component_provider.py:38-93); here that payload runs over RCCL.
+"""
+
+import json
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+REPO = Path(__file__).resolve().parent.parent
+
+
+def _train_steps(model, flat, ddp, opt, cfg, steps=8, B=2, S=128):
+    torch.manual_seed(7)
+    dev = next(iter(flat.flat_p16.values())).device
+    tokens = torch.randint(0, cfg.vocab_size, (B, S), device=dev)
+    targets = torch.roll(tokens, shifts=-1, dims=1)
+    losses = []
+    for _ in range(steps):
+        opt.zero_grad()
+        loss = model(tokens, targets)
+        loss.backward()
+        ddp.finish()
+        opt.step()
+        losses.append(float(loss.detach()))
+    return losses
+
+
+def test_llama_tiny_step_gpu():
+    from torchx_amd.models.llama import LlamaModel, llama_tiny
+    from torchx_amd.parallel import FlatAdamW, FlatDDP, FlatParams
+
+    cfg = llama_tiny()
+    dev = torch.device("cuda", 0)
+    model = LlamaModel(cfg, device=dev)
+    flat = FlatParams(model, dev)
+    ddp = FlatDDP(flat)
+    opt = FlatAdamW(flat, lr=1e-3)
+    losses = _train_steps(model, flat, ddp, opt, cfg)
+    assert all(l == l and l < 1e4 for l in losses)  # finite
+    # training on a fixed batch must make clear progress
+    assert losses[-1] < losses[0] - 0.5, losses
+
+
+def test_mixtral_tiny_step_gpu():
+    from torchx_amd.models.mixtral import MixtralModel, mixtral_tiny
+    from torchx_amd.parallel import FlatAdamW, FlatDDP, FlatParams
+
+    cfg = mixtral_tiny()
+    dev = torch.device("cuda", 0)
+    model = MixtralModel(cfg, device=dev)
+    flat = FlatParams(model, dev)
+    ddp = FlatDDP(flat)
+    opt = FlatAdamW(flat, lr=1e-3)
+    losses = _train_steps(model, flat, ddp, opt, cfg)
+    assert all(l == l and l < 1e4 for l in losses)
+    assert losses[-1] < losses[0] - 0.5, losses
+
+
+def test_launcher_rccl_e2e():
+    """torchx run -s local_cwd dist.ddp -j 1x1 compute_world_size over RCCL."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO) + os.pathsep + env.get("PYTHONPATH", "")
+    out = subprocess.run(
+        [sys.executable, "-m", "torchx_amd.cli.main", "run", "--wait",
+         "--scheduler", "local_cwd", "dist.ddp", "-j", "1x1",
+         "-m", "torchx_amd.apps.compute_world_size"],
+        cwd=str(REPO), env=env, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stdout + out.stderr

@@ -104,12 +104,18 @@ def create_args_parser(fn: Callable[..., Any]) -> argparse.ArgumentParser:
     )
     for name, p in sig.parameters.items():
         help_txt = helps.get(name, "")
+        # single-char params also get a short flag (-j 1x2, -m mod, -h named
+        # resource is taken by --help so it stays long-only); reference
+        # parity: torchx/specs/builders.py:113 Annotated short flags
+        flags = [f"--{name}"]
+        if len(name) == 1 and name != "h":
+            flags.append(f"-{name}")
         if p.kind == inspect.Parameter.VAR_POSITIONAL:
             parser.add_argument(name, nargs="*", default=[], help=help_txt)
         elif p.default is inspect.Parameter.empty:
-            parser.add_argument(f"--{name}", required=True, help=help_txt)
+            parser.add_argument(*flags, required=True, help=help_txt)
         else:
-            parser.add_argument(f"--{name}", default=p.default, help=help_txt)
+            parser.add_argument(*flags, default=p.default, help=help_txt)
     return parser
 
 
