@@ -39,10 +39,10 @@ def _train_steps(model, flat, ddp, opt, cfg, steps=8, B=2, S=128):
 
 
 def test_llama_tiny_step_gpu():
-    from torchx_amd.models.llama import LlamaModel, llama_tiny
+    from torchx_amd.models.llama import LlamaModel, llama_gpu_tiny
     from torchx_amd.parallel import FlatAdamW, FlatDDP, FlatParams
 
-    cfg = llama_tiny()
+    cfg = llama_gpu_tiny()
     dev = torch.device("cuda", 0)
     model = LlamaModel(cfg, device=dev)
     flat = FlatParams(model, dev)
@@ -55,10 +55,10 @@ def test_llama_tiny_step_gpu():
 
 
 def test_mixtral_tiny_step_gpu():
-    from torchx_amd.models.mixtral import MixtralModel, mixtral_tiny
+    from torchx_amd.models.mixtral import MixtralModel, mixtral_gpu_tiny
     from torchx_amd.parallel import FlatAdamW, FlatDDP, FlatParams
 
-    cfg = mixtral_tiny()
+    cfg = mixtral_gpu_tiny()
     dev = torch.device("cuda", 0)
     model = MixtralModel(cfg, device=dev)
     flat = FlatParams(model, dev)

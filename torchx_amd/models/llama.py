@@ -61,6 +61,16 @@ def llama_tiny(vocab: int = 512) -> LlamaConfig:
     )
 
 
+def llama_gpu_tiny(vocab: int = 512) -> LlamaConfig:
+    """Smallest config on the HIP-kernel hot path (the kernels specialize
+    on production shapes: hidden multiple of 2048, head_dim 128)."""
+    return LlamaConfig(
+        vocab_size=vocab, hidden_size=2048, intermediate_size=4096,
+        num_layers=2, num_heads=8, num_kv_heads=4, head_dim=128,
+        max_seq_len=512,
+    )
+
+
 class LlamaBlock(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()

@@ -54,6 +54,16 @@ def mixtral_tiny(vocab: int = 512) -> MixtralConfig:
     )
 
 
+def mixtral_gpu_tiny(vocab: int = 512) -> MixtralConfig:
+    """Smallest MoE config on the HIP-kernel hot path (hidden 2048,
+    head_dim 128 — see llama.llama_gpu_tiny)."""
+    return MixtralConfig(
+        vocab_size=vocab, hidden_size=2048, intermediate_size=2048,
+        num_layers=2, num_heads=8, num_kv_heads=4, head_dim=128,
+        max_seq_len=512, num_experts=4, top_k=2,
+    )
+
+
 class Expert(nn.Module):
     def __init__(self, cfg: MixtralConfig):
         super().__init__()

@@ -36,11 +36,19 @@ from torchx_amd.specs import (
     macros,
     runopts,
 )
+from torchx_amd.specs.overlays import OverlaySpec, apply_overlay
 
 from .api import DescribeAppResponse, ListAppResponse, Scheduler, Stream
 from .ids import make_unique
 
 log = logging.getLogger(__name__)
+
+# overlay surface: users patch the generated pod via
+# set_overlay(role, "kubernetes", "V1Pod", {...}); fields owned by the Role
+# are rejected at write time
+V1POD_OVERLAY = OverlaySpec(
+    "kubernetes", "V1Pod", blocklist=("command", "env", "image")
+)
 
 # reserve headroom for k8s daemons (parity: RESERVED_MILLICPU/MEMMB)
 RESERVED_MILLICPU = 100
@@ -188,6 +196,11 @@ def app_to_resource(app: AppDef, queue: str,
                 replica.env["TORCHX_RANK0_HOST"] = "localhost"
             name = sanitize_for_k8s(f"{role.name}-{replica_id}")
             pod = role_to_pod(name, replica, service_account)
+            # user pod overlay (reference parity: kubernetes_scheduler.py:164
+            # _apply_pod_overlay); command/env/resources come from the Role
+            pod_overlay = V1POD_OVERLAY.get(role)
+            if pod_overlay:
+                apply_overlay(pod, pod_overlay)
             pod["metadata"].setdefault("labels", {}).update(
                 {
                     LABEL_APP_NAME: sanitize_for_k8s(app.name),

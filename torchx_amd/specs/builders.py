@@ -154,11 +154,13 @@ def materialize_appdef(
         # everything from the first token that is not a known `--param value`
         # pair belongs to *args, preserving order (REMAINDER semantics that
         # also allow `--known` flags before the tail)
-        known_flags = {
-            f"--{n}"
-            for n, p in sig.parameters.items()
-            if p.kind != inspect.Parameter.VAR_POSITIONAL
-        }
+        known_flags = set()
+        for n, p in sig.parameters.items():
+            if p.kind == inspect.Parameter.VAR_POSITIONAL:
+                continue
+            known_flags.add(f"--{n}")
+            if len(n) == 1 and n != "h":
+                known_flags.add(f"-{n}")
         head: List[str] = []
         i = 0
         while i < len(args):

@@ -54,6 +54,20 @@ def _validate_component_fn(fn: Callable[..., Any], name: str) -> None:
 def _from_file(path: str, fn_name: str) -> _Component:
     if not os.path.isfile(path):
         raise ComponentNotFoundException(f"component file not found: {path}")
+    # AST lint before exec: typed args + AppDef return (reference parity:
+    # torchx/specs/finder.py:266 CustomComponentsFinder + file_linter)
+    from .file_linter import validate as _lint
+
+    msgs = _lint(path, fn_name)
+    if any("not found" in m.description for m in msgs):
+        raise ComponentNotFoundException(
+            f"function {fn_name!r} not found in {path}"
+        )
+    errors = [m for m in msgs if m.severity == "error"]
+    if errors:
+        raise ComponentValidationException(
+            "; ".join(f"{path}:{m.line}: {m.description}" for m in errors)
+        )
     ns = runpy.run_path(path)
     fn = ns.get(fn_name)
     if fn is None or not callable(fn):
