@@ -30,6 +30,8 @@ from torchx_amd.specs import (
     runopts,
 )
 
+from torchx_amd.workspace.docker_workspace import DockerWorkspaceMixin
+
 from .api import DescribeAppResponse, ListAppResponse, Scheduler, Stream
 from .devices import get_device_mounts
 from .ids import make_unique
@@ -84,9 +86,9 @@ def _replica_name(app_id: str, role: str, idx: int) -> str:
     return f"{app_id}-{role}-{idx}"
 
 
-class DockerScheduler(Scheduler[DockerJob]):
+class DockerScheduler(DockerWorkspaceMixin, Scheduler[DockerJob]):
     def __init__(self, session_name: str, client: Optional[Any] = None) -> None:
-        super().__init__("local_docker", session_name)
+        super().__init__("local_docker", session_name, docker_client=client)
         self.__client = client
 
     def _client(self) -> Any:

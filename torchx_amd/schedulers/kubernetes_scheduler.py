@@ -38,6 +38,8 @@ from torchx_amd.specs import (
 )
 from torchx_amd.specs.overlays import OverlaySpec, apply_overlay
 
+from torchx_amd.workspace.docker_workspace import DockerWorkspaceMixin
+
 from .api import DescribeAppResponse, ListAppResponse, Scheduler, Stream
 from .ids import make_unique
 
@@ -253,7 +255,7 @@ class KubernetesJob:
         return yaml.dump(self.resource, sort_keys=False)
 
 
-class KubernetesScheduler(Scheduler[KubernetesJob]):
+class KubernetesScheduler(DockerWorkspaceMixin, Scheduler[KubernetesJob]):
     def __init__(self, session_name: str, client: Optional[Any] = None) -> None:
         super().__init__("kubernetes", session_name)
         self.__client = client
@@ -295,6 +297,9 @@ class KubernetesScheduler(Scheduler[KubernetesJob]):
     def _submit_dryrun(
         self, app: AppDef, cfg: Mapping[str, Any]
     ) -> AppDryRunInfo[KubernetesJob]:
+        # locally-built sha256 images must be pushed to image_repo before
+        # the cluster can pull them (reference parity: k8s_scheduler.py:824)
+        images_to_push = self.dryrun_push_images(app, dict(cfg))
         resource = app_to_resource(
             app,
             queue=str(cfg.get("queue")),
@@ -302,12 +307,14 @@ class KubernetesScheduler(Scheduler[KubernetesJob]):
             priority_class=cfg.get("priority_class"),
         )
         job = KubernetesJob(
-            resource=resource, namespace=str(cfg.get("namespace", "default"))
+            resource=resource, images_to_push=images_to_push,
+            namespace=str(cfg.get("namespace", "default")),
         )
         return AppDryRunInfo(request=job, fmt=str)
 
     def schedule(self, dryrun_info: AppDryRunInfo[KubernetesJob]) -> str:
         req = dryrun_info.request
+        self.push_images(req.images_to_push)
         api = self._custom_api()
         resp = api.create_namespaced_custom_object(
             group="batch.volcano.sh",
