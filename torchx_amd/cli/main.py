@@ -165,10 +165,15 @@ def cmd_log(args: argparse.Namespace) -> int:
             if replica is not None and k != replica:
                 continue
             targets.append((r, k))
-    for r, k in targets:
-        for line in runner.log_lines(base, r, k, regex=args.regex,
-                                     should_tail=args.follow):
-            print(f"{r}/{k} {line}", flush=True)
+    from torchx_amd.utils.log_tee import print_log_lines
+
+    # one thread per (role, replica), colored interleaved prefixes
+    # (reference parity: cli/cmd_log.py:144-163)
+    print_log_lines(
+        targets,
+        lambda r, k: runner.log_lines(base, r, k, regex=args.regex,
+                                      should_tail=args.follow),
+    )
     return 0
 
 

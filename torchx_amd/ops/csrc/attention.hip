@@ -135,8 +135,89 @@ __device__ __forceinline__ void stage_tr(
 }
 
 // ---------------------------------------------------------------------------
-// Forward
+// Forward v2 (guide Appendix B ladder): KVBLK=64, double-buffered K tiles
+// arriving by async global_load_lds into an XOR-swizzled LDS image (T2+T3),
+// V reg-staged and transposed AFTER the barrier (T14 async-STAGE split),
+// exp2-domain online softmax (v_exp_f32 native).
 // ---------------------------------------------------------------------------
+
+#define FKV 64                      // kv rows per staged tile
+#define KIMG_BYTES (FKV * 256)      // [64 rows][256 B], byte ^= (row&15)<<4
+#define VIMG_BYTES (HD * FKV * 2)   // V^T [128 d][128 B], byte ^= (d&7)<<4
+#define LOG2E 1.44269504088896340736f
+#define LN2 0.69314718055994530942f
+
+// K fragment from the swizzled K image: row k = sub*32 + (lane&31),
+// d-slice c -> 16 B at [32c + (lane>>5)*16] ^ swizzle.
+__device__ __forceinline__ mbf16x8 kimg_frag(const char* kimg, int sub, int c) {
+  const int lane = threadIdx.x & 63;
+  const int row = sub * 32 + (lane & 31);
+  const int byte = (c * 32 + ((lane >> 5) * 16)) ^ ((row & 15) << 4);
+  return *(const mbf16x8*)(kimg + row * 256 + byte);
+}
+
+// V^T fragment: row d = dt*32 + (lane&31), k-slice ks -> 16 B.
+__device__ __forceinline__ mbf16x8 vimg_frag(const char* vimg, int dt, int ks) {
+  const int lane = threadIdx.x & 63;
+  const int d = dt * 32 + (lane & 31);
+  const int byte = (ks * 32 + ((lane >> 5) * 16)) ^ ((d & 7) << 4);
+  return *(const mbf16x8*)(vimg + d * 128 + byte);
+}
+
+// Async-stage one 64-row K tile: 16 global_load_lds_dwordx4 per block
+// (4 per wave), source-permuted so the lane-linear LDS image lands
+// swizzled (T2 note: swizzle moves to the SOURCE with glds staging).
+// Rows beyond S clamp to S-1 (values masked in softmax).
+__device__ __forceinline__ void stage_k_glds(
+    const unsigned short* __restrict__ kb, long kv0, long stride_elems,
+    int S, char* kimg) {
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  #pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int i = wave * 4 + j;
+    const int row = i * 4 + (lane >> 4);
+    const int colbyte = ((lane & 15) * 16) ^ ((row & 15) << 4);
+    long srow = kv0 + row;
+    if (srow >= S) srow = S - 1;
+    const char* src = (const char*)(kb + srow * stride_elems) + colbyte;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(kimg + i * 1024),
+        16, 0, 0);
+  }
+}
+
+// V tile -> registers (4 x b128 per thread); written transposed+swizzled
+// into the LDS V^T image after the barrier.
+__device__ __forceinline__ void load_v_regs(
+    const unsigned short* __restrict__ vb, long kv0, long stride_elems,
+    int S, ushort8 vr[4]) {
+  const int tid = threadIdx.x;
+  #pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int c = tid + p * 256;          // 1024 chunks = 64 rows x 16
+    const int r = c >> 4, g = c & 15;     // k row, d granule [8g, 8g+8)
+    long srow = kv0 + r;
+    if (srow >= S) srow = S - 1;
+    vr[p] = *(const ushort8*)(vb + srow * stride_elems + g * 8);
+  }
+}
+
+__device__ __forceinline__ void write_v_tr(const ushort8 vr[4], char* vimg) {
+  const int tid = threadIdx.x;
+  #pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int c = tid + p * 256;
+    const int r = c >> 4, g = c & 15;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d = 8 * g + j;
+      const int byte = (r * 2) ^ ((d & 7) << 4);
+      *(unsigned short*)(vimg + d * 128 + byte) = vr[p][j];
+    }
+  }
+}
 
 __global__ void __launch_bounds__(256, 2)
 attn_fwd_kernel(const unsigned short* __restrict__ q,
@@ -145,9 +226,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
                 unsigned short* __restrict__ o,
                 float* __restrict__ lse,  // [B,Hq,S]
                 int B, int S, int Hq, int Hkv, float scale, int causal) {
-  __shared__ __align__(16) unsigned short smem[KBLK * NAT_STRIDE + HD * TR_STRIDE];
-  unsigned short* k_nat = smem;
-  unsigned short* v_tr = smem + KBLK * NAT_STRIDE;
+  __shared__ __align__(16) char smem[2 * (KIMG_BYTES + VIMG_BYTES)];
 
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
   int bid = blockIdx.x;
@@ -185,75 +264,110 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
   }
 
   f32x16 acc_o[4] = {};
-  float m_run = -3.0e38f, l_run = 0.f;
+  float m_run = -3.0e38f, l_run = 0.f;   // m in exp2 domain
+  const float sc2 = scale * LOG2E;
 
   const int kv_limit = causal ? min(S, q0_blk + BLOCK_Q) : S;
-  const int ntiles = (kv_limit + KBLK - 1) / KBLK;
+  const int ntiles = (kv_limit + FKV - 1) / FKV;
   const int qw_max = min(qw0 + QBLK - 1, S - 1);
 
+  char* kcur = smem;
+  char* vcur = smem + KIMG_BYTES;
+  char* knxt = smem + KIMG_BYTES + VIMG_BYTES;
+  char* vnxt = knxt + KIMG_BYTES;
+
+  ushort8 vr[4];
+  // prologue: stage tile 0
+  stage_k_glds(kb, 0, kv_seq_stride, S, kcur);
+  load_v_regs(vb, 0, kv_seq_stride, S, vr);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __syncthreads();
+  write_v_tr(vr, vcur);
+  __syncthreads();
+
   for (int t = 0; t < ntiles; ++t) {
-    const int kv0 = t * KBLK;
-    __syncthreads();
-    stage_nat(kb, kv0, kv_seq_stride, KBLK, S, k_nat);
-    stage_tr(vb, kv0, kv_seq_stride, KBLK, S, v_tr, 1.0f);
-    __syncthreads();
+    const int kv0 = t * FKV;
+    const bool has_next = (t + 1) < ntiles;
+    if (has_next) {
+      // issue next tile's loads BEFORE compute (T14: HBM latency hides
+      // under this tile's MFMAs)
+      stage_k_glds(kb, kv0 + FKV, kv_seq_stride, S, knxt);
+      load_v_regs(vb, kv0 + FKV, kv_seq_stride, S, vr);
+    }
 
     const bool needed = wave_active && (!causal || kv0 <= qw_max);
-    if (!needed) continue;
-
-    // S^T[k, q] = K . Q^T
-    f32x16 acc = {};
-    #pragma unroll
-    for (int c = 0; c < 8; ++c) {
-      acc = mfma32(lds_frag(k_nat, NAT_STRIDE, c * 16), qfrag[c], acc);
-    }
-
-    float sv[16];
-    const bool mask_tile =
-        (causal && kv0 + KBLK - 1 > qw0) || (kv0 + KBLK > S) || (q_row >= S);
-    #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      float s = acc[r] * scale;
-      if (mask_tile) {
-        const long kg = kv0 + c_row(r, hi);
-        if (kg >= S || q_row >= S || (causal && kg > q_row)) s = -3.0e38f;
-      }
-      sv[r] = s;
-    }
-
-    // online softmax (lane owns q row; partner holds the other 16 k's)
-    float m_tile = sv[0];
-    #pragma unroll
-    for (int r = 1; r < 16; ++r) m_tile = fmaxf(m_tile, sv[r]);
-    m_tile = fmaxf(m_tile, __shfl_xor(m_tile, 32, 64));
-    const float m_new = fmaxf(m_run, m_tile);
-    const float alpha = __expf(m_run - m_new);
-
-    float p[16], row_sum = 0.f;
-    #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      p[r] = __expf(sv[r] - m_new);
-      row_sum += p[r];
-    }
-    row_sum += __shfl_xor(row_sum, 32, 64);
-    l_run = l_run * alpha + row_sum;
-    m_run = m_new;
-    #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
+    if (needed) {
+      // S^T[k, q] = K . Q^T over both 32-row k sub-tiles
+      f32x16 acc0 = {}, acc1 = {};
       #pragma unroll
-      for (int r = 0; r < 16; ++r) acc_o[dt][r] *= alpha;
+      for (int c = 0; c < 8; ++c) {
+        acc0 = mfma32(kimg_frag(kcur, 0, c), qfrag[c], acc0);
+        acc1 = mfma32(kimg_frag(kcur, 1, c), qfrag[c], acc1);
+      }
+
+      float sv[32];
+      #pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        sv[r] = acc0[r] * sc2;
+        sv[16 + r] = acc1[r] * sc2;
+      }
+      const bool mask_tile =
+          (causal && kv0 + FKV - 1 > qw0) || (kv0 + FKV > S) || (q_row >= S);
+      if (mask_tile) {
+        #pragma unroll
+        for (int r = 0; r < 32; ++r) {
+          const long kg = kv0 + (r >> 4) * 32 + c_row(r & 15, hi);
+          if (kg >= S || q_row >= S || (causal && kg > q_row)) {
+            sv[r] = -3.0e38f;
+          }
+        }
+      }
+
+      // online softmax, exp2 domain (lane owns q row; partner lane+32
+      // holds the other 16 k's of each sub-tile)
+      float m_tile = sv[0];
+      #pragma unroll
+      for (int r = 1; r < 32; ++r) m_tile = fmaxf(m_tile, sv[r]);
+      m_tile = fmaxf(m_tile, __shfl_xor(m_tile, 32, 64));
+      const float m_new = fmaxf(m_run, m_tile);
+      const float alpha = __builtin_amdgcn_exp2f(m_run - m_new);
+
+      float p[32], row_sum = 0.f;
+      #pragma unroll
+      for (int r = 0; r < 32; ++r) {
+        p[r] = __builtin_amdgcn_exp2f(sv[r] - m_new);
+        row_sum += p[r];
+      }
+      row_sum += __shfl_xor(row_sum, 32, 64);
+      l_run = l_run * alpha + row_sum;
+      m_run = m_new;
+      #pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) acc_o[dt][r] *= alpha;
+      }
+
+      mbf16x8 pf[4];
+      cvals_to_frags(p, hi, &pf[0], &pf[1]);
+      cvals_to_frags(p + 16, hi, &pf[2], &pf[3]);
+
+      // O^T[d, q] += V^T . P  (A = V^T from the swizzled LDS image)
+      #pragma unroll
+      for (int dt = 0; dt < 4; ++dt) {
+        #pragma unroll
+        for (int ks = 0; ks < 4; ++ks) {
+          acc_o[dt] = mfma32(vimg_frag(vcur, dt, ks), pf[ks], acc_o[dt]);
+        }
+      }
     }
 
-    mbf16x8 pf0, pf1;
-    cvals_to_frags(p, hi, &pf0, &pf1);
-
-    // O^T[d, q] += V^T . P  (A = V^T from transposed LDS image)
-    #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      acc_o[dt] = mfma32(lds_frag(v_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
-                         pf0, acc_o[dt]);
-      acc_o[dt] = mfma32(lds_frag(v_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
-                         pf1, acc_o[dt]);
+    if (has_next) {
+      asm volatile("s_waitcnt vmcnt(0)");
+      __syncthreads();
+      write_v_tr(vr, vnxt);
+      __syncthreads();
+      char* tk = kcur; kcur = knxt; knxt = tk;
+      char* tv = vcur; vcur = vnxt; vnxt = tv;
     }
   }
 
@@ -275,7 +389,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
     }
   }
   if (!hi) {
-    lse[((long)b * Hq + hq) * S + q_row] = m_run + __logf(l_run);
+    // lse stays in the natural-log domain for the backward kernels
+    lse[((long)b * Hq + hq) * S + q_row] = (m_run + __log2f(l_run)) * LN2;
   }
 }
 
