@@ -3,7 +3,11 @@
 (B=4, S=4096, Hq=32, Hkv=8, D=128, causal bf16)."""
 
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
@@ -33,6 +37,8 @@ def main():
     p.add_argument("--Hq", type=int, default=32)
     p.add_argument("--Hkv", type=int, default=8)
     p.add_argument("--iters", type=int, default=20)
+    p.add_argument("--mode", type=str, default="both",
+                   choices=["fwd", "bwd", "both"])
     args = p.parse_args()
     B, S, Hq, Hkv, D = args.B, args.S, args.Hq, args.Hkv, 128
 
@@ -50,17 +56,18 @@ def main():
         return hip.attn_fwd(q, k, v, scale, True)
 
     o, lse = fwd()
-    t_fwd = time_fn(fwd, args.iters)
-    tf_fwd = flops_fwd(B, S, Hq, D, True) / t_fwd / 1e12
+    if args.mode in ("fwd", "both"):
+        t_fwd = time_fn(fwd, args.iters)
+        tf_fwd = flops_fwd(B, S, Hq, D, True) / t_fwd / 1e12
+        print(f"fwd: {t_fwd*1e3:.3f} ms  {tf_fwd:.0f} TF/s")
 
-    def bwd():
-        return hip.attn_bwd(q, k, v, o, do, lse, scale, True)
+    if args.mode in ("bwd", "both"):
+        def bwd():
+            return hip.attn_bwd(q, k, v, o, do, lse, scale, True)
 
-    t_bwd = time_fn(bwd, args.iters)
-    tf_bwd = flops_fwd(B, S, Hq, D, True) * 2.5 / t_bwd / 1e12
-
-    print(f"fwd: {t_fwd*1e3:.3f} ms  {tf_fwd:.0f} TF/s")
-    print(f"bwd: {t_bwd*1e3:.3f} ms  {tf_bwd:.0f} TF/s")
+        t_bwd = time_fn(bwd, args.iters)
+        tf_bwd = flops_fwd(B, S, Hq, D, True) * 2.5 / t_bwd / 1e12
+        print(f"bwd: {t_bwd*1e3:.3f} ms  {tf_bwd:.0f} TF/s")
 
 
 if __name__ == "__main__":

@@ -42,9 +42,13 @@ __device__ __forceinline__ f32x16 mfma32(mbf16x8 a, mbf16x8 b, f32x16 c) {
   return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
 }
 
+// v_cvt_pk_bf16_f32: 2 f32 -> packed 2x bf16 (RNE) in ONE instruction
+// (no builtin on gfx950 — guide T12 recipe; the manual RNE pack is ~6 VALU
+// ops per value and dominated the softmax)
 __device__ __forceinline__ unsigned int pack_bf16x2(float a, float b) {
-  return (unsigned int)f32_to_bf16(a) |
-         ((unsigned int)f32_to_bf16(b) << 16);
+  unsigned int r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(a), "v"(b));
+  return r;
 }
 
 // C-layout row index of accumulator register r for this lane-half.
@@ -324,28 +328,35 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
       }
 
       // online softmax, exp2 domain (lane owns q row; partner lane+32
-      // holds the other 16 k's of each sub-tile)
+      // holds the other 16 k's of each sub-tile).  T13 defer-max: skip the
+      // O-wide rescale while per-tile max growth stays under THR2
+      // (P bounded by 2^THR2 ~= e^8, bf16-accum tolerates; the previous
+      // tile's PV is complete before this decision — the safe order).
       float m_tile = sv[0];
       #pragma unroll
       for (int r = 1; r < 32; ++r) m_tile = fmaxf(m_tile, sv[r]);
       m_tile = fmaxf(m_tile, __shfl_xor(m_tile, 32, 64));
-      const float m_new = fmaxf(m_run, m_tile);
-      const float alpha = __builtin_amdgcn_exp2f(m_run - m_new);
+      const float THR2 = 11.54f;  // 8 * log2(e)
+      if (!__all(m_tile - m_run <= THR2)) {
+        const float m_new = fmaxf(m_run, m_tile);
+        const float alpha = __builtin_amdgcn_exp2f(m_run - m_new);
+        l_run *= alpha;
+        m_run = m_new;
+        #pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          #pragma unroll
+          for (int r = 0; r < 16; ++r) acc_o[dt][r] *= alpha;
+        }
+      }
 
       float p[32], row_sum = 0.f;
       #pragma unroll
       for (int r = 0; r < 32; ++r) {
-        p[r] = __builtin_amdgcn_exp2f(sv[r] - m_new);
+        p[r] = __builtin_amdgcn_exp2f(sv[r] - m_run);
         row_sum += p[r];
       }
       row_sum += __shfl_xor(row_sum, 32, 64);
-      l_run = l_run * alpha + row_sum;
-      m_run = m_new;
-      #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        #pragma unroll
-        for (int r = 0; r < 16; ++r) acc_o[dt][r] *= alpha;
-      }
+      l_run += row_sum;
 
       mbf16x8 pf[4];
       cvals_to_frags(p, hi, &pf[0], &pf[1]);
