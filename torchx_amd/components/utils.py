@@ -212,3 +212,62 @@ def booth(x1: float, x2: float, trial_idx: int = 0,
             )
         ],
     )
+
+
+def from_yaml(config: str, *overrides: str) -> specs.AppDef:
+    """Build an AppDef from a YAML config file (parity with the reference's
+    ``utils.hydra`` component, torchx/components/utils.py:330 — implemented
+    dependency-free on plain YAML + dotted-path overrides).
+
+    The file must have an ``app`` key describing the AppDef; ``role``
+    entries are dicts of Role fields.  TorchX macros are available as
+    ``${torchx.app_id}``, ``${torchx.replica_id}``, ``${torchx.rank0_env}``,
+    ``${torchx.img_root}``.
+
+    Args:
+        config: path to the YAML config file
+        overrides: dotted-path overrides, e.g. ``app.roles.0.num_replicas=2``
+    """
+    import yaml
+
+    with open(config) as f:
+        text = f.read()
+    for name, macro in (
+        ("app_id", specs.macros.app_id),
+        ("replica_id", specs.macros.replica_id),
+        ("rank0_env", specs.macros.rank0_env),
+        ("img_root", specs.macros.img_root),
+    ):
+        text = text.replace("${torchx.%s}" % name, macro)
+    cfg = yaml.safe_load(text)
+    if not isinstance(cfg, dict) or "app" not in cfg:
+        raise ValueError(f"{config} must contain a top-level `app` key")
+
+    for ov in overrides:
+        path, _, value = ov.partition("=")
+        keys = path.split(".")
+        node = cfg
+        for k in keys[:-1]:
+            node = node[int(k)] if isinstance(node, list) else node[k]
+        leaf = keys[-1]
+        parsed = yaml.safe_load(value)
+        if isinstance(node, list):
+            node[int(leaf)] = parsed
+        else:
+            node[leaf] = parsed
+
+    app = cfg["app"]
+    roles = []
+    for r in app.get("roles", []):
+        res = r.pop("resource", None)
+        resource = (
+            specs.Resource(**res) if isinstance(res, dict)
+            else specs.named_resources()[res]() if isinstance(res, str)
+            else specs.NULL_RESOURCE
+        )
+        roles.append(specs.Role(resource=resource, **r))
+    return specs.AppDef(
+        name=app.get("name", "app"),
+        roles=roles,
+        metadata=app.get("metadata", {}),
+    )
