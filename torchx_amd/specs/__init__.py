@@ -45,6 +45,7 @@ from .api import (  # noqa: F401
     runopts,
 )
 from .builders import materialize_appdef, parse_mounts  # noqa: F401
+from .capabilities import GFX_ARCH, HBM_GB, XGMI_LINKS, CapabilityKey
 from .named_resources import NAMED_RESOURCES
 
 _lock = threading.Lock()
