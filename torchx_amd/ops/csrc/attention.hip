@@ -235,6 +235,32 @@ __device__ __forceinline__ void write_v_tr(const ushort8 vr[4], char* vimg) {
   }
 }
 
+
+// T1 XCD-aware block swizzle: the dispatcher hands consecutive blockIdx to
+// consecutive XCDs (each with its own 4 MB L2).  Group the G*nqt blocks
+// that read the same (b, hkv) K/V stream onto ONE XCD so K/V go L2-resident
+// after the first touch (these kernels are otherwise HBM-bound at
+// AI ~= 128 FLOP/B).  Falls back to the linear map when B*Hkv % 8 != 0.
+__device__ __forceinline__ void map_block_fwd(int B, int Hq, int Hkv,
+                                              int nqt, int G,
+                                              int* b, int* hq, int* qt) {
+  int bid = blockIdx.x;
+  if (((B * Hkv) & 7) == 0) {
+    const int gsz = G * nqt;
+    const int xcd = bid & 7, slot = bid >> 3;
+    const int g = (slot / gsz) * 8 + xcd;
+    *b = g / Hkv;
+    const int r = slot % gsz;
+    *hq = (g % Hkv) * G + r / nqt;
+    *qt = r % nqt;
+  } else {
+    *b = bid / (Hq * nqt);
+    bid -= *b * Hq * nqt;
+    *hq = bid / nqt;
+    *qt = bid % nqt;
+  }
+}
+
 __global__ void __launch_bounds__(256, 2)
 attn_fwd_kernel(const unsigned short* __restrict__ q,
                 const unsigned short* __restrict__ k,
@@ -242,15 +268,15 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
                 unsigned short* __restrict__ o,
                 float* __restrict__ lse,  // [B,Hq,S]
                 int B, int S, int Hq, int Hkv, float scale, int causal) {
-  __shared__ __align__(16) char smem[2 * (KIMG_BYTES + VIMG_BYTES)];
+  // 3-deep K ring (glds prefetch 2 tiles ahead, counted vmcnt keeps the
+  // newest tile's loads in flight ACROSS the barrier — T3/T4) + 2-deep V
+  __shared__ __align__(16) char smem[3 * KIMG_BYTES + 2 * VIMG_BYTES];
 
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
-  int bid = blockIdx.x;
-  const int b = bid / (Hq * nqt);
-  bid -= b * Hq * nqt;
-  const int hq = bid / nqt;
-  const int qt = bid % nqt;
-  const int hkv = hq / (Hq / Hkv);
+  const int G = Hq / Hkv;
+  int b, hq, qt;
+  map_block_fwd(B, Hq, Hkv, nqt, G, &b, &hq, &qt);
+  const int hkv = hq / G;
 
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x & 63;
@@ -287,17 +313,22 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
   const int ntiles = (kv_limit + FKV - 1) / FKV;
   const int qw_max = min(qw0 + QBLK - 1, S - 1);
 
-  char* kcur = smem;
-  char* vcur = smem + KIMG_BYTES;
-  char* knxt = smem + KIMG_BYTES + VIMG_BYTES;
-  char* vnxt = knxt + KIMG_BYTES;
+  char* k0 = smem;                       // K tile t   (cur)
+  char* k1 = smem + KIMG_BYTES;          // K tile t+1 (staged, waited)
+  char* k2 = smem + 2 * KIMG_BYTES;      // K tile t+2 (in flight)
+  char* vcur = smem + 3 * KIMG_BYTES;
+  char* vnxt = vcur + VIMG_BYTES;
 
   ushort8 vr[4];
-  // prologue: stage tile 0
-  stage_k_glds(kb, 0, kv_seq_stride, S, kcur);
+  // prologue: K0, V0 (waited), K1 (stays in flight past the barrier)
+  stage_k_glds(kb, 0, kv_seq_stride, S, k0);
   load_v_regs(vb, 0, kv_seq_stride, S, vr);
-  asm volatile("s_waitcnt vmcnt(0)");
-  __syncthreads();
+  if (ntiles > 1) {
+    stage_k_glds(kb, FKV, kv_seq_stride, S, k1);
+    asm volatile("s_waitcnt vmcnt(4)");  // K0 + V0 done; K1 in flight
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)");
+  }
   write_v_tr(vr, vcur);
   __syncthreads();
 
@@ -305,10 +336,12 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
     const int kv0 = t * FKV;
     const bool has_next = (t + 1) < ntiles;
     if (has_next) {
-      // issue next tile's loads BEFORE compute (T14: HBM latency hides
-      // under this tile's MFMAs)
-      stage_k_glds(kb, kv0 + FKV, kv_seq_stride, S, knxt);
+      // issue next-V and next-next-K BEFORE compute (T14: HBM latency
+      // hides under this tile's MFMAs; K gets ~2 tiles of cover)
       load_v_regs(vb, kv0 + FKV, kv_seq_stride, S, vr);
+      if (t + 2 < ntiles) {
+        stage_k_glds(kb, kv0 + 2 * FKV, kv_seq_stride, S, k2);
+      }
     }
 
     const bool needed = wave_active && (!causal || kv0 <= qw_max);
@@ -317,8 +350,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
       f32x16 acc0 = {}, acc1 = {};
       #pragma unroll
       for (int c = 0; c < 8; ++c) {
-        acc0 = mfma32(kimg_frag(kcur, 0, c), qfrag[c], acc0);
-        acc1 = mfma32(kimg_frag(kcur, 1, c), qfrag[c], acc1);
+        acc0 = mfma32(kimg_frag(k0, 0, c), qfrag[c], acc0);
+        acc1 = mfma32(kimg_frag(k0, 1, c), qfrag[c], acc1);
       }
 
       float sv[32];
@@ -385,11 +418,18 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
     }
 
     if (has_next) {
-      asm volatile("s_waitcnt vmcnt(0)");
-      __syncthreads();
+      if (t + 2 < ntiles) {
+        // oldest 8 outstanding = K(t+1) + V(t+1); K(t+2) stays in flight
+        asm volatile("s_waitcnt vmcnt(4)");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)");
+      }
       write_v_tr(vr, vnxt);
+      // one barrier per tile: K(t+1)/V(t+1) stores visible to all waves;
+      // the buffers being overwritten were last read before the PREVIOUS
+      // barrier, so no front barrier is needed
       __syncthreads();
-      char* tk = kcur; kcur = knxt; knxt = tk;
+      char* tk = k0; k0 = k1; k1 = k2; k2 = tk;
       char* tv = vcur; vcur = vnxt; vnxt = tv;
     }
   }
@@ -463,12 +503,10 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   unsigned short* v_nat = k_tr + HD * TR_STRIDE;
 
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
-  int bid = blockIdx.x;
-  const int b = bid / (Hq * nqt);
-  bid -= b * Hq * nqt;
-  const int hq = bid / nqt;
-  const int qt = bid % nqt;
-  const int hkv = hq / (Hq / Hkv);
+  const int G = Hq / Hkv;
+  int b, hq, qt;
+  map_block_fwd(B, Hq, Hkv, nqt, G, &b, &hq, &qt);
+  const int hkv = hq / G;
 
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x & 63;
