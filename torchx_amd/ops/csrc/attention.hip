@@ -496,11 +496,16 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
                    const float* __restrict__ delta,
                    unsigned short* __restrict__ dq,
                    int B, int S, int Hq, int Hkv, float scale, int causal) {
-  __shared__ __align__(16) unsigned short smem[KBLK * NAT_STRIDE + HD * TR_STRIDE +
-                                 KBLK * NAT_STRIDE];
-  unsigned short* k_nat = smem;
-  unsigned short* k_tr = smem + KBLK * NAT_STRIDE;                // [128][40]
-  unsigned short* v_nat = k_tr + HD * TR_STRIDE;
+  // v2: KV tiles of 64; K and V natural images arrive by async
+  // global_load_lds (2-deep rings, swizzled); K^T single rot-placed image
+  // reg-staged with the T14 split; exp2-domain probabilities.  80 KiB LDS
+  // -> 2 blocks/CU.
+  __shared__ __align__(16) char smem[4 * KIMG_BYTES + VIMG_BYTES];
+  char* kcur = smem;
+  char* knxt = smem + KIMG_BYTES;
+  char* vcur = smem + 2 * KIMG_BYTES;
+  char* vnxt = smem + 3 * KIMG_BYTES;
+  char* ktr = smem + 4 * KIMG_BYTES;
 
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
   const int G = Hq / Hkv;
@@ -526,71 +531,108 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   const bool wave_active = qw0 < S;
   const bool row_valid = q_row < S;
 
-  mbf16x8 qfrag[8], dofrag[8];
-  float my_lse = 0.f, my_delta = 0.f;
+  mbf16x8 qfrag[8];
+  float my_lse2 = 0.f, my_delta = 0.f;
+  const long qrow_safe = row_valid ? q_row : (S - 1);
+  const unsigned short* dop =
+      dob + qrow_safe * q_seq_stride + (hi ? 8 : 0);
   {
-    const long row = row_valid ? q_row : (S - 1);
-    const unsigned short* qp = qb + row * q_seq_stride + (hi ? 8 : 0);
-    const unsigned short* dp = dob + row * q_seq_stride + (hi ? 8 : 0);
+    const unsigned short* qp = qb + qrow_safe * q_seq_stride + (hi ? 8 : 0);
     #pragma unroll
     for (int c = 0; c < 8; ++c) {
       qfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(qp + c * 16));
-      dofrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(dp + c * 16));
     }
     if (row_valid) {
-      my_lse = lse[((long)b * Hq + hq) * S + q_row];
+      my_lse2 = lse[((long)b * Hq + hq) * S + q_row] * LOG2E;
       my_delta = delta[((long)b * Hq + hq) * S + q_row];
     }
   }
 
   f32x16 acc_dq[4] = {};
+  const float sc2 = scale * LOG2E;
 
   const int kv_limit = causal ? min(S, q0_blk + BLOCK_Q) : S;
-  const int ntiles = (kv_limit + KBLK - 1) / KBLK;
+  const int ntiles = (kv_limit + FKV - 1) / FKV;
   const int qw_max = min(qw0 + QBLK - 1, S - 1);
 
+  ushort8 vr[4];
+  stage_k_glds(kb, 0, kv_seq_stride, S, kcur);
+  stage_k_glds(vb, 0, kv_seq_stride, S, vcur);
+  load_v_regs(kb, 0, kv_seq_stride, S, vr);
+  asm volatile("s_waitcnt vmcnt(0)");
+  __syncthreads();
+  write_v_tr(vr, ktr);
+  __syncthreads();
+
   for (int t = 0; t < ntiles; ++t) {
-    const int kv0 = t * KBLK;
-    __syncthreads();
-    stage_nat(kb, kv0, kv_seq_stride, KBLK, S, k_nat);
-    stage_tr(kb, kv0, kv_seq_stride, KBLK, S, k_tr, 1.0f);
-    stage_nat(vb, kv0, kv_seq_stride, KBLK, S, v_nat);
-    __syncthreads();
+    const int kv0 = t * FKV;
+    const bool has_next = (t + 1) < ntiles;
+    if (has_next) {
+      // next tile's glds issue BEFORE compute; with the XCD swizzle the
+      // K/V stream is L2-resident, one tile of cover is plenty
+      stage_k_glds(kb, kv0 + FKV, kv_seq_stride, S, knxt);
+      stage_k_glds(vb, kv0 + FKV, kv_seq_stride, S, vnxt);
+    }
 
     const bool needed = wave_active && (!causal || kv0 <= qw_max);
-    if (!needed) continue;
-
-    f32x16 acc_s = {}, acc_dp = {};
-    #pragma unroll
-    for (int c = 0; c < 8; ++c) {
-      acc_s = mfma32(lds_frag(k_nat, NAT_STRIDE, c * 16), qfrag[c], acc_s);
-      acc_dp = mfma32(lds_frag(v_nat, NAT_STRIDE, c * 16), dofrag[c], acc_dp);
-    }
-
-    const bool mask_tile =
-        (causal && kv0 + KBLK - 1 > qw0) || (kv0 + KBLK > S) || !row_valid;
-    float ds[16];
-    #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      float s = acc_s[r] * scale;
-      if (mask_tile) {
-        const long kg = kv0 + c_row(r, hi);
-        if (kg >= S || !row_valid || (causal && kg > q_row)) s = -3.0e38f;
+    if (needed) {
+      // dO fragments re-read each tile (L2-hot; keeping them resident
+      // costs 32 VGPR across the staging phase and spills)
+      mbf16x8 dofrag[8];
+      #pragma unroll
+      for (int c = 0; c < 8; ++c) {
+        dofrag[c] =
+            __builtin_bit_cast(mbf16x8, *(const ushort8*)(dop + c * 16));
       }
-      const float pr = __expf(s - my_lse);
-      ds[r] = scale * pr * (acc_dp[r] - my_delta);
+      // two 32-row k sub-tiles fully sequentially (no running max to
+      // couple them -> half the live accumulators of a fused pass; the
+      // loop must NOT unroll or both subs' accumulators go live at once)
+      #pragma clang loop unroll(disable)
+      for (int sb = 0; sb < 2; ++sb) {
+        f32x16 acc_s = {}, acc_dp = {};
+        #pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          acc_s = mfma32(kimg_frag(kcur, sb, c), qfrag[c], acc_s);
+          acc_dp = mfma32(kimg_frag(vcur, sb, c), dofrag[c], acc_dp);
+        }
+        const int k0 = kv0 + 32 * sb;
+        const bool mask_tile =
+            (causal && k0 + 31 > qw0) || (k0 + 32 > S) || !row_valid;
+        float ds[16];
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float s2 = acc_s[r] * sc2;
+          if (mask_tile) {
+            const long kg = k0 + c_row(r, hi);
+            if (kg >= S || !row_valid || (causal && kg > q_row)) {
+              s2 = -3.0e38f;
+            }
+          }
+          const float pr = __builtin_amdgcn_exp2f(s2 - my_lse2);
+          ds[r] = scale * pr * (acc_dp[r] - my_delta);
+        }
+        mbf16x8 df0, df1;
+        cvals_to_frags(ds, hi, &df0, &df1);
+        // dQ^T[d, q] += K^T . dS^T
+        #pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          acc_dq[dt] = mfma32(vimg_frag(ktr, dt, 2 * sb), df0, acc_dq[dt]);
+          acc_dq[dt] =
+              mfma32(vimg_frag(ktr, dt, 2 * sb + 1), df1, acc_dq[dt]);
+        }
+      }
     }
 
-    mbf16x8 df0, df1;
-    cvals_to_frags(ds, hi, &df0, &df1);
-
-    // dQ^T[d, q] += K^T . dS^T
-    #pragma unroll
-    for (int dt = 0; dt < 4; ++dt) {
-      acc_dq[dt] = mfma32(lds_frag(k_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
-                          df0, acc_dq[dt]);
-      acc_dq[dt] = mfma32(lds_frag(k_tr + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
-                          df1, acc_dq[dt]);
+    if (has_next) {
+      // K^T regs loaded here (L2-hit) so vr is live only briefly;
+      // keeping it across compute costs 16 VGPR at the pressure peak
+      load_v_regs(kb, kv0 + FKV, kv_seq_stride, S, vr);
+      asm volatile("s_waitcnt vmcnt(0)");
+      __syncthreads();   // all waves done reading ktr for tile t
+      write_v_tr(vr, ktr);
+      __syncthreads();   // ktr + glds images visible
+      char* tk = kcur; kcur = knxt; knxt = tk;
+      char* tv = vcur; vcur = vnxt; vnxt = tv;
     }
   }
 
@@ -619,6 +661,19 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
 // wave/SIMD with zero latency hiding; the S^T recompute in the second pass
 // costs 8 extra MFMAs/tile but doubles occupancy).
 // ---------------------------------------------------------------------------
+// Natural swizzled write of a reg-staged [64][128] tile (same image form
+// as the glds K staging: byte ^= (row&15)<<4; vectorized b128, no
+// conflicts: 16 consecutive lanes write one row's 16 granules permuted).
+__device__ __forceinline__ void write_nat_sw(const ushort8 vr[4], char* img) {
+  const int tid = threadIdx.x;
+  #pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int c = tid + p * 256;
+    const int r = c >> 4, g = c & 15;
+    *(ushort8*)(img + r * 256 + ((g * 16) ^ ((r & 15) << 4))) = vr[p];
+  }
+}
+
 template <bool DK>
 __global__ void __launch_bounds__(256, 2)
 attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
@@ -629,21 +684,39 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     const float* __restrict__ delta,
                     unsigned short* __restrict__ out,  // dk or dv
                     int B, int S, int Hq, int Hkv, float scale, int causal) {
-  // dK pass stages {Q nat, dO nat, Q^T}; dV pass stages {Q nat, dO^T}
-  __shared__ __align__(16) unsigned short smem[2 * QBLK * NAT_STRIDE +
-                                               HD * TR_STRIDE];
-  __shared__ float lse_s[QBLK], del_s[QBLK];
-  unsigned short* q_nat = smem;
-  unsigned short* do_nat = smem + QBLK * NAT_STRIDE;         // dK only
-  unsigned short* tr_tile = smem + 2 * QBLK * NAT_STRIDE;    // q_tr / do_tr
+  // v2: q tiles of 64; Q arrives by async global_load_lds (2-deep
+  // swizzled ring); the transposed tile (dO^T for dV / Q^T for dK) and the
+  // dK pass's natural dO are reg-staged after the barrier (T14).  Split
+  // dV/dK passes keep 2 waves/SIMD.  LDS: 32K Q ring + 16K TR (+16K dO
+  // nat for dK) + lse/delta rows.
+  __shared__ __align__(16) char smem[2 * KIMG_BYTES + VIMG_BYTES +
+                                     (DK ? KIMG_BYTES : 0)];
+  __shared__ float lse_s[FKV], del_s[FKV];
+  char* qcur = smem;
+  char* qnxt = smem + KIMG_BYTES;
+  char* tr_img = smem + 2 * KIMG_BYTES;              // dO^T / Q^T [128][64]
+  char* do_img = smem + 2 * KIMG_BYTES + VIMG_BYTES; // dK only, [64][256B]
 
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
-  int bid = blockIdx.x;
-  const int b = bid / (Hkv * nkt);
-  bid -= b * Hkv * nkt;
-  const int hkv = bid / nkt;
-  const int kt = bid % nkt;
   const int G = Hq / Hkv;
+  int b, hkv, kt;
+  {
+    // XCD-aware swizzle: the nkt blocks streaming the same (b, hkv)
+    // Q/dO group go to one XCD (same rationale as map_block_fwd)
+    int bid = blockIdx.x;
+    if (((B * Hkv) & 7) == 0) {
+      const int xcd = bid & 7, slot = bid >> 3;
+      const int g = (slot / nkt) * 8 + xcd;
+      b = g / Hkv;
+      hkv = g % Hkv;
+      kt = slot % nkt;
+    } else {
+      b = bid / (Hkv * nkt);
+      bid -= b * Hkv * nkt;
+      hkv = bid / nkt;
+      kt = bid % nkt;
+    }
+  }
 
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x & 63;
@@ -661,93 +734,163 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   const bool wave_active = kw0 < S;
   const bool krow_valid = k_row < S;
 
-  // K (and V for the dK pass) fragments resident in registers (lane = k row)
-  mbf16x8 kfrag[8], vfrag[8];
+  // K fragments resident in registers; the dK pass re-reads its V
+  // fragments per tile from L2 (keeping them resident costs 32 VGPR at
+  // the staging-phase pressure peak and spills)
+  mbf16x8 kfrag[8];
+  const long krow_safe = krow_valid ? k_row : (S - 1);
+  const unsigned short* vp = vb + krow_safe * kv_seq_stride + (hi ? 8 : 0);
   {
-    const long row = krow_valid ? k_row : (S - 1);
-    const unsigned short* kp = kb + row * kv_seq_stride + (hi ? 8 : 0);
-    const unsigned short* vp = vb + row * kv_seq_stride + (hi ? 8 : 0);
+    const unsigned short* kp = kb + krow_safe * kv_seq_stride + (hi ? 8 : 0);
     #pragma unroll
     for (int c = 0; c < 8; ++c) {
       kfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(kp + c * 16));
-      if (DK) {
-        vfrag[c] =
-            __builtin_bit_cast(mbf16x8, *(const ushort8*)(vp + c * 16));
-      }
     }
   }
 
   f32x16 acc[4] = {};
+  const float sc2 = scale * LOG2E;
 
-  const int t0 = causal ? (kv0_blk / QBLK) : 0;
-  const int nt = (S + QBLK - 1) / QBLK;
+  const int t0 = causal ? (kv0_blk / FKV) : 0;
+  const int nt = (S + FKV - 1) / FKV;
+  const int nt_eff = nt - t0;
+  const int total = G * nt_eff;      // flattened (gh, t) iterations
 
-  for (int gh = 0; gh < G; ++gh) {
+  // stage helpers for iteration idx: q tile t of head gh
+  auto stage_glds = [&](int idx, char* dst) {
+    const int gh = idx / nt_eff;
+    const int t = t0 + idx % nt_eff;
     const int hq = hkv * G + gh;
-    const unsigned short* qb = q + (long)b * S * q_seq_stride + (long)hq * HD;
-    const unsigned short* dob =
+    const unsigned short* qb =
+        q + (long)b * S * q_seq_stride + (long)hq * HD;
+    stage_k_glds(qb, (long)t * FKV, q_seq_stride, S, dst);
+  };
+  auto load_tr_regs = [&](int idx, ushort8 vr[4]) {
+    const int gh = idx / nt_eff;
+    const int t = t0 + idx % nt_eff;
+    const int hq = hkv * G + gh;
+    const unsigned short* src = DK ? q : dout;
+    const unsigned short* sb_ =
+        src + (long)b * S * q_seq_stride + (long)hq * HD;
+    load_v_regs(sb_, (long)t * FKV, q_seq_stride, S, vr);
+  };
+  auto load_nat_regs = [&](int idx, ushort8 vr[4]) {  // dK only: dO
+    const int gh = idx / nt_eff;
+    const int t = t0 + idx % nt_eff;
+    const int hq = hkv * G + gh;
+    const unsigned short* sb_ =
         dout + (long)b * S * q_seq_stride + (long)hq * HD;
+    load_v_regs(sb_, (long)t * FKV, q_seq_stride, S, vr);
+  };
+  auto stage_lse = [&](int idx) {
+    const int gh = idx / nt_eff;
+    const int t = t0 + idx % nt_eff;
+    const int hq = hkv * G + gh;
+    const int q0 = t * FKV;
     const float* lse_b = lse + ((long)b * Hq + hq) * S;
-    const float* del_b = delta + ((long)b * Hq + hq) * S;
+    if (threadIdx.x < FKV) {
+      const int qg = q0 + threadIdx.x;
+      lse_s[threadIdx.x] = (qg < S) ? lse_b[qg] * LOG2E : 0.f;
+    } else if (DK && threadIdx.x < 2 * FKV) {
+      const float* del_b = delta + ((long)b * Hq + hq) * S;
+      const int qg = q0 + (threadIdx.x - FKV);
+      del_s[threadIdx.x - FKV] = (qg < S) ? del_b[qg] : 0.f;
+    }
+  };
 
-    for (int t = t0; t < nt; ++t) {
-      const int q0 = t * QBLK;
-      __syncthreads();
-      stage_nat(qb, q0, q_seq_stride, QBLK, S, q_nat);
-      if (DK) {
-        stage_nat(dob, q0, q_seq_stride, QBLK, S, do_nat);
-        stage_tr(qb, q0, q_seq_stride, QBLK, S, tr_tile, 1.0f);
-      } else {
-        stage_tr(dob, q0, q_seq_stride, QBLK, S, tr_tile, 1.0f);
-      }
-      if (threadIdx.x < QBLK) {
-        const int qg = q0 + threadIdx.x;
-        lse_s[threadIdx.x] = (qg < S) ? lse_b[qg] : 0.f;
-        if (DK) del_s[threadIdx.x] = (qg < S) ? del_b[qg] : 0.f;
-      }
-      __syncthreads();
+  ushort8 trr[4];
+  // prologue: stage iteration 0
+  stage_glds(0, qcur);
+  load_tr_regs(0, trr);
+  if (DK) {
+    ushort8 dor[4];
+    load_nat_regs(0, dor);
+    asm volatile("s_waitcnt vmcnt(0)");
+    write_nat_sw(dor, do_img);
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)");
+  }
+  write_v_tr(trr, tr_img);
+  stage_lse(0);
+  __syncthreads();
 
-      const bool needed =
-          wave_active && (!causal || q0 + QBLK - 1 >= kw0);
-      if (!needed) continue;
+  for (int idx = 0; idx < total; ++idx) {
+    const int gh = idx / nt_eff;
+    const int t = t0 + idx % nt_eff;
+    const int q0 = t * FKV;
+    const bool has_next = (idx + 1) < total;
 
-      // S[q, k] (lane = k col); dP[q, k] for the dK pass
-      f32x16 acc_s = {}, acc_dp = {};
-      #pragma unroll
-      for (int c = 0; c < 8; ++c) {
-        acc_s = mfma32(lds_frag(q_nat, NAT_STRIDE, c * 16), kfrag[c], acc_s);
+    const bool needed =
+        wave_active && (!causal || q0 + FKV - 1 >= kw0);
+    if (needed) {
+      #pragma clang loop unroll(disable)
+      for (int sb = 0; sb < 2; ++sb) {
+        // S[q-sub, k] (lane owns k col); dP for the dK pass.  The two
+        // MFMA chains run sequentially so only one B-operand set (kfrag
+        // or the per-chain V fragments) is live at a time.
+        f32x16 acc_s = {}, acc_dp = {};
+        #pragma unroll
+        for (int c = 0; c < 8; ++c) {
+          acc_s = mfma32(kimg_frag(qcur, sb, c), kfrag[c], acc_s);
+        }
         if (DK) {
-          acc_dp =
-              mfma32(lds_frag(do_nat, NAT_STRIDE, c * 16), vfrag[c], acc_dp);
+          #pragma unroll
+          for (int c = 0; c < 8; ++c) {
+            const mbf16x8 vf =
+                __builtin_bit_cast(mbf16x8, *(const ushort8*)(vp + c * 16));
+            acc_dp = mfma32(kimg_frag(do_img, sb, c), vf, acc_dp);
+          }
+        }
+        const int qs0 = q0 + 32 * sb;
+        const bool mask_tile = causal || (qs0 + 32 > S) || !krow_valid;
+        float cv[16];
+        #pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int ql = 32 * sb + c_row(r, hi);  // q offset within tile
+          float s2 = acc_s[r] * sc2;
+          if (mask_tile) {
+            const long qg = q0 + ql;
+            if (qg >= S || !krow_valid || (causal && k_row > qg)) {
+              s2 = -3.0e38f;
+            }
+          }
+          const float pr = __builtin_amdgcn_exp2f(s2 - lse_s[ql]);
+          cv[r] = DK ? scale * pr * (acc_dp[r] - del_s[ql]) : pr;
+        }
+
+        mbf16x8 f0, f1;
+        cvals_to_frags(cv, hi, &f0, &f1);
+
+        // dV[k,d] += P^T . dO   |   dK[k,d] += dS^T . Q
+        #pragma unroll
+        for (int dt = 0; dt < 4; ++dt) {
+          acc[dt] = mfma32(f0, vimg_frag(tr_img, dt, 2 * sb), acc[dt]);
+          acc[dt] =
+              mfma32(f1, vimg_frag(tr_img, dt, 2 * sb + 1), acc[dt]);
         }
       }
+    }
 
-      const bool mask_tile = causal || (q0 + QBLK > S) || !krow_valid;
-      float cv[16];
-      #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int qg = q0 + c_row(r, hi);
-        float s = acc_s[r] * scale;
-        if (mask_tile) {
-          if (qg >= S || !krow_valid || (causal && k_row > qg)) s = -3.0e38f;
-        }
-        const float pr = __expf(s - lse_s[qg - q0]);
-        cv[r] = DK ? scale * pr * (acc_dp[r] - del_s[qg - q0]) : pr;
+    if (has_next) {
+      // all staging grouped here (glds included): interleaving the glds
+      // address math with the MFMA region spilled ~50 dwords/lane
+      stage_glds(idx + 1, qnxt);
+      load_tr_regs(idx + 1, trr);
+      if (DK) {
+        ushort8 dor[4];
+        load_nat_regs(idx + 1, dor);
+        asm volatile("s_waitcnt vmcnt(0)");
+        __syncthreads();   // everyone done with tr_img/do_img/lse_s
+        write_nat_sw(dor, do_img);
+        write_v_tr(trr, tr_img);
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)");
+        __syncthreads();
+        write_v_tr(trr, tr_img);
       }
-
-      mbf16x8 f0, f1;
-      cvals_to_frags(cv, hi, &f0, &f1);
-
-      // dV[k,d] += P^T . dO   |   dK[k,d] += dS^T . Q
-      #pragma unroll
-      for (int dt = 0; dt < 4; ++dt) {
-        acc[dt] =
-            mfma32(f0, lds_frag(tr_tile + dt * 32 * TR_STRIDE, TR_STRIDE, 0),
-                   acc[dt]);
-        acc[dt] =
-            mfma32(f1, lds_frag(tr_tile + dt * 32 * TR_STRIDE, TR_STRIDE, 16),
-                   acc[dt]);
-      }
+      stage_lse(idx + 1);
+      __syncthreads();     // staged images visible
+      char* tq = qcur; qcur = qnxt; qnxt = tq;
     }
   }
 
