@@ -188,3 +188,44 @@ def test_adamw(dev):
     assert rel_err(m, mr) < 1e-4
     assert rel_err(v, vr) < 1e-4
     assert rel_err(p16, p16r) < 1e-2
+
+
+def test_fused_attention_qkv(dev):
+    """Fused rope+attention off the packed qkv buffer vs the unfused ops."""
+    torch.manual_seed(3)
+    B, S, Hq, Hkv, D = 2, 256, 8, 4, 128
+    qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D, device=dev,
+                      dtype=torch.bfloat16, requires_grad=True)
+    cos, sin = ops.rope_tables(S, D, device=dev)
+    out = ops.fused_attention_qkv(qkv, cos, sin, Hq, Hkv)
+    g = torch.randn_like(out)
+    out.backward(g)
+    dqkv = qkv.grad.clone()
+
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    q, k, v = qkv2.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    qr = ops.rope(q.reshape(B, S, Hq, D), cos, sin)
+    kr = ops.rope(k.reshape(B, S, Hkv, D), cos, sin)
+    ref = ops.flash_attention(qr, kr, v.reshape(B, S, Hkv, D).contiguous(),
+                              causal=True)
+    ref.backward(g)
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(dqkv.float(), qkv2.grad.float(),
+                          atol=5e-2, rtol=5e-2)
+
+
+def test_swiglu_packed(dev):
+    torch.manual_seed(4)
+    gu = torch.randn(64, 512, device=dev, dtype=torch.bfloat16,
+                     requires_grad=True)
+    out = ops.swiglu_packed(gu)
+    g = torch.randn_like(out)
+    out.backward(g)
+    dgu = gu.grad.clone()
+
+    gu2 = gu.detach().clone().requires_grad_(True)
+    a, b = gu2.chunk(2, dim=-1)
+    ref = ops.swiglu(a.contiguous(), b.contiguous())
+    ref.backward(g)
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2)
+    assert torch.allclose(dgu.float(), gu2.grad.float(), atol=2e-2, rtol=2e-2)

@@ -90,22 +90,17 @@ class LlamaBlock(nn.Module):
                 sin: torch.Tensor) -> torch.Tensor:
         cfg = self.cfg
         B, S, H = x.shape
-        # attention
+        # attention: rope + flash straight off the packed qkv GEMM output
+        # (no split/cat/contiguous traffic — ops.fused_attention_qkv)
         xn = ops.rmsnorm(x, self.attn_norm, cfg.rms_eps)
         qkv = self.wqkv(xn)
-        q, k, v = qkv.split([cfg.q_dim, cfg.kv_dim, cfg.kv_dim], dim=-1)
-        q = q.view(B, S, cfg.num_heads, cfg.head_dim)
-        k = k.view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        v = v.view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        q = ops.rope(q, cos, sin)
-        k = ops.rope(k, cos, sin)
-        attn = ops.flash_attention(q, k, v, causal=True)
+        attn = ops.fused_attention_qkv(
+            qkv, cos, sin, cfg.num_heads, cfg.num_kv_heads, causal=True
+        )
         x = x + self.wo(attn.reshape(B, S, cfg.q_dim))
-        # mlp
+        # mlp: swiglu over the packed gate|up buffer
         xn = ops.rmsnorm(x, self.mlp_norm, cfg.rms_eps)
-        gu = self.wgu(xn)
-        g, u = gu.chunk(2, dim=-1)
-        x = x + self.wdown(ops.swiglu(g.contiguous(), u.contiguous()))
+        x = x + self.wdown(ops.swiglu_packed(self.wgu(xn)))
         return x
 
 

@@ -73,8 +73,7 @@ class Expert(nn.Module):
                                bias=False, dtype=torch.bfloat16)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        g, u = self.wgu(x).chunk(2, dim=-1)
-        return self.wdown(ops.swiglu(g.contiguous(), u.contiguous()))
+        return self.wdown(ops.swiglu_packed(self.wgu(x)))
 
 
 class MoELayer(nn.Module):
@@ -208,11 +207,9 @@ class MixtralBlock(nn.Module):
         B, S, H = x.shape
         xn = ops.rmsnorm(x, self.attn_norm, cfg.rms_eps)
         qkv = self.wqkv(xn)
-        q, kk, v = qkv.split([cfg.q_dim, cfg.kv_dim, cfg.kv_dim], dim=-1)
-        q = ops.rope(q.view(B, S, cfg.num_heads, cfg.head_dim), cos, sin)
-        kk = ops.rope(kk.view(B, S, cfg.num_kv_heads, cfg.head_dim), cos, sin)
-        v = v.view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        attn = ops.flash_attention(q, kk, v, causal=True)
+        attn = ops.fused_attention_qkv(
+            qkv, cos, sin, cfg.num_heads, cfg.num_kv_heads, causal=True
+        )
         x = x + self.wo(attn.reshape(B, S, cfg.q_dim))
         xn = ops.rmsnorm(x, self.mlp_norm, cfg.rms_eps)
         return x + self.moe(xn)

@@ -245,3 +245,75 @@ def adamw_step(
 
 def mfma_probe(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
     return hip_ops().mfma_probe(a, b)
+
+
+# ---------------------------------------------------------------------------
+# Fused qkv attention: rope(q,k) + flash attention straight off the packed
+# qkv GEMM output — no split/contiguous copies forward, no torch.cat
+# backward (the grads dq/dk/dv are written strided into one dqkv buffer and
+# inverse-roped in place).
+# ---------------------------------------------------------------------------
+
+
+class _FusedAttentionQKV(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkv: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                Hq: int, Hkv: int, causal: bool, scale: float):
+        hip = hip_ops()
+        roped = hip.rope_qkv(qkv.contiguous(), cos, sin, Hq, Hkv, 1.0, False)
+        o, lse = hip.attn_fwd_qkv(roped, Hq, Hkv, scale, causal)
+        ctx.save_for_backward(roped, o, lse, cos, sin)
+        ctx.dims = (Hq, Hkv, causal, scale)
+        return o
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        roped, o, lse, cos, sin = ctx.saved_tensors
+        Hq, Hkv, causal, scale = ctx.dims
+        hip = hip_ops()
+        dqkv = hip.attn_bwd_qkv(roped, o, dout.contiguous(), lse, Hq, Hkv,
+                                scale, causal)
+        # inverse rotation on the dq/dk regions, in place on our own buffer
+        hip.rope_qkv(dqkv, cos, sin, Hq, Hkv, -1.0, True)
+        return dqkv, None, None, None, None, None, None
+
+
+def fused_attention_qkv(
+    qkv: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+    Hq: int, Hkv: int, causal: bool = True,
+    scale: Optional[float] = None,
+) -> torch.Tensor:
+    """qkv [B, S, (Hq+2*Hkv)*128] bf16 -> o [B, S, Hq, 128]."""
+    D = qkv.shape[-1] // (Hq + 2 * Hkv)
+    if scale is None:
+        scale = 1.0 / math.sqrt(D)
+    if _on_gpu(qkv):
+        return _FusedAttentionQKV.apply(qkv, cos, sin, Hq, Hkv, causal, scale)
+    # CPU reference path: split + rope + attention
+    B, S, _ = qkv.shape
+    q, k, v = qkv.split([Hq * D, Hkv * D, Hkv * D], dim=-1)
+    q = reference.rope(q.reshape(B, S, Hq, D), cos, sin)
+    k = reference.rope(k.reshape(B, S, Hkv, D), cos, sin)
+    return reference.attention(q, k, v.reshape(B, S, Hkv, D),
+                               causal=causal, scale=scale)
+
+
+class _SwiGLUPacked(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu: torch.Tensor):
+        gu = gu.contiguous()
+        ctx.save_for_backward(gu)
+        return hip_ops().swiglu_gu_fwd(gu)
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        (gu,) = ctx.saved_tensors
+        return hip_ops().swiglu_gu_bwd(dout.contiguous(), gu)
+
+
+def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
+    """gu [..., 2I] (gate | up) -> silu(gate) * up, [..., I]."""
+    if _on_gpu(gu):
+        return _SwiGLUPacked.apply(gu)
+    g, u = gu.chunk(2, dim=-1)
+    return reference.swiglu(g, u)

@@ -267,7 +267,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
                 const unsigned short* __restrict__ v,
                 unsigned short* __restrict__ o,
                 float* __restrict__ lse,  // [B,Hq,S]
-                int B, int S, int Hq, int Hkv, float scale, int causal) {
+                int B, int S, int Hq, int Hkv, float scale, int causal,
+                long q_rs, long kv_rs) {  // per-seq-row element strides
   // 3-deep K ring (glds prefetch 2 tiles ahead, counted vmcnt keeps the
   // newest tile's loads in flight ACROSS the barrier — T3/T4) + 2-deep V
   __shared__ __align__(16) char smem[3 * KIMG_BYTES + 2 * VIMG_BYTES];
@@ -283,8 +284,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
   const bool hi = lane >= 32;
   const int qr = lane & 31;
 
-  const long q_seq_stride = (long)Hq * HD;
-  const long kv_seq_stride = (long)Hkv * HD;
+  const long q_seq_stride = q_rs;
+  const long kv_seq_stride = kv_rs;
   const unsigned short* qb = q + (long)b * S * q_seq_stride + (long)hq * HD;
   const unsigned short* kb = k + (long)b * S * kv_seq_stride + (long)hkv * HD;
   const unsigned short* vb = v + (long)b * S * kv_seq_stride + (long)hkv * HD;
@@ -437,7 +438,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
   if (!wave_active || q_row >= S) return;
 
   const float inv_l = (l_run > 0.f) ? 1.0f / l_run : 0.f;
-  unsigned short* ob = o + ((long)b * S + q_row) * q_seq_stride + (long)hq * HD;
+  unsigned short* ob =
+      o + (((long)b * S + q_row) * Hq + hq) * HD;  // o is dense BSHD
   #pragma unroll
   for (int dt = 0; dt < 4; ++dt) {
     #pragma unroll
@@ -495,7 +497,8 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
                    const float* __restrict__ lse,
                    const float* __restrict__ delta,
                    unsigned short* __restrict__ dq,
-                   int B, int S, int Hq, int Hkv, float scale, int causal) {
+                   int B, int S, int Hq, int Hkv, float scale, int causal,
+                   long q_rs, long kv_rs, long dq_rs) {
   // v2: KV tiles of 64; K and V natural images arrive by async
   // global_load_lds (2-deep rings, swizzled); K^T single rot-placed image
   // reg-staged with the T14 split; exp2-domain probabilities.  80 KiB LDS
@@ -518,12 +521,13 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   const bool hi = lane >= 32;
   const int qr = lane & 31;
 
-  const long q_seq_stride = (long)Hq * HD;
-  const long kv_seq_stride = (long)Hkv * HD;
+  const long q_seq_stride = q_rs;
+  const long kv_seq_stride = kv_rs;
   const unsigned short* qb = q + (long)b * S * q_seq_stride + (long)hq * HD;
   const unsigned short* kb = k + (long)b * S * kv_seq_stride + (long)hkv * HD;
   const unsigned short* vb = v + (long)b * S * kv_seq_stride + (long)hkv * HD;
-  const unsigned short* dob = dout + (long)b * S * q_seq_stride + (long)hq * HD;
+  const unsigned short* dob =
+      dout + ((long)b * S) * ((long)Hq * HD) + (long)hq * HD;  // dense
 
   const int q0_blk = qt * BLOCK_Q;
   const int qw0 = q0_blk + wave * QBLK;
@@ -535,7 +539,7 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   float my_lse2 = 0.f, my_delta = 0.f;
   const long qrow_safe = row_valid ? q_row : (S - 1);
   const unsigned short* dop =
-      dob + qrow_safe * q_seq_stride + (hi ? 8 : 0);
+      dob + qrow_safe * ((long)Hq * HD) + (hi ? 8 : 0);
   {
     const unsigned short* qp = qb + qrow_safe * q_seq_stride + (hi ? 8 : 0);
     #pragma unroll
@@ -638,7 +642,7 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
 
   if (!wave_active || !row_valid) return;
   unsigned short* dqb =
-      dq + ((long)b * S + q_row) * q_seq_stride + (long)hq * HD;
+      dq + ((long)b * S + q_row) * dq_rs + (long)hq * HD;
   #pragma unroll
   for (int dt = 0; dt < 4; ++dt) {
     #pragma unroll
@@ -683,7 +687,8 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     const float* __restrict__ lse,
                     const float* __restrict__ delta,
                     unsigned short* __restrict__ out,  // dk or dv
-                    int B, int S, int Hq, int Hkv, float scale, int causal) {
+                    int B, int S, int Hq, int Hkv, float scale, int causal,
+                    long q_rs, long kv_rs, long dout_rs, long out_rs) {
   // v2: q tiles of 64; Q arrives by async global_load_lds (2-deep
   // swizzled ring); the transposed tile (dO^T for dV / Q^T for dK) and the
   // dK pass's natural dO are reg-staged after the barrier (T14).  Split
@@ -723,8 +728,8 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   const bool hi = lane >= 32;
   const int kr = lane & 31;
 
-  const long q_seq_stride = (long)Hq * HD;
-  const long kv_seq_stride = (long)Hkv * HD;
+  const long q_seq_stride = q_rs;
+  const long kv_seq_stride = kv_rs;
   const unsigned short* kb = k + (long)b * S * kv_seq_stride + (long)hkv * HD;
   const unsigned short* vb = v + (long)b * S * kv_seq_stride + (long)hkv * HD;
 
@@ -734,17 +739,21 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   const bool wave_active = kw0 < S;
   const bool krow_valid = k_row < S;
 
-  // K fragments resident in registers; the dK pass re-reads its V
-  // fragments per tile from L2 (keeping them resident costs 32 VGPR at
-  // the staging-phase pressure peak and spills)
-  mbf16x8 kfrag[8];
+  // K (and V for the dK pass) fragments resident in registers; the
+  // resulting spill is confined to the kernel prologue/epilogue (a
+  // per-tile scattered V re-read costs far more than the cold spill)
+  mbf16x8 kfrag[8], vfrag[8];
   const long krow_safe = krow_valid ? k_row : (S - 1);
-  const unsigned short* vp = vb + krow_safe * kv_seq_stride + (hi ? 8 : 0);
   {
     const unsigned short* kp = kb + krow_safe * kv_seq_stride + (hi ? 8 : 0);
+    const unsigned short* vp = vb + krow_safe * kv_seq_stride + (hi ? 8 : 0);
     #pragma unroll
     for (int c = 0; c < 8; ++c) {
       kfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(kp + c * 16));
+      if (DK) {
+        vfrag[c] =
+            __builtin_bit_cast(mbf16x8, *(const ushort8*)(vp + c * 16));
+      }
     }
   }
 
@@ -770,17 +779,18 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
     const int t = t0 + idx % nt_eff;
     const int hq = hkv * G + gh;
     const unsigned short* src = DK ? q : dout;
+    const long rs = DK ? q_seq_stride : dout_rs;
     const unsigned short* sb_ =
-        src + (long)b * S * q_seq_stride + (long)hq * HD;
-    load_v_regs(sb_, (long)t * FKV, q_seq_stride, S, vr);
+        src + (long)b * S * rs + (long)hq * HD;
+    load_v_regs(sb_, (long)t * FKV, rs, S, vr);
   };
   auto load_nat_regs = [&](int idx, ushort8 vr[4]) {  // dK only: dO
     const int gh = idx / nt_eff;
     const int t = t0 + idx % nt_eff;
     const int hq = hkv * G + gh;
     const unsigned short* sb_ =
-        dout + (long)b * S * q_seq_stride + (long)hq * HD;
-    load_v_regs(sb_, (long)t * FKV, q_seq_stride, S, vr);
+        dout + (long)b * S * dout_rs + (long)hq * HD;
+    load_v_regs(sb_, (long)t * FKV, dout_rs, S, vr);
   };
   auto stage_lse = [&](int idx) {
     const int gh = idx / nt_eff;
@@ -836,9 +846,7 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
         if (DK) {
           #pragma unroll
           for (int c = 0; c < 8; ++c) {
-            const mbf16x8 vf =
-                __builtin_bit_cast(mbf16x8, *(const ushort8*)(vp + c * 16));
-            acc_dp = mfma32(kimg_frag(do_img, sb, c), vf, acc_dp);
+            acc_dp = mfma32(kimg_frag(do_img, sb, c), vfrag[c], acc_dp);
           }
         }
         const int qs0 = q0 + 32 * sb;
@@ -917,7 +925,7 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
           for (int j = 0; j < 4; ++j) {
             w[j] = (short)f32_to_bf16(tr[kk * 36 + d0 + j]);
           }
-          *(bf16x4_raw*)(out + ((long)b * S + kw0 + kk) * kv_seq_stride +
+          *(bf16x4_raw*)(out + ((long)b * S + kw0 + kk) * out_rs +
                          (long)hkv * HD + dt * 32 + d0) = w;
         }
       }
@@ -934,13 +942,13 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
 extern "C" void attn_fwd_launch(const void* q, const void* k, const void* v,
                                 void* o, void* lse, int B, int S, int Hq,
                                 int Hkv, float scale, int causal,
-                                hipStream_t stream) {
+                                long q_rs, long kv_rs, hipStream_t stream) {
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
   hipLaunchKernelGGL(attn_fwd_kernel, dim3(B * Hq * nqt), dim3(256), 0,
                      stream, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v,
                      (unsigned short*)o, (float*)lse, B, S, Hq, Hkv, scale,
-                     causal);
+                     causal, q_rs, kv_rs);
 }
 
 extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
@@ -948,7 +956,8 @@ extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
                                 const void* lse, void* delta, void* dq,
                                 void* dk, void* dv, int B, int S, int Hq,
                                 int Hkv, float scale, int causal,
-                                hipStream_t stream) {
+                                long q_rs, long kv_rs, long dqkv_q_rs,
+                                long dqkv_kv_rs, hipStream_t stream) {
   const long rows = (long)B * S * Hq;
   hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3((int)((rows + 3) / 4)),
                      dim3(256), 0, stream, (const unsigned short*)dout,
@@ -959,18 +968,19 @@ extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
                      (const unsigned short*)k, (const unsigned short*)v,
                      (const unsigned short*)dout, (const float*)lse,
                      (const float*)delta, (unsigned short*)dq, B, S, Hq, Hkv,
-                     scale, causal);
+                     scale, causal, q_rs, kv_rs, dqkv_q_rs);
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
+  const long dout_rs = (long)Hq * HD;
   hipLaunchKernelGGL((attn_bwd_dkv_kernel<false>), dim3(B * Hkv * nkt),
                      dim3(256), 0, stream, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v,
                      (const unsigned short*)dout, (const float*)lse,
                      (const float*)delta, (unsigned short*)dv, B, S, Hq, Hkv,
-                     scale, causal);
+                     scale, causal, q_rs, kv_rs, dout_rs, dqkv_kv_rs);
   hipLaunchKernelGGL((attn_bwd_dkv_kernel<true>), dim3(B * Hkv * nkt),
                      dim3(256), 0, stream, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v,
                      (const unsigned short*)dout, (const float*)lse,
                      (const float*)delta, (unsigned short*)dk, B, S, Hq, Hkv,
-                     scale, causal);
+                     scale, causal, q_rs, kv_rs, dout_rs, dqkv_kv_rs);
 }

@@ -22,10 +22,16 @@ void ce_fwd_launch(const void*, const void*, void*, void*, long, int,
 void ce_bwd_launch(const void*, const void*, const void*, const void*, void*,
                    long, int, hipStream_t);
 void attn_fwd_launch(const void*, const void*, const void*, void*, void*, int,
-                     int, int, int, float, int, hipStream_t);
+                     int, int, int, float, int, long, long, hipStream_t);
 void attn_bwd_launch(const void*, const void*, const void*, const void*,
                      const void*, const void*, void*, void*, void*, void*,
-                     int, int, int, int, float, int, hipStream_t);
+                     int, int, int, int, float, int, long, long, long, long,
+                     hipStream_t);
+void rope_qkv_launch(const void*, void*, const void*, const void*, long, int,
+                     long, int, float, hipStream_t);
+void swiglu_gu_fwd_launch(const void*, void*, long, long, hipStream_t);
+void swiglu_gu_bwd_launch(const void*, const void*, void*, long, long,
+                          hipStream_t);
 void mfma_probe_launch(const void*, const void*, void*, hipStream_t);
 }
 
@@ -171,7 +177,7 @@ std::vector<at::Tensor> attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v,
   auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
   attn_fwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                   lse.data_ptr(), B, S, Hq, Hkv, (float)scale, causal ? 1 : 0,
-                  cur_stream());
+                  (long)Hq * 128, (long)Hkv * 128, cur_stream());
   return {o, lse};
 }
 
@@ -188,8 +194,94 @@ std::vector<at::Tensor> attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v,
   attn_bwd_launch(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
                   dout.data_ptr(), lse.data_ptr(), delta.data_ptr(),
                   dq.data_ptr(), dk.data_ptr(), dv.data_ptr(), B, S, Hq, Hkv,
-                  (float)scale, causal ? 1 : 0, cur_stream());
+                  (float)scale, causal ? 1 : 0, (long)Hq * 128,
+                  (long)Hkv * 128, (long)Hq * 128, (long)Hkv * 128,
+                  cur_stream());
   return {dq, dk, dv};
+}
+
+// ---- fused qkv attention (strided views over the packed qkv buffer) ------
+// qkv: [B, S, (Hq+2*Hkv)*128] bf16 from the fused qkv GEMM.  q starts at 0,
+// k at Hq*128, v at (Hq+Hkv)*128; row stride = (Hq+2*Hkv)*128.  Saves the
+// split/contiguous copies forward and the torch.cat backward.
+std::vector<at::Tensor> attn_fwd_qkv(at::Tensor qkv, long Hq, long Hkv,
+                                     double scale, bool causal) {
+  check_bf16(qkv, "qkv");
+  const int B = qkv.size(0), S = qkv.size(1);
+  const long rs = (Hq + 2 * Hkv) * 128;
+  TORCH_CHECK(qkv.size(2) == rs, "qkv last dim mismatch");
+  auto o = at::empty({B, S, Hq, 128}, qkv.options());
+  auto lse = at::empty({B, Hq, S}, qkv.options().dtype(at::kFloat));
+  const unsigned short* base = (const unsigned short*)qkv.data_ptr();
+  attn_fwd_launch(base, base + Hq * 128, base + (Hq + Hkv) * 128,
+                  o.data_ptr(), lse.data_ptr(), B, S, (int)Hq, (int)Hkv,
+                  (float)scale, causal ? 1 : 0, rs, rs, cur_stream());
+  return {o, lse};
+}
+
+at::Tensor attn_bwd_qkv(at::Tensor qkv, at::Tensor o, at::Tensor dout,
+                        at::Tensor lse, long Hq, long Hkv, double scale,
+                        bool causal) {
+  check_bf16(dout, "dout");
+  const int B = qkv.size(0), S = qkv.size(1);
+  const long rs = (Hq + 2 * Hkv) * 128;
+  auto dqkv = at::empty_like(qkv);
+  auto delta = at::empty({B, Hq, S}, qkv.options().dtype(at::kFloat));
+  const unsigned short* base = (const unsigned short*)qkv.data_ptr();
+  unsigned short* dbase = (unsigned short*)dqkv.data_ptr();
+  attn_bwd_launch(base, base + Hq * 128, base + (Hq + Hkv) * 128,
+                  o.data_ptr(), dout.data_ptr(), lse.data_ptr(),
+                  delta.data_ptr(), dbase, dbase + Hq * 128,
+                  dbase + (Hq + Hkv) * 128, B, S, (int)Hq, (int)Hkv,
+                  (float)scale, causal ? 1 : 0, rs, rs, rs, rs,
+                  cur_stream());
+  return dqkv;
+}
+
+// RoPE over the packed qkv buffer's q+k heads (y may alias x for in-place
+// gradient rotation; the v region is copied when y != x).
+at::Tensor rope_qkv(at::Tensor x, at::Tensor cos_t, at::Tensor sin_t,
+                    long Hq, long Hkv, double sign, bool inplace) {
+  check_bf16(x, "x");
+  check_f32(cos_t, "cos");
+  check_f32(sin_t, "sin");
+  const int B = x.size(0), S = x.size(1);
+  const long rs = (Hq + 2 * Hkv) * 128;
+  TORCH_CHECK(x.size(2) == rs, "qkv last dim mismatch");
+  at::Tensor y = inplace ? x : at::empty_like(x);
+  rope_qkv_launch(x.data_ptr(), y.data_ptr(), cos_t.data_ptr(),
+                  sin_t.data_ptr(), (long)B * S, (int)(Hq + Hkv), rs, S,
+                  (float)sign, cur_stream());
+  if (!inplace) {
+    const long v0 = (Hq + Hkv) * 128;
+    y.narrow(2, v0, Hkv * 128).copy_(x.narrow(2, v0, Hkv * 128));
+  }
+  return y;
+}
+
+// ---- packed swiglu --------------------------------------------------------
+at::Tensor swiglu_gu_fwd(at::Tensor gu) {
+  check_bf16(gu, "gu");
+  const long twoI = gu.size(-1);
+  TORCH_CHECK(twoI % 16 == 0, "gu last dim must be 2I, I % 8 == 0");
+  const long I = twoI / 2;
+  const long rows = gu.numel() / twoI;
+  auto sizes = gu.sizes().vec();
+  sizes.back() = I;
+  auto out = at::empty(sizes, gu.options());
+  swiglu_gu_fwd_launch(gu.data_ptr(), out.data_ptr(), rows, I, cur_stream());
+  return out;
+}
+
+at::Tensor swiglu_gu_bwd(at::Tensor dout, at::Tensor gu) {
+  check_bf16(dout, "dout");
+  const long twoI = gu.size(-1);
+  const long I = twoI / 2;
+  const long rows = gu.numel() / twoI;
+  auto dgu = at::empty_like(gu);
+  swiglu_gu_bwd_launch(dout.data_ptr(), gu.data_ptr(), dgu.data_ptr(), rows,
+                       I, cur_stream());
+  return dgu;
 }
 
 // ---- probe ----------------------------------------------------------------
@@ -214,5 +306,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
+  m.def("attn_fwd_qkv", &attn_fwd_qkv);
+  m.def("attn_bwd_qkv", &attn_bwd_qkv);
+  m.def("rope_qkv", &rope_qkv);
+  m.def("swiglu_gu_fwd", &swiglu_gu_fwd);
+  m.def("swiglu_gu_bwd", &swiglu_gu_bwd);
   m.def("mfma_probe", &mfma_probe);
 }

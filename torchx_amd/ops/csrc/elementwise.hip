@@ -147,3 +147,141 @@ extern "C" void swiglu_bwd_launch(const void* dout, const void* g,
                      (const unsigned short*)g, (const unsigned short*)u,
                      (unsigned short*)dg, (unsigned short*)du, n8);
 }
+
+// ---------------------------------------------------------------------------
+// Strided in-place-capable RoPE over the fused qkv buffer: rows = B*S, each
+// row is [q: Hq*128 | k: Hkv*128 | v: Hkv*128] at row_stride elements.
+// Rotates the first (Hq+Hkv) heads; the v region is untouched (pass y == x
+// for in-place, e.g. on the dqkv gradient buffer).
+// ---------------------------------------------------------------------------
+__global__ void rope_qkv_kernel(
+    const unsigned short* __restrict__ x,
+    unsigned short* __restrict__ y,
+    const float* __restrict__ cos_t,   // [S, 64]
+    const float* __restrict__ sin_t,
+    long np4,        // rows * nh * 16  (4-pair groups, half_d = 64)
+    int nh,          // Hq + Hkv
+    long row_stride, // elements per row
+    int seq_len,
+    float sign) {
+  typedef unsigned short us4 __attribute__((ext_vector_type(4)));
+  typedef float f4 __attribute__((ext_vector_type(4)));
+  const int half_d = 64;
+  for (long p4 = grid_stride_begin(); p4 < np4; p4 += grid_stride()) {
+    const long hp = p4 >> 4;           // (row, head)
+    const int i = (int)(p4 & 15) * 4;  // pair index within head
+    const long row = hp / nh;
+    const int h = (int)(hp % nh);
+    const int pos = (int)(row % seq_len);
+    const long base = row * row_stride + (long)h * 128;
+    us4 u1 = *(const us4*)(x + base + i);
+    us4 u2 = *(const us4*)(x + base + half_d + i);
+    f4 c = *(const f4*)(cos_t + (long)pos * half_d + i);
+    f4 s = *(const f4*)(sin_t + (long)pos * half_d + i);
+    us4 o1, o2;
+    #pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float x1 = bf16_to_f32(u1[j]);
+      float x2 = bf16_to_f32(u2[j]);
+      o1[j] = f32_to_bf16(fmaf(x1, c[j], -sign * s[j] * x2));
+      o2[j] = f32_to_bf16(fmaf(x2, c[j], sign * s[j] * x1));
+    }
+    *(us4*)(y + base + i) = o1;
+    *(us4*)(y + base + half_d + i) = o2;
+  }
+}
+
+extern "C" void rope_qkv_launch(
+    const void* x, void* y, const void* cos_t, const void* sin_t,
+    long rows, int nh, long row_stride, int seq_len, float sign,
+    hipStream_t stream) {
+  const long np4 = rows * nh * 16;
+  const int block = 256;
+  long grid = (np4 + block - 1) / block;
+  if (grid > 65535L * 8) grid = 65535L * 8;
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL(rope_qkv_kernel, dim3((int)grid), dim3(block), 0,
+                     stream, (const unsigned short*)x, (unsigned short*)y,
+                     (const float*)cos_t, (const float*)sin_t, np4, nh,
+                     row_stride, seq_len, sign);
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU over the packed gate-up buffer gu = [rows, 2I]: out = silu(g)*u
+// with g = gu[:, :I], u = gu[:, I:].  Avoids the two .contiguous() copies
+// of the chunked views and the torch.cat in backward.
+// ---------------------------------------------------------------------------
+__global__ void swiglu_gu_fwd_kernel(
+    const unsigned short* __restrict__ gu,
+    unsigned short* __restrict__ out,
+    long n8,     // rows * I / 8
+    long i8) {   // I / 8
+  for (long p = grid_stride_begin(); p < n8; p += grid_stride()) {
+    const long row = p / i8, c8 = p % i8;
+    const long base = row * 2 * i8 * 8 + c8 * 8;
+    ushort8 gv = *(const ushort8*)(gu + base);
+    ushort8 uv = *(const ushort8*)(gu + base + i8 * 8);
+    ushort8 ov;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float gf = bf16_to_f32(gv[j]);
+      float uf = bf16_to_f32(uv[j]);
+      float sig = 1.0f / (1.0f + __expf(-gf));
+      ov[j] = f32_to_bf16(gf * sig * uf);
+    }
+    *(ushort8*)(out + p * 8) = ov;
+  }
+}
+
+__global__ void swiglu_gu_bwd_kernel(
+    const unsigned short* __restrict__ dout,
+    const unsigned short* __restrict__ gu,
+    unsigned short* __restrict__ dgu,
+    long n8, long i8) {
+  for (long p = grid_stride_begin(); p < n8; p += grid_stride()) {
+    const long row = p / i8, c8 = p % i8;
+    const long base = row * 2 * i8 * 8 + c8 * 8;
+    ushort8 dov = *(const ushort8*)(dout + p * 8);
+    ushort8 gv = *(const ushort8*)(gu + base);
+    ushort8 uv = *(const ushort8*)(gu + base + i8 * 8);
+    ushort8 dgv, duv;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float dof = bf16_to_f32(dov[j]);
+      float gf = bf16_to_f32(gv[j]);
+      float uf = bf16_to_f32(uv[j]);
+      float sig = 1.0f / (1.0f + __expf(-gf));
+      float silu = gf * sig;
+      dgv[j] = f32_to_bf16(dof * uf * sig * (1.0f + gf * (1.0f - sig)));
+      duv[j] = f32_to_bf16(dof * silu);
+    }
+    *(ushort8*)(dgu + base) = dgv;
+    *(ushort8*)(dgu + base + i8 * 8) = duv;
+  }
+}
+
+extern "C" void swiglu_gu_fwd_launch(const void* gu, void* out, long rows,
+                                     long I, hipStream_t stream) {
+  const long n8 = rows * I / 8;
+  const int block = 256;
+  long grid = (n8 + block - 1) / block;
+  if (grid > 65535L * 8) grid = 65535L * 8;
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL(swiglu_gu_fwd_kernel, dim3((int)grid), dim3(block), 0,
+                     stream, (const unsigned short*)gu, (unsigned short*)out,
+                     n8, I / 8);
+}
+
+extern "C" void swiglu_gu_bwd_launch(const void* dout, const void* gu,
+                                     void* dgu, long rows, long I,
+                                     hipStream_t stream) {
+  const long n8 = rows * I / 8;
+  const int block = 256;
+  long grid = (n8 + block - 1) / block;
+  if (grid > 65535L * 8) grid = 65535L * 8;
+  if (grid == 0) grid = 1;
+  hipLaunchKernelGGL(swiglu_gu_bwd_kernel, dim3((int)grid), dim3(block), 0,
+                     stream, (const unsigned short*)dout,
+                     (const unsigned short*)gu, (unsigned short*)dgu, n8,
+                     I / 8);
+}
