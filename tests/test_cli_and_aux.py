@@ -182,3 +182,19 @@ def test_devices_partition():
     # oversubscription -> no pinning
     assign2 = partition_devices({"t": 4}, {"t": 4}, total_gpus=8)
     assert assign2["t"] == [None] * 4
+
+
+def test_cli_run_stdin_json(tmp_path):
+    import json as _json
+    import subprocess
+    import sys as _sys
+
+    spec = {"component": "utils.echo",
+            "component_args": ["--msg", "stdin-mode"],
+            "scheduler": "local_cwd", "dryrun": True}
+    out = subprocess.run(
+        [_sys.executable, "-m", "torchx_amd.cli.main", "run", "--stdin"],
+        input=_json.dumps(spec), capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stderr
+    assert "stdin-mode" in out.stdout

@@ -54,6 +54,28 @@ def _component_defaults() -> Dict[str, Dict[str, str]]:
 
 
 def cmd_run(args: argparse.Namespace) -> int:
+    if getattr(args, "stdin", False):
+        # JSON run spec from stdin: {"component": ..., "component_args":
+        # [...], "scheduler": ..., "scheduler_args": {...}, "dryrun": bool}
+        # (reference parity: cli/cmd_run.py:366-399)
+        try:
+            spec = json.load(sys.stdin)
+        except (json.JSONDecodeError, EOFError):
+            print("invalid JSON on stdin for `torchx run`", file=sys.stderr)
+            return 1
+        if not isinstance(spec, dict) or "component" not in spec:
+            print("stdin JSON must be a dict with a `component` key",
+                  file=sys.stderr)
+            return 1
+        args.component_name_and_args = (
+            [spec["component"]] + [str(a) for a in
+                                   spec.get("component_args", [])]
+        )
+        args.scheduler = spec.get("scheduler", args.scheduler)
+        args.scheduler_args = ",".join(
+            f"{k}={v}" for k, v in spec.get("scheduler_args", {}).items()
+        )
+        args.dryrun = bool(spec.get("dryrun", args.dryrun))
     component, comp_args = _parse_run_args(args.component_name_and_args)
     scheduler = args.scheduler
     runner = get_runner(component_defaults=_component_defaults())
@@ -275,6 +297,8 @@ def create_parser() -> argparse.ArgumentParser:
     p.add_argument("-cfg", "--scheduler_args", type=str, default="",
                    help="scheduler runopts, e.g. k1=v1,k2=v2")
     p.add_argument("--dryrun", action="store_true")
+    p.add_argument("--stdin", action="store_true",
+                   help="read a JSON run spec from stdin")
     p.add_argument("--wait", action="store_true")
     p.add_argument("--log", action="store_true")
     p.add_argument("--workspace", type=str, default=None)
