@@ -192,6 +192,7 @@ def test_adamw(dev):
 
 def test_fused_attention_qkv(dev):
     """Fused rope+attention off the packed qkv buffer vs the unfused ops."""
+    ops = _hip()
     torch.manual_seed(3)
     B, S, Hq, Hkv, D = 2, 256, 8, 4, 128
     qkv = torch.randn(B, S, (Hq + 2 * Hkv) * D, device=dev,
@@ -215,6 +216,7 @@ def test_fused_attention_qkv(dev):
 
 
 def test_swiglu_packed(dev):
+    ops = _hip()
     torch.manual_seed(4)
     gu = torch.randn(64, 512, device=dev, dtype=torch.bfloat16,
                      requires_grad=True)
