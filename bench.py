@@ -43,6 +43,13 @@ def main() -> int:
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
         backend = "nccl"  # RCCL on ROCm
+        # tuned hipBLASLt algorithm table (tools/gemm_tune.py), read-only
+        tuned = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                             "torchx_amd", "ops", "tunableop_gfx950.csv")
+        if os.path.exists(tuned):
+            torch.cuda.tunable.enable(True)
+            torch.cuda.tunable.tuning_enable(False)
+            torch.cuda.tunable.read_file(tuned)
     else:
         device = torch.device("cpu")
         backend = "gloo"

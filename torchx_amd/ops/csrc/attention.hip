@@ -665,49 +665,94 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
 // wave/SIMD with zero latency hiding; the S^T recompute in the second pass
 // costs 8 extra MFMAs/tile but doubles occupancy).
 // ---------------------------------------------------------------------------
-// Natural swizzled write of a reg-staged [64][128] tile (same image form
-// as the glds K staging: byte ^= (row&15)<<4; vectorized b128, no
-// conflicts: 16 consecutive lanes write one row's 16 granules permuted).
-__device__ __forceinline__ void write_nat_sw(const ushort8 vr[4], char* img) {
-  const int tid = threadIdx.x;
+// 512-thread (8-wave) staging variants: same image formats as the 4-wave
+// helpers, work split across twice the threads.
+__device__ __forceinline__ void stage_k_glds8(
+    const unsigned short* __restrict__ kb, long kv0, long stride_elems,
+    int S, char* kimg) {
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
   #pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    const int c = tid + p * 256;
-    const int r = c >> 4, g = c & 15;
-    *(ushort8*)(img + r * 256 + ((g * 16) ^ ((r & 15) << 4))) = vr[p];
+  for (int j = 0; j < 2; ++j) {
+    const int i = wave * 2 + j;
+    const int row = i * 4 + (lane >> 4);
+    const int colbyte = ((lane & 15) * 16) ^ ((row & 15) << 4);
+    long srow = kv0 + row;
+    if (srow >= S) srow = S - 1;
+    const char* src = (const char*)(kb + srow * stride_elems) + colbyte;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) unsigned int*)src,
+        (__attribute__((address_space(3))) unsigned int*)(kimg + i * 1024),
+        16, 0, 0);
   }
 }
 
-template <bool DK>
-__global__ void __launch_bounds__(256, 2)
+__device__ __forceinline__ void load_v_regs8(
+    const unsigned short* __restrict__ vb, long kv0, long stride_elems,
+    int S, ushort8 vr[2]) {
+  const int tid = threadIdx.x;
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int c = tid + p * 512;
+    const int r = c >> 4, g = c & 15;
+    long srow = kv0 + r;
+    if (srow >= S) srow = S - 1;
+    vr[p] = *(const ushort8*)(vb + srow * stride_elems + g * 8);
+  }
+}
+
+__device__ __forceinline__ void write_v_tr8(const ushort8 vr[2], char* vimg) {
+  const int tid = threadIdx.x;
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int c = tid + p * 512;
+    const int r = c >> 4, g = c & 15;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d = 8 * g + j;
+      const int gr = ((r >> 3) + vrot(d)) & 7;
+      *(unsigned short*)(vimg + d * 128 + gr * 16 + (r & 7) * 2) = vr[p][j];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Backward dK+dV, one 8-wave kernel.  Waves w and w+4 own the SAME 32 kv
+// rows of a 128-row block: w accumulates dV, w+4 accumulates dK — so the
+// Q/dO streams are staged ONCE for both outputs (the two-pass version
+// streamed them 2-3x).  C-layouts are flipped vs the fwd kernel (swapped
+// operand order): each lane owns one q COLUMN, so lse/delta are direct
+// per-lane global loads (no LDS staging).  Four staged images per tile —
+// Q nat (S^T chain B-op), dO nat (dP^T chain B-op), Q^T (dK accumulate
+// B-op), dO^T (dV accumulate B-op) — double-buffered in 128 KiB LDS, one
+// barrier per tile.
+// ---------------------------------------------------------------------------
+__global__ void __launch_bounds__(512, 2)
 attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     const unsigned short* __restrict__ k,
                     const unsigned short* __restrict__ v,
                     const unsigned short* __restrict__ dout,
                     const float* __restrict__ lse,
                     const float* __restrict__ delta,
-                    unsigned short* __restrict__ out,  // dk or dv
+                    unsigned short* __restrict__ dk,
+                    unsigned short* __restrict__ dv,
                     int B, int S, int Hq, int Hkv, float scale, int causal,
                     long q_rs, long kv_rs, long dout_rs, long out_rs) {
-  // v2: q tiles of 64; Q arrives by async global_load_lds (2-deep
-  // swizzled ring); the transposed tile (dO^T for dV / Q^T for dK) and the
-  // dK pass's natural dO are reg-staged after the barrier (T14).  Split
-  // dV/dK passes keep 2 waves/SIMD.  LDS: 32K Q ring + 16K TR (+16K dO
-  // nat for dK) + lse/delta rows.
-  __shared__ __align__(16) char smem[2 * KIMG_BYTES + VIMG_BYTES +
-                                     (DK ? KIMG_BYTES : 0)];
+  __shared__ __align__(16) char smem[8 * KIMG_BYTES];
   __shared__ float lse_s[FKV], del_s[FKV];
-  char* qcur = smem;
-  char* qnxt = smem + KIMG_BYTES;
-  char* tr_img = smem + 2 * KIMG_BYTES;              // dO^T / Q^T [128][64]
-  char* do_img = smem + 2 * KIMG_BYTES + VIMG_BYTES; // dK only, [64][256B]
+  char* qn_c = smem;                       // Q natural cur
+  char* don_c = smem + KIMG_BYTES;         // dO natural cur
+  char* qt_c = smem + 2 * KIMG_BYTES;      // Q^T cur
+  char* dot_c = smem + 3 * KIMG_BYTES;     // dO^T cur
+  char* qn_n = smem + 4 * KIMG_BYTES;
+  char* don_n = smem + 5 * KIMG_BYTES;
+  char* qt_n = smem + 6 * KIMG_BYTES;
+  char* dot_n = smem + 7 * KIMG_BYTES;
 
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
   const int G = Hq / Hkv;
   int b, hkv, kt;
   {
-    // XCD-aware swizzle: the nkt blocks streaming the same (b, hkv)
-    // Q/dO group go to one XCD (same rationale as map_block_fwd)
     int bid = blockIdx.x;
     if (((B * Hkv) & 7) == 0) {
       const int xcd = bid & 7, slot = bid >> 3;
@@ -724,6 +769,7 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   }
 
   const int wave = threadIdx.x / 64;
+  const bool is_dk = wave >= 4;
   const int lane = threadIdx.x & 63;
   const bool hi = lane >= 32;
   const int kr = lane & 31;
@@ -734,23 +780,21 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   const unsigned short* vb = v + (long)b * S * kv_seq_stride + (long)hkv * HD;
 
   const int kv0_blk = kt * BLOCK_K;
-  const int kw0 = kv0_blk + wave * KBLK;       // wave's first kv row
-  const long k_row = kw0 + kr;                  // lane's kv row
+  const int kw0 = kv0_blk + (wave & 3) * KBLK;  // this wave's 32 kv rows
+  const long k_row = kw0 + kr;
   const bool wave_active = kw0 < S;
   const bool krow_valid = k_row < S;
 
-  // K (and V for the dK pass) fragments resident in registers; the
-  // resulting spill is confined to the kernel prologue/epilogue (a
-  // per-tile scattered V re-read costs far more than the cold spill)
+  // K resident in all waves; V resident in the dK waves
   mbf16x8 kfrag[8], vfrag[8];
-  const long krow_safe = krow_valid ? k_row : (S - 1);
   {
-    const unsigned short* kp = kb + krow_safe * kv_seq_stride + (hi ? 8 : 0);
-    const unsigned short* vp = vb + krow_safe * kv_seq_stride + (hi ? 8 : 0);
+    const long row = krow_valid ? k_row : (S - 1);
+    const unsigned short* kp = kb + row * kv_seq_stride + (hi ? 8 : 0);
+    const unsigned short* vp = vb + row * kv_seq_stride + (hi ? 8 : 0);
     #pragma unroll
     for (int c = 0; c < 8; ++c) {
       kfrag[c] = __builtin_bit_cast(mbf16x8, *(const ushort8*)(kp + c * 16));
-      if (DK) {
+      if (is_dk) {
         vfrag[c] =
             __builtin_bit_cast(mbf16x8, *(const ushort8*)(vp + c * 16));
       }
@@ -763,92 +807,73 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   const int t0 = causal ? (kv0_blk / FKV) : 0;
   const int nt = (S + FKV - 1) / FKV;
   const int nt_eff = nt - t0;
-  const int total = G * nt_eff;      // flattened (gh, t) iterations
+  const int total = G * nt_eff;
 
-  // stage helpers for iteration idx: q tile t of head gh
-  auto stage_glds = [&](int idx, char* dst) {
-    const int gh = idx / nt_eff;
-    const int t = t0 + idx % nt_eff;
-    const int hq = hkv * G + gh;
-    const unsigned short* qb =
-        q + (long)b * S * q_seq_stride + (long)hq * HD;
-    stage_k_glds(qb, (long)t * FKV, q_seq_stride, S, dst);
+  auto head_q = [&](int gh) {
+    return q + (long)b * S * q_seq_stride + (long)(hkv * G + gh) * HD;
   };
-  auto load_tr_regs = [&](int idx, ushort8 vr[4]) {
-    const int gh = idx / nt_eff;
-    const int t = t0 + idx % nt_eff;
-    const int hq = hkv * G + gh;
-    const unsigned short* src = DK ? q : dout;
-    const long rs = DK ? q_seq_stride : dout_rs;
-    const unsigned short* sb_ =
-        src + (long)b * S * rs + (long)hq * HD;
-    load_v_regs(sb_, (long)t * FKV, rs, S, vr);
+  auto head_do = [&](int gh) {
+    return dout + (long)b * S * dout_rs + (long)(hkv * G + gh) * HD;
   };
-  auto load_nat_regs = [&](int idx, ushort8 vr[4]) {  // dK only: dO
-    const int gh = idx / nt_eff;
-    const int t = t0 + idx % nt_eff;
-    const int hq = hkv * G + gh;
-    const unsigned short* sb_ =
-        dout + (long)b * S * dout_rs + (long)hq * HD;
-    load_v_regs(sb_, (long)t * FKV, dout_rs, S, vr);
-  };
-  auto stage_lse = [&](int idx) {
-    const int gh = idx / nt_eff;
-    const int t = t0 + idx % nt_eff;
-    const int hq = hkv * G + gh;
-    const int q0 = t * FKV;
-    const float* lse_b = lse + ((long)b * Hq + hq) * S;
+  auto stage_lse = [&](int gh, int t) {
+    const int hq_ = hkv * G + gh;
+    const int q0_ = t * FKV;
     if (threadIdx.x < FKV) {
-      const int qg = q0 + threadIdx.x;
+      const float* lse_b = lse + ((long)b * Hq + hq_) * S;
+      const int qg = q0_ + threadIdx.x;
       lse_s[threadIdx.x] = (qg < S) ? lse_b[qg] * LOG2E : 0.f;
-    } else if (DK && threadIdx.x < 2 * FKV) {
-      const float* del_b = delta + ((long)b * Hq + hq) * S;
-      const int qg = q0 + (threadIdx.x - FKV);
+    } else if (threadIdx.x < 2 * FKV) {
+      const float* del_b = delta + ((long)b * Hq + hq_) * S;
+      const int qg = q0_ + (threadIdx.x - FKV);
       del_s[threadIdx.x - FKV] = (qg < S) ? del_b[qg] : 0.f;
     }
   };
 
-  ushort8 trr[4];
-  // prologue: stage iteration 0
-  stage_glds(0, qcur);
-  load_tr_regs(0, trr);
-  if (DK) {
-    ushort8 dor[4];
-    load_nat_regs(0, dor);
-    asm volatile("s_waitcnt vmcnt(0)");
-    write_nat_sw(dor, do_img);
-  } else {
-    asm volatile("s_waitcnt vmcnt(0)");
-  }
-  write_v_tr(trr, tr_img);
-  stage_lse(0);
+  ushort8 trq[2], trdo[2];
+  // prologue: tile (gh=0, t=t0)
+  stage_k_glds8(head_q(0), (long)t0 * FKV, q_seq_stride, S, qn_c);
+  stage_k_glds8(head_do(0), (long)t0 * FKV, dout_rs, S, don_c);
+  load_v_regs8(head_q(0), (long)t0 * FKV, q_seq_stride, S, trq);
+  load_v_regs8(head_do(0), (long)t0 * FKV, dout_rs, S, trdo);
+  asm volatile("s_waitcnt vmcnt(0)");
+  write_v_tr8(trq, qt_c);
+  write_v_tr8(trdo, dot_c);
+  stage_lse(0, t0);
   __syncthreads();
 
   for (int idx = 0; idx < total; ++idx) {
     const int gh = idx / nt_eff;
     const int t = t0 + idx % nt_eff;
     const int q0 = t * FKV;
+    const int hq = hkv * G + gh;
     const bool has_next = (idx + 1) < total;
+    if (has_next) {
+      const int nidx = idx + 1;
+      const int ngh = nidx / nt_eff;
+      const long nrow0 = (long)(t0 + nidx % nt_eff) * FKV;
+      stage_k_glds8(head_q(ngh), nrow0, q_seq_stride, S, qn_n);
+      stage_k_glds8(head_do(ngh), nrow0, dout_rs, S, don_n);
+      load_v_regs8(head_q(ngh), nrow0, q_seq_stride, S, trq);
+      load_v_regs8(head_do(ngh), nrow0, dout_rs, S, trdo);
+    }
 
     const bool needed =
         wave_active && (!causal || q0 + FKV - 1 >= kw0);
     if (needed) {
       #pragma clang loop unroll(disable)
       for (int sb = 0; sb < 2; ++sb) {
-        // S[q-sub, k] (lane owns k col); dP for the dK pass.  The two
-        // MFMA chains run sequentially so only one B-operand set (kfrag
-        // or the per-chain V fragments) is live at a time.
+        // S[q, k own] (lane owns the k column; q rows via c_row so the
+        // contraction dim of the accumulate MFMAs is the C-row dim);
+        // dP[q, k own] for the dK waves
         f32x16 acc_s = {}, acc_dp = {};
         #pragma unroll
         for (int c = 0; c < 8; ++c) {
-          acc_s = mfma32(kimg_frag(qcur, sb, c), kfrag[c], acc_s);
-        }
-        if (DK) {
-          #pragma unroll
-          for (int c = 0; c < 8; ++c) {
-            acc_dp = mfma32(kimg_frag(do_img, sb, c), vfrag[c], acc_dp);
+          acc_s = mfma32(kimg_frag(qn_c, sb, c), kfrag[c], acc_s);
+          if (is_dk) {
+            acc_dp = mfma32(kimg_frag(don_c, sb, c), vfrag[c], acc_dp);
           }
         }
+
         const int qs0 = q0 + 32 * sb;
         const bool mask_tile = causal || (qs0 + 32 > S) || !krow_valid;
         float cv[16];
@@ -863,47 +888,45 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
             }
           }
           const float pr = __builtin_amdgcn_exp2f(s2 - lse_s[ql]);
-          cv[r] = DK ? scale * pr * (acc_dp[r] - del_s[ql]) : pr;
+          cv[r] = is_dk ? scale * pr * (acc_dp[r] - del_s[ql]) : pr;
         }
 
         mbf16x8 f0, f1;
         cvals_to_frags(cv, hi, &f0, &f1);
 
-        // dV[k,d] += P^T . dO   |   dK[k,d] += dS^T . Q
+        // dV[k,d] += P^T . dO  (B rows from dO^T image)
+        // dK[k,d] += dS^T . Q  (B rows from Q^T image)
+        const char* acc_img = is_dk ? qt_c : dot_c;
         #pragma unroll
         for (int dt = 0; dt < 4; ++dt) {
-          acc[dt] = mfma32(f0, vimg_frag(tr_img, dt, 2 * sb), acc[dt]);
+          acc[dt] = mfma32(f0, vimg_frag(acc_img, dt, 2 * sb), acc[dt]);
           acc[dt] =
-              mfma32(f1, vimg_frag(tr_img, dt, 2 * sb + 1), acc[dt]);
+              mfma32(f1, vimg_frag(acc_img, dt, 2 * sb + 1), acc[dt]);
         }
       }
     }
 
     if (has_next) {
-      // all staging grouped here (glds included): interleaving the glds
-      // address math with the MFMA region spilled ~50 dwords/lane
-      stage_glds(idx + 1, qnxt);
-      load_tr_regs(idx + 1, trr);
-      if (DK) {
-        ushort8 dor[4];
-        load_nat_regs(idx + 1, dor);
-        asm volatile("s_waitcnt vmcnt(0)");
-        __syncthreads();   // everyone done with tr_img/do_img/lse_s
-        write_nat_sw(dor, do_img);
-        write_v_tr(trr, tr_img);
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)");
-        __syncthreads();
-        write_v_tr(trr, tr_img);
-      }
-      stage_lse(idx + 1);
-      __syncthreads();     // staged images visible
-      char* tq = qcur; qcur = qnxt; qnxt = tq;
+      asm volatile("s_waitcnt vmcnt(0)");
+      // nxt tr buffers were last read two tiles ago — safe to fill now.
+      // lse_s is single-buffered: wait for all waves' reads of tile t
+      // before overwriting, then one barrier for visibility.
+      write_v_tr8(trq, qt_n);
+      write_v_tr8(trdo, dot_n);
+      __syncthreads();
+      const int nidx2 = idx + 1;
+      stage_lse(nidx2 / nt_eff, t0 + nidx2 % nt_eff);
+      __syncthreads();
+      char* tp;
+      tp = qn_c; qn_c = qn_n; qn_n = tp;
+      tp = don_c; don_c = don_n; don_n = tp;
+      tp = qt_c; qt_c = qt_n; qt_n = tp;
+      tp = dot_c; dot_c = dot_n; dot_n = tp;
     }
   }
 
-  // Store via LDS transpose so global writes are 8-B vectors.
-  // Each wave owns a private [32 k][36 d] f32 region (reuses staging smem).
+  // store via LDS transpose (8-B global writes); wave-private f32 region
+  unsigned short* out = is_dk ? dk : dv;
   __syncthreads();
   float* tr = (float*)smem + wave * (32 * 36);
   if (wave_active) {
@@ -971,16 +994,11 @@ extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
                      scale, causal, q_rs, kv_rs, dqkv_q_rs);
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
   const long dout_rs = (long)Hq * HD;
-  hipLaunchKernelGGL((attn_bwd_dkv_kernel<false>), dim3(B * Hkv * nkt),
-                     dim3(256), 0, stream, (const unsigned short*)q,
+  hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(B * Hkv * nkt),
+                     dim3(512), 0, stream, (const unsigned short*)q,
                      (const unsigned short*)k, (const unsigned short*)v,
                      (const unsigned short*)dout, (const float*)lse,
-                     (const float*)delta, (unsigned short*)dv, B, S, Hq, Hkv,
-                     scale, causal, q_rs, kv_rs, dout_rs, dqkv_kv_rs);
-  hipLaunchKernelGGL((attn_bwd_dkv_kernel<true>), dim3(B * Hkv * nkt),
-                     dim3(256), 0, stream, (const unsigned short*)q,
-                     (const unsigned short*)k, (const unsigned short*)v,
-                     (const unsigned short*)dout, (const float*)lse,
-                     (const float*)delta, (unsigned short*)dk, B, S, Hq, Hkv,
+                     (const float*)delta, (unsigned short*)dk,
+                     (unsigned short*)dv, B, S, Hq, Hkv,
                      scale, causal, q_rs, kv_rs, dout_rs, dqkv_kv_rs);
 }
