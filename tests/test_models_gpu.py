@@ -80,3 +80,31 @@ def test_launcher_rccl_e2e():
         cwd=str(REPO), env=env, capture_output=True, text=True, timeout=300,
     )
     assert out.returncode == 0, out.stdout + out.stderr
+
+
+def test_launcher_elastic_restart_rccl(tmp_path):
+    """BASELINE config 4: worker kill -> re-rendezvous -> success over RCCL."""
+    script = tmp_path / "flaky_gpu.py"
+    marker = tmp_path / "marker"
+    script.write_text(
+        f"""
+import os, sys
+sys.path.insert(0, {str(REPO)!r})
+from torchx_amd.apps.compute_world_size import compute_world_size
+marker = {str(marker)!r}
+if not os.path.exists(marker):
+    open(marker, "w").close()
+    sys.exit(17)
+compute_world_size()
+"""
+    )
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO) + os.pathsep + env.get("PYTHONPATH", "")
+    out = subprocess.run(
+        [sys.executable, "-m", "torchx_amd.cli.main", "run", "--wait",
+         "--scheduler", "local_cwd", "dist.ddp", "-j", "1x1",
+         "--max_retries", "1", "--script", str(script)],
+        cwd=str(REPO), env=env, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-2000:]
+    assert marker.exists()

@@ -47,6 +47,7 @@ from .api import (  # noqa: F401
 from .builders import materialize_appdef, parse_mounts  # noqa: F401
 from .capabilities import GFX_ARCH, HBM_GB, XGMI_LINKS, CapabilityKey
 from .named_resources import NAMED_RESOURCES
+from .named_resources_cloud import NAMED_RESOURCES as CLOUD_NAMED_RESOURCES
 
 _lock = threading.Lock()
 _extra_named_resources: Dict[str, Callable[[], Resource]] = {}
@@ -77,8 +78,11 @@ def _load_custom_from_env() -> None:
 
 
 def named_resources() -> Dict[str, Callable[[], Resource]]:
+    from .named_resources_cloud import NAMED_RESOURCES as CLOUD
+
     _load_custom_from_env()
     out = dict(NAMED_RESOURCES)
+    out.update(CLOUD)
     out.update(_extra_named_resources)
     return out
 
