@@ -30,6 +30,10 @@ def main():
     assert torch.cuda.is_available()
     torch.cuda.tunable.enable(True)
     torch.cuda.tunable.tuning_enable(True)
+    # cap per-candidate cost: the defaults (30 ms / 100 iters per
+    # solution x ~600 hipblaslt solutions) take ~2.5 min per GEMM shape
+    torch.cuda.tunable.set_max_tuning_duration(5)
+    torch.cuda.tunable.set_max_tuning_iterations(10)
     # filename must be set BEFORE ops run; write_file at the end
     torch.cuda.tunable.set_filename(args.out)
 

@@ -518,7 +518,10 @@ class runopts:
                 last_key = k
             elif last_key is not None:
                 prev = cfg[last_key]
-                if isinstance(prev, list):
+                if isinstance(prev, dict) and ":" in token:
+                    dk, _, dv = token.partition(":")
+                    prev[dk.strip()] = dv.strip()
+                elif isinstance(prev, list):
                     prev.append(token)
                 else:
                     cfg[last_key] = [str(prev), token]
