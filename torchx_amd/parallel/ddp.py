@@ -57,6 +57,11 @@ class FlatDDP:
         if bucket_bytes is None:
             bucket_bytes = int(os.environ.get("TORCHX_AMD_BUCKET_MB", "64")) << 20
         self.bucket_bytes = bucket_bytes
+        # RCCL supports AVG in-collective: skips a whole-grad-buffer divide
+        # sweep at finish (gloo lacks AVG -> reduce SUM + divide once)
+        self.use_avg = (
+            self.enabled and dist.get_backend(process_group) == "nccl"
+        )
         self.sync_enabled = True
         self.buckets: List[_Bucket] = []
         self.param2bucket: Dict[int, _Bucket] = {}
@@ -114,8 +119,8 @@ class FlatDDP:
 
     def _launch(self, b: _Bucket) -> None:
         buf = self.flat.flat_grad[b.group][b.start:b.end]
-        b.work = dist.all_reduce(buf, op=dist.ReduceOp.SUM, group=self.pg,
-                                 async_op=True)
+        op = dist.ReduceOp.AVG if self.use_avg else dist.ReduceOp.SUM
+        b.work = dist.all_reduce(buf, op=op, group=self.pg, async_op=True)
         b.launched = True
 
     def no_sync(self):
@@ -144,7 +149,11 @@ class FlatDDP:
             if b.work is not None:
                 b.work.wait()
         ws = dist.get_world_size(self.pg)
-        for _, _, grad in self.flat.groups():
+        for g, _, grad in self.flat.groups():
+            # AVG path: reduced buckets are already averaged; only the
+            # non-reduced (EP-local) groups still need the 1/ws factor
+            if self.use_avg and g not in self.local_groups:
+                continue
             grad.div_(ws)
         self._reset()
 
