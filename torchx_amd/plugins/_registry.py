@@ -86,6 +86,11 @@ class PluginRegistry:
                 ("torchx_amd.schedulers", PluginType.SCHEDULER),
                 ("torchx_amd.tracker", PluginType.TRACKER),
                 ("torchx_amd.named_resources", PluginType.NAMED_RESOURCE),
+                # legacy reference group names keep existing torchx
+                # plugin packages working unchanged (wire-contract parity)
+                ("torchx.schedulers", PluginType.SCHEDULER),
+                ("torchx.tracker", PluginType.TRACKER),
+                ("torchx.named_resources", PluginType.NAMED_RESOURCE),
             ):
                 found = (
                     eps.select(group=group)
