@@ -72,3 +72,18 @@ def test_tracker_artifact_written(tmp_path, monkeypatch):
                        "--checkpoint-dir", ckpt, "--seq-len", "32",
                        "--micro-batch", "1"])
     assert rc == 0
+
+
+def test_datapreproc_synthetic(tmp_path):
+    import numpy as np
+
+    from torchx_amd.apps import datapreproc
+
+    out = tmp_path / "tokens.bin"
+    rc = datapreproc.main(["--synthetic", "20", "--output", str(out),
+                           "--vocab-size", "512"])
+    assert rc == 0
+    ids = np.frombuffer(out.read_bytes(), dtype=np.uint32)
+    assert len(ids) > 20
+    assert ids.max() < 512
+    assert (ids == 1).sum() == 20  # one EOS per doc
