@@ -145,3 +145,34 @@ def test_runner_list(tmp_path):
         _wait(runner, handle)
         apps = runner.list("local_cwd")
         assert any(handle.endswith(a.app_id) for a in apps)
+
+
+def test_launch_latency_tool():
+    """BASELINE headline latency metric is measurable end to end."""
+    import json
+    import subprocess
+    import sys as _sys
+    from pathlib import Path
+
+    repo = Path(__file__).resolve().parent.parent
+    out = subprocess.run(
+        [_sys.executable, str(repo / "tools" / "launch_latency.py"),
+         "--nproc", "2"],
+        capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stdout + out.stderr
+    result = json.loads(out.stdout.strip().splitlines()[-1])
+    assert result["metric"] == "launch_to_first_step_seconds"
+    assert result["value"] and result["value"] > 0
+
+
+def test_ddp_rocprof_argv():
+    """--rocprof reaches the agent argv (SURVEY §5.1 profiling bridge)."""
+    from torchx_amd.specs.builders import materialize_appdef
+    from torchx_amd.specs.finder import get_component
+
+    comp = get_component("dist.ddp")
+    app = materialize_appdef(
+        comp.fn, ["-j", "1x2", "--script", "t.py", "--rocprof", "True"],
+    )
+    assert "--rocprof" in app.roles[0].args

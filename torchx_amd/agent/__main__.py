@@ -90,6 +90,11 @@ def spawn_workers(args, rdzv: RendezvousResult, restart_count: int,
     else:
         cmd_base = [sys.executable, args.script]
     for lr in range(nproc):
+        prof_prefix: List[str] = []
+        if args.rocprof:
+            prof_dir = os.path.join(log_dir, f"rocprof_rank{lr}")
+            prof_prefix = ["rocprofv3", "--kernel-trace", "--stats",
+                           "-d", prof_dir, "-o", "prof", "--"]
         grank = rdzv.node_rank * nproc + lr
         error_file = os.path.join(log_dir, f"worker_{grank}_error.json")
         if os.path.exists(error_file):
@@ -114,7 +119,7 @@ def spawn_workers(args, rdzv: RendezvousResult, restart_count: int,
         stdout = subprocess.PIPE if args.tee else None
         stderr = subprocess.PIPE if args.tee else None
         proc = subprocess.Popen(
-            cmd_base + args.script_args,
+            prof_prefix + cmd_base + args.script_args,
             env=env,
             stdout=stdout,
             stderr=stderr,
@@ -196,6 +201,9 @@ def main(argv: Optional[List[str]] = None) -> int:
     p.add_argument("--tee", action="store_true", default=True)
     p.add_argument("--no-tee", dest="tee", action="store_false")
     p.add_argument("--no-python", action="store_true")
+    p.add_argument("--rocprof", action="store_true",
+                   help="wrap each worker in rocprofv3 --kernel-trace "
+                        "--stats (output under the log dir)")
     p.add_argument("--log-dir", type=str, default=None)
     p.add_argument("script", type=str)
     p.add_argument("script_args", nargs=argparse.REMAINDER)

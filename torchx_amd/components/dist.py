@@ -60,6 +60,7 @@ def ddp(
     mounts: Optional[List[str]] = None,
     debug: bool = False,
     tee: bool = True,
+    rocprof: bool = False,
 ) -> specs.AppDef:
     """Distributed data-parallel application (elastic, one agent per node).
 
@@ -81,6 +82,9 @@ def ddp(
         mounts: mount specs (type=bind,src=...,dst=...[,readonly])
         debug: enable the RCCL/ROCm debug env preset
         tee: prefix worker output with rank labels
+        rocprof: wrap each worker in `rocprofv3 --kernel-trace --stats`
+            (profiles land in $PET_LOG_DIR; SURVEY §5.1 launcher->profiling
+            bridge)
     """
     if (script is None) == (m is None):
         raise ValueError("exactly one of --script / -m must be set")
@@ -113,6 +117,8 @@ def ddp(
     ]
     if not tee:
         agent_args.append("--no-tee")
+    if rocprof:
+        agent_args.append("--rocprof")
     if m is not None:
         # agent execs `python3 -m <module>` via --no-python
         agent_args += ["--no-python", "python3", "-m", m]
