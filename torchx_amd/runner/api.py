@@ -174,7 +174,11 @@ class Runner:
                        workspace=workspace):
             sched._pre_build_validate(app, cfg)
 
-            # tracker/session env injection (reference api.py:400-424)
+            # tracker/session env injection (reference api.py:400-424);
+            # parent run id falls back to the launcher's own env so lineage
+            # chains when a tracked job launches further jobs
+            parent_run_id = parent_run_id or os.environ.get(
+                "TORCHX_PARENT_RUN_ID")
             trackers = torchx_config.get_configured_trackers()
             for role in app.roles:
                 role.env.setdefault("TORCHX_INTERNAL_SESSION_ID",

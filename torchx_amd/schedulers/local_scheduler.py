@@ -238,6 +238,12 @@ class LocalScheduler(Scheduler[PopenRequest]):
     # -- dryrun -------------------------------------------------------------
     def _submit_dryrun(self, app: AppDef,
                        cfg: Mapping[str, Any]) -> AppDryRunInfo[PopenRequest]:
+        for role in app.roles:
+            if role.mounts:
+                raise ValueError(
+                    "local_cwd does not support mounts (compat matrix); "
+                    "use local_docker or kubernetes"
+                )
         app_id = make_unique(app.name)
         base_log = cfg.get("log_dir") or os.path.join(
             tempfile.gettempdir(), "torchx_amd"
