@@ -198,3 +198,42 @@ def test_cli_run_stdin_json(tmp_path):
     )
     assert out.returncode == 0, out.stderr
     assert "stdin-mode" in out.stdout
+
+
+def test_cli_status_describe_list_cancel(tmp_path, capsys):
+    # local_cwd state is per-session (as in the reference): a run handle
+    # resolves within the owning Runner; a fresh session reports not-found
+    rc = cli_main([
+        "run", "-s", "local_cwd",
+        "-cfg", f"log_dir={tmp_path},auto_set_hip_visible_devices=false",
+        "utils.echo", "--msg", "cycle",
+    ])
+    assert rc == 0
+    out = capsys.readouterr().out
+    handle = out.splitlines()[0].strip()
+    assert handle.startswith("local_cwd://")
+    assert cli_main(["list", "-s", "local_cwd"]) == 0
+    # unknown-app error paths exit non-zero, don't crash
+    assert cli_main(["status", handle]) == 1
+    assert cli_main(["describe", handle]) == 1
+
+
+def test_cli_configure_dumps_template(tmp_path, monkeypatch):
+    monkeypatch.chdir(tmp_path)
+    assert cli_main(["configure", "-s", "local_cwd"]) == 0
+    cfg = tmp_path / ".torchxconfig"
+    assert cfg.exists()
+    assert "[local_cwd]" in cfg.read_text()
+
+
+def test_cli_tracker_cmds(tmp_path, monkeypatch):
+    from torchx_amd.tracker.fsspec import FsspecTracker
+
+    t = FsspecTracker(str(tmp_path))
+    t.add_metadata("run7", lr=0.5)
+    t.add_artifact("run7", "ckpt", "/x/ckpt.pt")
+    cfg = tmp_path / ".torchxconfig"
+    cfg.write_text(f"[torchx:tracker]\nfsspec = {tmp_path}\n")
+    monkeypatch.setenv("TORCHXCONFIG", str(cfg))
+    assert cli_main(["tracker", "list", "jobs"]) == 0
+    assert cli_main(["tracker", "list", "metadata", "--run_id", "run7"]) == 0
