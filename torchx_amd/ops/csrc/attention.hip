@@ -261,6 +261,30 @@ __device__ __forceinline__ void map_block_fwd(int B, int Hq, int Hkv,
   }
 }
 
+// Build the rot-placed transposed image from a staged swizzled natural
+// image: per thread NCHUNK b128 LDS reads + 8 scalar writes each.  The
+// source tile arrived by glds earlier in the iteration, so the transpose
+// costs an LDS round trip (~50 cyc) instead of a global re-read (~200+).
+template <int NTHREADS>
+__device__ __forceinline__ void lds_nat_to_tr(const char* nat, char* tr_img) {
+  const int tid = threadIdx.x;
+  #pragma unroll
+  for (int p = 0; p < 1024 / NTHREADS; ++p) {
+    const int c = tid + p * NTHREADS;
+    const int r = c >> 4, g = c & 15;
+    const ushort8 v = *(const ushort8*)(
+        nat + r * 256 + ((g * 16) ^ ((r & 15) << 4)));
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d = 8 * g + j;
+      const int gr = ((r >> 3) + vrot(d)) & 7;
+      *(unsigned short*)(tr_img + d * 128 + gr * 16 + (r & 7) * 2) = v[j];
+    }
+  }
+}
+
+
+
 __global__ void __launch_bounds__(256, 2)
 attn_fwd_kernel(const unsigned short* __restrict__ q,
                 const unsigned short* __restrict__ k,
@@ -559,13 +583,11 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   const int ntiles = (kv_limit + FKV - 1) / FKV;
   const int qw_max = min(qw0 + QBLK - 1, S - 1);
 
-  ushort8 vr[4];
   stage_k_glds(kb, 0, kv_seq_stride, S, kcur);
   stage_k_glds(vb, 0, kv_seq_stride, S, vcur);
-  load_v_regs(kb, 0, kv_seq_stride, S, vr);
   asm volatile("s_waitcnt vmcnt(0)");
   __syncthreads();
-  write_v_tr(vr, ktr);
+  lds_nat_to_tr<256>(kcur, ktr);
   __syncthreads();
 
   for (int t = 0; t < ntiles; ++t) {
@@ -628,13 +650,12 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
     }
 
     if (has_next) {
-      // K^T regs loaded here (L2-hit) so vr is live only briefly;
-      // keeping it across compute costs 16 VGPR at the pressure peak
-      load_v_regs(kb, kv0 + FKV, kv_seq_stride, S, vr);
       asm volatile("s_waitcnt vmcnt(0)");
-      __syncthreads();   // all waves done reading ktr for tile t
-      write_v_tr(vr, ktr);
-      __syncthreads();   // ktr + glds images visible
+      __syncthreads();   // all waves done reading ktr; nxt glds visible
+      // K^T rebuilt from the just-landed K natural image: an LDS round
+      // trip instead of a second global K stream
+      lds_nat_to_tr<256>(knxt, ktr);
+      __syncthreads();   // ktr visible
       char* tk = kcur; kcur = knxt; knxt = tk;
       char* tv = vcur; vcur = vnxt; vnxt = tv;
     }
@@ -727,6 +748,7 @@ __device__ __forceinline__ void write_v_tr8(const ushort8 vr[2], char* vimg) {
 // B-op), dO^T (dV accumulate B-op) — double-buffered in 128 KiB LDS, one
 // barrier per tile.
 // ---------------------------------------------------------------------------
+
 __global__ void __launch_bounds__(512, 2)
 attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     const unsigned short* __restrict__ k,
@@ -829,15 +851,14 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
     }
   };
 
-  ushort8 trq[2], trdo[2];
-  // prologue: tile (gh=0, t=t0)
+  // prologue: tile (gh=0, t=t0) — glds the natural images, then build
+  // the transposed images from LDS
   stage_k_glds8(head_q(0), (long)t0 * FKV, q_seq_stride, S, qn_c);
   stage_k_glds8(head_do(0), (long)t0 * FKV, dout_rs, S, don_c);
-  load_v_regs8(head_q(0), (long)t0 * FKV, q_seq_stride, S, trq);
-  load_v_regs8(head_do(0), (long)t0 * FKV, dout_rs, S, trdo);
   asm volatile("s_waitcnt vmcnt(0)");
-  write_v_tr8(trq, qt_c);
-  write_v_tr8(trdo, dot_c);
+  __syncthreads();
+  lds_nat_to_tr<512>(qn_c, qt_c);
+  lds_nat_to_tr<512>(don_c, dot_c);
   stage_lse(0, t0);
   __syncthreads();
 
@@ -847,14 +868,16 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
     const int q0 = t * FKV;
     const int hq = hkv * G + gh;
     const bool has_next = (idx + 1) < total;
+    const int ngh = (idx + 1) / nt_eff;
+    const long nrow0 = (long)(t0 + (idx + 1) % nt_eff) * FKV;
     if (has_next) {
-      const int nidx = idx + 1;
-      const int ngh = nidx / nt_eff;
-      const long nrow0 = (long)(t0 + nidx % nt_eff) * FKV;
+      // async glds issue BEFORE compute (their latency hides under the
+      // MFMAs; the address temporaries are consumed by the instruction
+      // immediately so this does not raise register pressure).  The
+      // reg-staged tr loads stay in the tail: their 32 data registers
+      // across the compute phase are what spilled.
       stage_k_glds8(head_q(ngh), nrow0, q_seq_stride, S, qn_n);
       stage_k_glds8(head_do(ngh), nrow0, dout_rs, S, don_n);
-      load_v_regs8(head_q(ngh), nrow0, q_seq_stride, S, trq);
-      load_v_regs8(head_do(ngh), nrow0, dout_rs, S, trdo);
     }
 
     const bool needed =
@@ -907,15 +930,13 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
     }
 
     if (has_next) {
-      asm volatile("s_waitcnt vmcnt(0)");
-      // nxt tr buffers were last read two tiles ago — safe to fill now.
-      // lse_s is single-buffered: wait for all waves' reads of tile t
-      // before overwriting, then one barrier for visibility.
-      write_v_tr8(trq, qt_n);
-      write_v_tr8(trdo, dot_n);
-      __syncthreads();
-      const int nidx2 = idx + 1;
-      stage_lse(nidx2 / nt_eff, t0 + nidx2 % nt_eff);
+      asm volatile("s_waitcnt vmcnt(0)");  // own glds of the nxt tiles done
+      __syncthreads();                     // all waves' glds visible
+      // transposed images built from the LDS natural images: no global
+      // re-read; the second barrier also covers the lse_s overwrite
+      lds_nat_to_tr<512>(qn_n, qt_n);
+      lds_nat_to_tr<512>(don_n, dot_n);
+      stage_lse(ngh, t0 + (idx + 1) % nt_eff);
       __syncthreads();
       char* tp;
       tp = qn_c; qn_c = qn_n; qn_n = tp;
