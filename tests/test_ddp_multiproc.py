@@ -83,3 +83,39 @@ def test_flatddp_gloo_world2(tmp_path):
     out = proc.stdout.decode() + proc.stderr.decode()
     assert proc.returncode == 0, out[-4000:]
     assert "DDP_MULTIPROC_OK" in out, out[-4000:]
+
+
+def test_agent_two_nodes(tmp_path):
+    """Two agent processes (nnodes=2) rendezvous and form one world of 4
+    (the multi-node path of the launcher, on one host via 127.0.0.1)."""
+    import socket
+    import time
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO) + os.pathsep + env.get("PYTHONPATH", "")
+    procs = []
+    for node in range(2):
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m", "torchx_amd.agent",
+             "--nnodes", "2:2", "--nproc-per-node", "2",
+             "--rdzv-endpoint", f"127.0.0.1:{port}",
+             "--rdzv-id", "twonode", "--log-dir", str(tmp_path / str(node)),
+             "--no-python", sys.executable, "-m",
+             "torchx_amd.apps.compute_world_size"],
+            env=env, cwd=str(REPO),
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        ))
+    outs = []
+    for p in procs:
+        try:
+            out, _ = p.communicate(timeout=240)
+        except subprocess.TimeoutExpired:
+            p.kill()
+            out, _ = p.communicate()
+        outs.append(out)
+    assert all(p.returncode == 0 for p in procs), outs
+    assert any("computed world size = 4" in o for o in outs), outs
