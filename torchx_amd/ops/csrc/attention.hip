@@ -34,10 +34,6 @@ typedef __bf16 mbf16x8 __attribute__((ext_vector_type(8)));
 #define BLOCK_Q (QBLK * NWAVES)  // q rows per block (fwd / dq)
 #define BLOCK_K (KBLK * NWAVES)  // kv rows per block (dkv)
 
-// padded LDS strides (bank-conflict-free ds_read_b128, see guide §6 G4)
-#define NAT_STRIDE 136   // [32][136] natural tile (row-major, 8-col pad)
-#define TR_STRIDE 40     // [128][40] transposed tile (d-major, 8-col pad)
-
 __device__ __forceinline__ f32x16 mfma32(mbf16x8 a, mbf16x8 b, f32x16 c) {
   return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
 }
@@ -80,62 +76,6 @@ __device__ __forceinline__ void cvals_to_frags(const float* p, bool hi,
   uint4_v u1 = {w[1][0], w[1][1], w[1][2], w[1][3]};
   *f0 = __builtin_bit_cast(mbf16x8, u0);
   *f1 = __builtin_bit_cast(mbf16x8, u1);
-}
-
-// A/B-style fragment read from an LDS tile: lane l -> row (l&31),
-// cols [col0 + (l>>5)*8, +8).  One ds_read_b128 when 16-B aligned.
-__device__ __forceinline__ mbf16x8 lds_frag(const unsigned short* tile,
-                                            int row_stride, int col0) {
-  const int lane = threadIdx.x & 63;
-  const unsigned short* p =
-      tile + (lane & 31) * row_stride + col0 + (lane >> 5) * 8;
-  return __builtin_bit_cast(mbf16x8, *(const ushort8*)p);
-}
-
-// Natural-layout cooperative staging of a [rows<=32][128] bf16 tile from a
-// BSHD tensor into LDS (guarded, zero-filled beyond seq).  256 threads.
-__device__ __forceinline__ void stage_nat(
-    const unsigned short* __restrict__ src,  // tensor base
-    long row0,                                // first seq row
-    long seq_stride,                          // H*128 (elements)
-    int rows, int S, unsigned short* dst) {
-  const int id0 = threadIdx.x;               // 512 chunks of 8, 2 per thread
-  #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    int id = id0 + p * 256;
-    int r = id / 16, c8 = (id % 16) * 8;
-    ushort8 v;
-    if (r < rows && row0 + r < S) {
-      v = *(const ushort8*)(src + (row0 + r) * seq_stride + c8);
-    } else {
-      v = (ushort8)0;
-    }
-    *(ushort8*)(dst + r * NAT_STRIDE + c8) = v;
-  }
-}
-
-// Transposed staging: dst[d][k] = src[k][d], dst is [128][TR_STRIDE].
-__device__ __forceinline__ void stage_tr(
-    const unsigned short* __restrict__ src, long row0, long seq_stride,
-    int rows, int S, unsigned short* dst, float scale) {
-  const int id0 = threadIdx.x;
-  #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    int id = id0 + p * 256;
-    int r = id / 16, c8 = (id % 16) * 8;
-    ushort8 v;
-    if (r < rows && row0 + r < S) {
-      v = *(const ushort8*)(src + (row0 + r) * seq_stride + c8);
-    } else {
-      v = (ushort8)0;
-    }
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      unsigned short u = v[j];
-      if (scale != 1.0f) u = f32_to_bf16(bf16_to_f32(u) * scale);
-      dst[(c8 + j) * TR_STRIDE + r] = u;
-    }
-  }
 }
 
 // ---------------------------------------------------------------------------
@@ -705,35 +645,6 @@ __device__ __forceinline__ void stage_k_glds8(
         (const __attribute__((address_space(1))) unsigned int*)src,
         (__attribute__((address_space(3))) unsigned int*)(kimg + i * 1024),
         16, 0, 0);
-  }
-}
-
-__device__ __forceinline__ void load_v_regs8(
-    const unsigned short* __restrict__ vb, long kv0, long stride_elems,
-    int S, ushort8 vr[2]) {
-  const int tid = threadIdx.x;
-  #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    const int c = tid + p * 512;
-    const int r = c >> 4, g = c & 15;
-    long srow = kv0 + r;
-    if (srow >= S) srow = S - 1;
-    vr[p] = *(const ushort8*)(vb + srow * stride_elems + g * 8);
-  }
-}
-
-__device__ __forceinline__ void write_v_tr8(const ushort8 vr[2], char* vimg) {
-  const int tid = threadIdx.x;
-  #pragma unroll
-  for (int p = 0; p < 2; ++p) {
-    const int c = tid + p * 512;
-    const int r = c >> 4, g = c & 15;
-    #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int d = 8 * g + j;
-      const int gr = ((r >> 3) + vrot(d)) & 7;
-      *(unsigned short*)(vimg + d * 128 + gr * 16 + (r & 7) * 2) = vr[p][j];
-    }
   }
 }
 
