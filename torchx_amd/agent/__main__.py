@@ -1,4 +1,6 @@
-"""torchx_amd elastic agent — the native torchrun replacement.
+"""torchx_amd elastic agent — the native torchrun replacement (the
+reference invokes `torchrun` as a subprocess, torchx/components/dist.py:262,
+delegating all of this to torchelastic).
 
 Launched by the dist.ddp component (one agent per node), it:
   * joins a c10d TCPStore rendezvous (rendezvous.py),
