@@ -9,8 +9,7 @@ replica, with a c10d rendezvous at the scheduler-provided rank-0 host
 from __future__ import annotations
 
 import os
-import shlex
-from typing import Dict, Iterable, List, Optional, Tuple
+from typing import Dict, List, Optional, Tuple
 
 import torchx_amd.specs as specs
 

@@ -8,7 +8,7 @@ test contract: HIP kernel vs plain PyTorch fp32 reference of the same op).
 from __future__ import annotations
 
 import math
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

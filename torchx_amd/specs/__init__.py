@@ -45,9 +45,13 @@ from .api import (  # noqa: F401
     runopts,
 )
 from .builders import materialize_appdef, parse_mounts  # noqa: F401
-from .capabilities import GFX_ARCH, HBM_GB, XGMI_LINKS, CapabilityKey
+from .capabilities import (  # noqa: F401 — public re-exports
+    GFX_ARCH,
+    HBM_GB,
+    XGMI_LINKS,
+    CapabilityKey,
+)
 from .named_resources import NAMED_RESOURCES
-from .named_resources_cloud import NAMED_RESOURCES as CLOUD_NAMED_RESOURCES
 
 _lock = threading.Lock()
 _extra_named_resources: Dict[str, Callable[[], Resource]] = {}
