@@ -311,13 +311,16 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
 
     const bool needed = wave_active && (!causal || kv0 <= qw_max);
     if (needed) {
-      // S^T[k, q] = K . Q^T over both 32-row k sub-tiles
+      // S^T[k, q] = K . Q^T over both 32-row k sub-tiles (T5: setprio
+      // keeps the MFMA pipe fed while the other wave issues memory ops)
       f32x16 acc0 = {}, acc1 = {};
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int c = 0; c < 8; ++c) {
         acc0 = mfma32(kimg_frag(k0, 0, c), qfrag[c], acc0);
         acc1 = mfma32(kimg_frag(k0, 1, c), qfrag[c], acc1);
       }
+      __builtin_amdgcn_s_setprio(0);
 
       float sv[32];
       #pragma unroll
@@ -373,6 +376,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
       cvals_to_frags(sv + 16, hi, &pf[2], &pf[3]);
 
       // O^T[d, q] += V^T . P  (A = V^T from the swizzled LDS image)
+      __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
         #pragma unroll
@@ -380,6 +384,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
           acc_o[dt] = mfma32(vimg_frag(vcur, dt, ks), pf[ks], acc_o[dt]);
         }
       }
+      __builtin_amdgcn_s_setprio(0);
     }
 
     if (has_next) {
