@@ -66,7 +66,7 @@ def test_rendezvous_single_node():
     assert res.node_rank == 0
     assert res.num_nodes == 1
     assert res.master_port
-    rdzv.close()
+    rdzv._store = None  # drop the TCPStore
 
 
 def test_free_port_is_bindable():
