@@ -665,6 +665,28 @@ __device__ __forceinline__ void stage_k_glds8(
 // barrier per tile.
 // ---------------------------------------------------------------------------
 
+// Wave-local variant for 8-wave blocks: wave w transposes ONLY the 8
+// rows it glds-staged itself ([8w, 8w+8)), so its own s_waitcnt vmcnt(0)
+// is the only ordering needed before the transpose — no pre-barrier.
+__device__ __forceinline__ void lds_nat_to_tr_own8(const char* nat,
+                                                   char* tr_img) {
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  #pragma unroll
+  for (int pp = 0; pp < 2; ++pp) {
+    const int c = wave * 128 + pp * 64 + lane;
+    const int r = c >> 4, g = c & 15;
+    const ushort8 v = *(const ushort8*)(
+        nat + r * 256 + ((g * 16) ^ ((r & 15) << 4)));
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d = 8 * g + j;
+      const int gr = ((r >> 3) + vrot(d)) & 7;
+      *(unsigned short*)(tr_img + d * 128 + gr * 16 + (r & 7) * 2) = v[j];
+    }
+  }
+}
+
 __global__ void __launch_bounds__(512, 2)
 attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     const unsigned short* __restrict__ k,
@@ -677,7 +699,7 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     int B, int S, int Hq, int Hkv, float scale, int causal,
                     long q_rs, long kv_rs, long dout_rs, long out_rs) {
   __shared__ __align__(16) char smem[8 * KIMG_BYTES];
-  __shared__ float lse_s[FKV], del_s[FKV];
+  __shared__ float lse_buf[2][FKV], del_buf[2][FKV];
   char* qn_c = smem;                       // Q natural cur
   char* don_c = smem + KIMG_BYTES;         // dO natural cur
   char* qt_c = smem + 2 * KIMG_BYTES;      // Q^T cur
@@ -753,17 +775,17 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   auto head_do = [&](int gh) {
     return dout + (long)b * S * dout_rs + (long)(hkv * G + gh) * HD;
   };
-  auto stage_lse = [&](int gh, int t) {
+  auto stage_lse = [&](int gh, int t, int buf) {
     const int hq_ = hkv * G + gh;
     const int q0_ = t * FKV;
     if (threadIdx.x < FKV) {
       const float* lse_b = lse + ((long)b * Hq + hq_) * S;
       const int qg = q0_ + threadIdx.x;
-      lse_s[threadIdx.x] = (qg < S) ? lse_b[qg] * LOG2E : 0.f;
+      lse_buf[buf][threadIdx.x] = (qg < S) ? lse_b[qg] * LOG2E : 0.f;
     } else if (threadIdx.x < 2 * FKV) {
       const float* del_b = delta + ((long)b * Hq + hq_) * S;
       const int qg = q0_ + (threadIdx.x - FKV);
-      del_s[threadIdx.x - FKV] = (qg < S) ? del_b[qg] : 0.f;
+      del_buf[buf][threadIdx.x - FKV] = (qg < S) ? del_b[qg] : 0.f;
     }
   };
 
@@ -772,12 +794,14 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   stage_k_glds8(head_q(0), (long)t0 * FKV, q_seq_stride, S, qn_c);
   stage_k_glds8(head_do(0), (long)t0 * FKV, dout_rs, S, don_c);
   asm volatile("s_waitcnt vmcnt(0)");
-  __syncthreads();
-  lds_nat_to_tr<512>(qn_c, qt_c);
-  lds_nat_to_tr<512>(don_c, dot_c);
-  stage_lse(0, t0);
+  lds_nat_to_tr_own8(qn_c, qt_c);
+  lds_nat_to_tr_own8(don_c, dot_c);
+  stage_lse(0, t0, 0);
   __syncthreads();
 
+  const float* lse_cur = lse_buf[0];
+  const float* del_cur = del_buf[0];
+  int lse_nxt = 1;
   for (int idx = 0; idx < total; ++idx) {
     const int gh = idx / nt_eff;
     const int t = t0 + idx % nt_eff;
@@ -826,8 +850,9 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
               s2 = -3.0e38f;
             }
           }
-          const float pr = __builtin_amdgcn_exp2f(s2 - lse_s[ql]);
-          cv[r] = is_dk ? scale * pr * (acc_dp[r] - del_s[ql]) : pr;
+          const float pr =
+              __builtin_amdgcn_exp2f(s2 - lse_cur[ql]);
+          cv[r] = is_dk ? scale * pr * (acc_dp[r] - del_cur[ql]) : pr;
         }
 
         mbf16x8 f0, f1;
@@ -847,13 +872,15 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
 
     if (has_next) {
       asm volatile("s_waitcnt vmcnt(0)");  // own glds of the nxt tiles done
-      __syncthreads();                     // all waves' glds visible
-      // transposed images built from the LDS natural images: no global
-      // re-read; the second barrier also covers the lse_s overwrite
-      lds_nat_to_tr<512>(qn_n, qt_n);
-      lds_nat_to_tr<512>(don_n, dot_n);
-      stage_lse(ngh, t0 + (idx + 1) % nt_eff);
-      __syncthreads();
+      // wave-local transpose of the rows THIS wave staged (no pre-barrier)
+      // into the double-buffered tr images; lse double-buffered likewise
+      lds_nat_to_tr_own8(qn_n, qt_n);
+      lds_nat_to_tr_own8(don_n, dot_n);
+      stage_lse(ngh, t0 + (idx + 1) % nt_eff, lse_nxt);
+      __syncthreads();   // ONE barrier per tile publishes everything
+      lse_cur = lse_buf[lse_nxt];
+      del_cur = del_buf[lse_nxt];
+      lse_nxt ^= 1;
       char* tp;
       tp = qn_c; qn_c = qn_n; qn_n = tp;
       tp = don_c; don_c = don_n; don_n = tp;
