@@ -19,18 +19,25 @@ _NODE_CPU = 192
 _NODE_MEM_MB = int(2_304 * 1024 * 0.96)
 HBM_GB_PER_GPU = 288
 
-_CAP = {"amd.com/gpu_arch": "gfx950", "amd.com/hbm_gb": HBM_GB_PER_GPU}
-
 
 def _mi355x(num_gpus: int) -> Callable[[], Resource]:
     def factory() -> Resource:
+        from .capabilities import GFX_ARCH, HBM_GB, XGMI_LINKS
+
         frac = num_gpus / 8
-        return Resource(
+        r = Resource(
             cpu=int(_NODE_CPU * frac),
             gpu=num_gpus,
             memMB=int(_NODE_MEM_MB * frac),
-            capabilities=dict(_CAP),
+            capabilities={"amd.com/gpu_arch": "gfx950",
+                          "amd.com/hbm_gb": HBM_GB_PER_GPU},
         )
+        # typed capability keys (specs/capabilities.py) for scheduler-side
+        # topology decisions: 7 point-to-point xGMI links per MI355X
+        GFX_ARCH.set(r, "gfx950")
+        HBM_GB.set(r, HBM_GB_PER_GPU)
+        XGMI_LINKS.set(r, 7)
+        return r
 
     return factory
 
