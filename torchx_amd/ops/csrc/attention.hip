@@ -225,6 +225,27 @@ __device__ __forceinline__ void lds_nat_to_tr(const char* nat, char* tr_img) {
 
 
 
+// Wave-local variant for 4-wave blocks: wave w transposes the 16 rows it
+// glds-staged itself ([16w, 16w+16)) — own vmcnt(0) orders the read.
+__device__ __forceinline__ void lds_nat_to_tr_own4(const char* nat,
+                                                   char* tr_img) {
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  #pragma unroll
+  for (int pp = 0; pp < 4; ++pp) {
+    const int c = wave * 256 + pp * 64 + lane;
+    const int r = c >> 4, g = c & 15;
+    const ushort8 v = *(const ushort8*)(
+        nat + r * 256 + ((g * 16) ^ ((r & 15) << 4)));
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int d = 8 * g + j;
+      const int gr = ((r >> 3) + vrot(d)) & 7;
+      *(unsigned short*)(tr_img + d * 128 + gr * 16 + (r & 7) * 2) = v[j];
+    }
+  }
+}
+
 __global__ void __launch_bounds__(256, 2)
 attn_fwd_kernel(const unsigned short* __restrict__ q,
                 const unsigned short* __restrict__ k,
@@ -233,8 +254,8 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
                 float* __restrict__ lse,  // [B,Hq,S]
                 int B, int S, int Hq, int Hkv, float scale, int causal,
                 long q_rs, long kv_rs) {  // per-seq-row element strides
-  // 3-deep K ring (glds prefetch 2 tiles ahead, counted vmcnt keeps the
-  // newest tile's loads in flight ACROSS the barrier — T3/T4) + 2-deep V
+  // K 2-ring + single V natural scratch (glds) + 2-deep V^T images,
+  // each transposed wave-locally from the scratch (as in the backward)
   __shared__ __align__(16) char smem[3 * KIMG_BYTES + 2 * VIMG_BYTES];
 
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
@@ -279,34 +300,26 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
   const int qw_max = min(qw0 + QBLK - 1, S - 1);
 
   char* k0 = smem;                       // K tile t   (cur)
-  char* k1 = smem + KIMG_BYTES;          // K tile t+1 (staged, waited)
-  char* k2 = smem + 2 * KIMG_BYTES;      // K tile t+2 (in flight)
-  char* vcur = smem + 3 * KIMG_BYTES;
-  char* vnxt = vcur + VIMG_BYTES;
+  char* k1 = smem + KIMG_BYTES;          // K tile t+1 (in flight)
+  char* vnat = smem + 2 * KIMG_BYTES;    // V natural glds scratch
+  char* vcur = smem + 3 * KIMG_BYTES;    // V^T cur
+  char* vnxt = vcur + VIMG_BYTES;        // V^T nxt
 
-  ushort8 vr[4];
-  // prologue: K0, V0 (waited), K1 (stays in flight past the barrier)
+  // prologue: glds K0 + V0 natural, wave-local transpose V0
   stage_k_glds(kb, 0, kv_seq_stride, S, k0);
-  load_v_regs(vb, 0, kv_seq_stride, S, vr);
-  if (ntiles > 1) {
-    stage_k_glds(kb, FKV, kv_seq_stride, S, k1);
-    asm volatile("s_waitcnt vmcnt(4)");  // K0 + V0 done; K1 in flight
-  } else {
-    asm volatile("s_waitcnt vmcnt(0)");
-  }
-  write_v_tr(vr, vcur);
+  stage_k_glds(vb, 0, kv_seq_stride, S, vnat);
+  asm volatile("s_waitcnt vmcnt(0)");
+  lds_nat_to_tr_own4(vnat, vcur);
   __syncthreads();
 
   for (int t = 0; t < ntiles; ++t) {
     const int kv0 = t * FKV;
     const bool has_next = (t + 1) < ntiles;
     if (has_next) {
-      // issue next-V and next-next-K BEFORE compute (T14: HBM latency
-      // hides under this tile's MFMAs; K gets ~2 tiles of cover)
-      load_v_regs(vb, kv0 + FKV, kv_seq_stride, S, vr);
-      if (t + 2 < ntiles) {
-        stage_k_glds(kb, kv0 + 2 * FKV, kv_seq_stride, S, k2);
-      }
+      // issue next K/V glds BEFORE compute (T14: their latency hides
+      // under this tile's MFMAs; the stream is L2-resident anyway)
+      stage_k_glds(kb, kv0 + FKV, kv_seq_stride, S, k1);
+      stage_k_glds(vb, kv0 + FKV, kv_seq_stride, S, vnat);
     }
 
     const bool needed = wave_active && (!causal || kv0 <= qw_max);
@@ -388,18 +401,13 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
     }
 
     if (has_next) {
-      if (t + 2 < ntiles) {
-        // oldest 8 outstanding = K(t+1) + V(t+1); K(t+2) stays in flight
-        asm volatile("s_waitcnt vmcnt(4)");
-      } else {
-        asm volatile("s_waitcnt vmcnt(0)");
-      }
-      write_v_tr(vr, vnxt);
-      // one barrier per tile: K(t+1)/V(t+1) stores visible to all waves;
-      // the buffers being overwritten were last read before the PREVIOUS
-      // barrier, so no front barrier is needed
+      asm volatile("s_waitcnt vmcnt(0)");  // own K/V glds landed
+      // wave-local transpose of the V rows THIS wave staged; one barrier
+      // per tile publishes K(t+1) + V^T(t+1) (the overwritten buffers
+      // were last read before the PREVIOUS barrier)
+      lds_nat_to_tr_own4(vnat, vnxt);
       __syncthreads();
-      char* tk = k0; k0 = k1; k1 = k2; k2 = tk;
+      char* tk = k0; k0 = k1; k1 = tk;
       char* tv = vcur; vcur = vnxt; vnxt = tv;
     }
   }
