@@ -119,3 +119,25 @@ def test_agent_two_nodes(tmp_path):
         outs.append(out)
     assert all(p.returncode == 0 for p in procs), outs
     assert any("computed world size = 4" in o for o in outs), outs
+
+
+def test_bench_contract_world2_cpu(tmp_path):
+    """The driver's bench invocation shape (torch.distributed.run, one rank
+    per device) must work end to end and print ONE valid JSON line."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO) + os.pathsep + env.get("PYTHONPATH", "")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes", "1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--standalone", str(REPO / "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--model", "tiny", "--device", "cpu"],
+        env=env, cwd=str(REPO), capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    result = json.loads(line)
+    assert result["metric"] == "tokens_per_second"
+    assert result["n_gpus"] == 2
+    assert result["value"] > 0
+    assert result["config"]["parallelism"] == "dp2"
