@@ -176,3 +176,41 @@ def test_ddp_rocprof_argv():
         comp.fn, ["-j", "1x2", "--script", "t.py", "--rocprof", "True"],
     )
     assert "--rocprof" in app.roles[0].args
+
+
+def test_ddp_elastic_nnodes_and_name():
+    from torchx_amd.specs.builders import materialize_appdef
+    from torchx_amd.specs.finder import get_component
+
+    comp = get_component("dist.ddp")
+    app = materialize_appdef(comp.fn, [
+        "-j", "2:4x8", "--script", "train.py", "--name", "exp1/run9",
+        "--env", "FOO=bar", "--debug", "True",
+    ])
+    role = app.roles[0]
+    assert app.name == "run9"
+    assert role.num_replicas == 4          # max nodes
+    assert role.min_replicas == 2          # elastic minimum
+    assert "--nnodes" in role.args
+    assert role.args[role.args.index("--nnodes") + 1] == "2:4"
+    assert role.env["FOO"] == "bar"
+    assert role.env["NCCL_DESYNC_DEBUG"] == "1"   # debug preset
+    # multi-node rendezvous points at the scheduler-provided rank0 host
+    ep = role.args[role.args.index("--rdzv-endpoint") + 1]
+    assert "TORCHX_RANK0_HOST" in ep
+
+
+def test_ddp_mounts_parse():
+    from torchx_amd.specs import BindMount
+    from torchx_amd.specs.builders import materialize_appdef
+    from torchx_amd.specs.finder import get_component
+
+    comp = get_component("dist.ddp")
+    app = materialize_appdef(comp.fn, [
+        "-j", "1x2", "--script", "t.py",
+        "--mounts", "type=bind,src=/data,dst=/mnt/data,readonly",
+    ])
+    m = app.roles[0].mounts[0]
+    assert isinstance(m, BindMount)
+    assert m.src_path == "/data" and m.dst_path == "/mnt/data"
+    assert m.read_only
