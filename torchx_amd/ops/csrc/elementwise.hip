@@ -86,7 +86,8 @@ __global__ void swiglu_fwd_kernel(
     for (int j = 0; j < 8; ++j) {
       float gf = bf16_to_f32(gv[j]);
       float uf = bf16_to_f32(uv[j]);
-      float sig = 1.0f / (1.0f + __expf(-gf));
+      float sig =
+          1.0f / (1.0f + __builtin_amdgcn_exp2f(-1.44269504f * gf));
       ov[j] = f32_to_bf16(gf * sig * uf);
     }
     *(ushort8*)(out + i) = ov;
@@ -111,7 +112,8 @@ __global__ void swiglu_bwd_kernel(
       float dof = bf16_to_f32(dov[j]);
       float gf = bf16_to_f32(gv[j]);
       float uf = bf16_to_f32(uv[j]);
-      float sig = 1.0f / (1.0f + __expf(-gf));
+      float sig =
+          1.0f / (1.0f + __builtin_amdgcn_exp2f(-1.44269504f * gf));
       float silu = gf * sig;
       float dsilu = sig * (1.0f + gf * (1.0f - sig));
       dgv[j] = f32_to_bf16(dof * uf * dsilu);
@@ -226,7 +228,8 @@ __global__ void swiglu_gu_fwd_kernel(
     for (int j = 0; j < 8; ++j) {
       float gf = bf16_to_f32(gv[j]);
       float uf = bf16_to_f32(uv[j]);
-      float sig = 1.0f / (1.0f + __expf(-gf));
+      float sig =
+          1.0f / (1.0f + __builtin_amdgcn_exp2f(-1.44269504f * gf));
       ov[j] = f32_to_bf16(gf * sig * uf);
     }
     *(ushort8*)(out + p * 8) = ov;
@@ -250,7 +253,8 @@ __global__ void swiglu_gu_bwd_kernel(
       float dof = bf16_to_f32(dov[j]);
       float gf = bf16_to_f32(gv[j]);
       float uf = bf16_to_f32(uv[j]);
-      float sig = 1.0f / (1.0f + __expf(-gf));
+      float sig =
+          1.0f / (1.0f + __builtin_amdgcn_exp2f(-1.44269504f * gf));
       float silu = gf * sig;
       dgv[j] = f32_to_bf16(dof * uf * sig * (1.0f + gf * (1.0f - sig)));
       duv[j] = f32_to_bf16(dof * silu);

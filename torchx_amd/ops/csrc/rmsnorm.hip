@@ -52,10 +52,14 @@ rmsnorm_fwd_kernel(
 
 // dx = r*(dy*w) - x * (r^3/H) * sum(dy*w*x);  dw = sum_rows(dy * x * r).
 // Each block walks ROWS_PER_BLOCK rows, keeping its dw partial in registers,
-// then does ONE atomicAdd per element at the end — 32x fewer contending
+// then does ONE atomicAdd per element at the end — far fewer contending
 // atomics than a per-row scheme (the naive version spent 14% of a full
-// Llama-3-8B step serializing 67M atomics on 4096 addresses).
-#define RMS_ROWS_PER_BLOCK 32
+// Llama-3-8B step serializing 67M atomics on 4096 addresses).  8 rows per
+// block: 32 starved the chip (rows/32 = 512 blocks = 2 per CU, and each
+// block walks its rows SERIALLY behind a per-row block reduce — measured
+// 25% of HBM roofline; 8 rows -> 2048 blocks keeps the atomic count low
+// while filling the CUs).
+#define RMS_ROWS_PER_BLOCK 8
 
 template <int ITERS>
 __global__ void __launch_bounds__(256)
