@@ -214,3 +214,57 @@ def test_ddp_mounts_parse():
     assert isinstance(m, BindMount)
     assert m.src_path == "/data" and m.dst_path == "/mnt/data"
     assert m.read_only
+
+
+def test_success_manifest_written(tmp_path):
+    import os
+
+    with get_runner("t") as runner:
+        handle = runner.run_component(
+            "utils.echo", ["--msg", "done"], scheduler="local_cwd",
+            cfg={"log_dir": str(tmp_path),
+                 "auto_set_hip_visible_devices": False},
+        )
+        status = _wait(runner, handle, timeout=60)
+        assert status.state == AppState.SUCCEEDED
+    # the local scheduler writes a SUCCESS manifest on terminal close
+    found = []
+    for root, _, files in os.walk(tmp_path):
+        if "SUCCESS" in files:
+            found.append(root)
+    assert found, f"no SUCCESS manifest under {tmp_path}"
+
+
+def test_local_scheduler_lru_eviction(tmp_path):
+    from torchx_amd.schedulers.local_scheduler import LocalScheduler
+    from torchx_amd.specs import AppDef, Resource, Role
+
+    s = LocalScheduler("t", cache_size=2)
+    handles = []
+    for i in range(3):
+        app = AppDef(name=f"e{i}", roles=[
+            Role(name="r", image="/", entrypoint="true", num_replicas=1,
+                 resource=Resource(cpu=1, gpu=0, memMB=32)),
+        ])
+        info = s.submit_dryrun(
+            app, {"log_dir": str(tmp_path),
+                  "auto_set_hip_visible_devices": False})
+        handles.append(s.schedule(info))
+    import time
+
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        live = [s.describe(h) for h in handles]
+        if all(d is None or d.state == AppState.SUCCEEDED for d in live):
+            break
+        time.sleep(0.2)
+    # submitting one more evicts the oldest finished app beyond cache_size
+    app = AppDef(name="e3", roles=[
+        Role(name="r", image="/", entrypoint="true", num_replicas=1,
+             resource=Resource(cpu=1, gpu=0, memMB=32)),
+    ])
+    info = s.submit_dryrun(
+        app, {"log_dir": str(tmp_path), "auto_set_hip_visible_devices": False})
+    s.schedule(info)
+    assert s.describe(handles[0]) is None  # evicted
+    s.close()
