@@ -1,0 +1,96 @@
+"""Unit tests: threaded log merge, in-job distributed helpers (trivial pg),
+tracker lineage, and the from_yaml component."""
+
+import io
+import os
+
+import pytest
+import torch.distributed as dist
+
+
+class TestLogTee:
+    def test_threaded_merge_prefixes(self):
+        from torchx_amd.utils.log_tee import print_log_lines
+
+        out = io.StringIO()
+        lines = {("train", 0): ["a1", "a2"], ("train", 1): ["b1"],
+                 ("ps", 0): ["c1"]}
+
+        print_log_lines(
+            list(lines), lambda r, k: lines[(r, k)], stream=out,
+            colored=False,
+        )
+        text = out.getvalue().splitlines()
+        assert sorted(text) == sorted(
+            ["train/0 a1", "train/0 a2", "train/1 b1", "ps/0 c1"]
+        )
+
+    def test_error_propagates(self):
+        from torchx_amd.utils.log_tee import print_log_lines
+
+        def boom(r, k):
+            raise RuntimeError("pull failed")
+
+        with pytest.raises(RuntimeError):
+            print_log_lines([("r", 0)], boom, stream=io.StringIO(),
+                            colored=False)
+
+
+class TestDistributedHelpers:
+    def test_trivial_pg_when_not_launched(self, monkeypatch):
+        # without RANK/WORLD_SIZE, init_pg makes a world-1 group
+        for var in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+                    "MASTER_PORT", "TORCHELASTIC_RUN_ID"):
+            monkeypatch.delenv(var, raising=False)
+        from torchx_amd.distributed import init_pg, rank, world_size
+
+        if dist.is_initialized():
+            dist.destroy_process_group()
+        init_pg()
+        try:
+            assert world_size() == 1
+            assert rank() == 0
+        finally:
+            if dist.is_initialized():
+                dist.destroy_process_group()
+
+    def test_rank_helpers_from_env(self, monkeypatch):
+        monkeypatch.setenv("RANK", "3")
+        monkeypatch.setenv("WORLD_SIZE", "8")
+        monkeypatch.setenv("LOCAL_RANK", "1")
+        from torchx_amd.distributed import local_rank, rank, world_size
+
+        if not dist.is_initialized():
+            assert rank() == 3
+            assert world_size() == 8
+            assert local_rank() == 1
+
+    def test_on_rank0_first_single(self, monkeypatch):
+        for var in ("RANK", "WORLD_SIZE"):
+            monkeypatch.delenv(var, raising=False)
+        from torchx_amd.distributed import on_rank0_first
+
+        order = []
+        with on_rank0_first():
+            order.append("inside")
+        assert order == ["inside"]
+
+
+class TestTrackerLineage:
+    def test_descendants_and_lineage(self, tmp_path):
+        from torchx_amd.tracker.fsspec import FsspecTracker
+
+        t = FsspecTracker(str(tmp_path))
+        t.add_source("child1", "parent")
+        t.add_source("child2", "parent")
+        t.add_source("grandchild", "child1")
+        assert set(t.lineage("parent")) >= {"child1", "child2"}
+        assert list(t.sources("grandchild")) == ["child1"]
+
+    def test_artifacts_roundtrip_metadata_types(self, tmp_path):
+        from torchx_amd.tracker.fsspec import FsspecTracker
+
+        t = FsspecTracker(str(tmp_path))
+        t.add_metadata("r", lr=0.1, steps=10, name="x", flag=True)
+        md = t.metadata("r")
+        assert md["lr"] == 0.1 and md["steps"] == 10 and md["flag"] is True

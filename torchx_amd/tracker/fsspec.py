@@ -97,6 +97,17 @@ class FsspecTracker(TrackerBase):
                     out.append(e["source"])
         return out
 
+    def lineage(self, run_id: str) -> Iterable[str]:
+        """Direct descendants of ``run_id`` (runs that declared it a
+        source) — read from the reverse edges add_source writes."""
+        d = self._dir(run_id, "descendants")
+        out = []
+        if self._fs.exists(d):
+            for p in self._fs.ls(d):
+                with fsspec.open(p, "r") as f:
+                    out.append(json.load(f)["descendant"])
+        return out
+
     def run_ids(self, **kwargs: str) -> Iterable[str]:
         if not self._fs.exists(self._root):
             return []
