@@ -185,3 +185,50 @@ class TestLocalSchedulerEdges:
         params = next(iter(info.request.role_params.values()))
         assert params[0].env.get("HIP_VISIBLE_DEVICES") == "0,1"
         assert params[1].env.get("HIP_VISIBLE_DEVICES") == "2,3"
+
+
+class TestRegistryAndRunopts:
+    def test_all_four_builtin_schedulers(self):
+        from torchx_amd.schedulers import get_scheduler_factories
+
+        factories = get_scheduler_factories()
+        for name in ("local_cwd", "local_docker", "slurm", "kubernetes"):
+            assert name in factories, factories
+
+    def test_default_scheduler_is_first(self):
+        from torchx_amd.schedulers import get_default_scheduler_name
+
+        assert get_default_scheduler_name() == "local_cwd"
+
+    def test_every_scheduler_has_runopts_help(self):
+        from torchx_amd.runner import get_runner
+
+        with get_runner("t") as runner:
+            for name in ("local_cwd", "local_docker", "slurm", "kubernetes"):
+                opts = runner.scheduler_run_opts(name)
+                assert opts is not None
+                # every option carries help text
+                for oname, opt in getattr(opts, "_opts", {}).items():
+                    assert opt.help, f"{name}.{oname} has no help"
+
+
+class TestSlurmList:
+    def test_list_merges_hetjob_groups(self):
+        from unittest.mock import patch as _patch
+
+        from torchx_amd.schedulers.slurm_scheduler import SlurmScheduler
+
+        payload = {"jobs": [
+            {"job_id": 300, "het_job_id": 300, "het_job_offset": 0,
+             "name": "t-0", "job_state": ["RUNNING"]},
+            {"job_id": 301, "het_job_id": 300, "het_job_offset": 1,
+             "name": "t-1", "job_state": ["RUNNING"]},
+        ]}
+        import json as _json
+
+        with _patch("subprocess.check_output",
+                    return_value=_json.dumps(payload).encode()):
+            apps = SlurmScheduler("t").list()
+        # two het groups of one job -> ONE listed app
+        assert len(apps) == 1
+        assert apps[0].app_id == "300"
