@@ -252,3 +252,37 @@ def test_k8s_pod_name_sanitized():
 
     assert sanitize_for_k8s("My_App.Name") == "my-app-name"
     assert len(sanitize_for_k8s("x" * 100)) == 63
+
+
+def test_docker_cancel_and_log_iter_mocked():
+    from unittest.mock import MagicMock
+
+    from torchx_amd.schedulers.docker_scheduler import DockerScheduler
+
+    client = MagicMock()
+    c = MagicMock()
+    c.labels = {"torchx.ai/app-id": "app1", "torchx.ai/role": "train",
+                "torchx.ai/replica": "0"}
+    c.logs.return_value = b"alpha\nbeta\n"
+    client.containers.list.return_value = [c]
+    s = DockerScheduler("t", client=client)
+    lines = list(s.log_iter("app1", "train", 0))
+    assert lines == ["alpha", "beta"]
+    s._cancel_existing("app1")
+    c.stop.assert_called_once()
+
+
+def test_docker_list_mocked():
+    from unittest.mock import MagicMock, patch
+
+    from torchx_amd.schedulers.docker_scheduler import DockerScheduler
+    from torchx_amd.specs import AppState
+
+    client = MagicMock()
+    c = MagicMock()
+    c.labels = {"torchx.ai/app-id": "appZ"}
+    client.containers.list.return_value = [c]
+    s = DockerScheduler("t", client=client)
+    with patch.object(DockerScheduler, "describe", return_value=None):
+        apps = s.list()
+    assert [a.app_id for a in apps] == ["appZ"]
