@@ -268,3 +268,33 @@ def test_local_scheduler_lru_eviction(tmp_path):
     s.schedule(info)
     assert s.describe(handles[0]) is None  # evicted
     s.close()
+
+
+def test_integ_matrix_all_cheap_components(tmp_path):
+    """Reference-style integration matrix: dryrun-print then actually run
+    every cheap builtin through local_cwd (parity:
+    torchx/components/integration_tests/integ_tests.py)."""
+    providers = [
+        ("utils.echo", ["--msg", "integ-echo"]),
+        ("utils.touch", ["--file", str(tmp_path / "touched.txt")]),
+        ("utils.sh", ["echo", "integ-sh"]),
+        ("utils.python", ["-c", "print('integ-python')"]),
+        ("utils.booth", ["--x1", "1.0", "--x2", "3.0",
+                         "--tracker_base", str(tmp_path / "booth")]),
+    ]
+    with get_runner("integ") as runner:
+        for name, args in providers:
+            info = runner.dryrun_component(
+                name, args, scheduler="local_cwd",
+                cfg={"log_dir": str(tmp_path),
+                     "auto_set_hip_visible_devices": False},
+            )
+            assert info.request is not None, name
+            handle = runner.run_component(
+                name, args, scheduler="local_cwd",
+                cfg={"log_dir": str(tmp_path),
+                     "auto_set_hip_visible_devices": False},
+            )
+            status = _wait(runner, handle, timeout=120)
+            assert status.state == AppState.SUCCEEDED, (name, status)
+    assert (tmp_path / "touched.txt").exists()
