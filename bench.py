@@ -32,6 +32,10 @@ def main() -> int:
                    choices=["llama3_8b", "llama3_8b_small", "tiny",
                             "mixtral8x7b", "mixtral_small"])
     p.add_argument("--device", type=str, default=None)
+    p.add_argument("--dtype", type=str, default="bf16",
+                   choices=["bf16", "fp8"],
+                   help="fp8: OCP e4m3/e5m2 GEMMs via hipBLASLt scaled-mm "
+                        "(opt-in; the headline metric is bf16)")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -105,6 +109,10 @@ def main() -> int:
         ddp = FlatDDP(flat, local_groups={"expert"})
     else:
         model = LlamaModel(cfg, device=device)
+        if args.dtype == "fp8":
+            from torchx_amd.parallel import convert_to_fp8
+
+            convert_to_fp8(model)
         flat = FlatParams(model, device)
         ddp = FlatDDP(flat)
     opt = FlatAdamW(flat, lr=3e-4)
@@ -168,7 +176,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": args.model,

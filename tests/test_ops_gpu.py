@@ -248,3 +248,64 @@ def test_attention_fwd_long_seq_8wave(dev):
     ref = reference.attention(q.float(), k.float(), v.float(), causal=True,
                               scale=D ** -0.5)
     assert rel_err(out, ref) < 3e-2
+
+
+def test_fp8_linear_numerics(dev):
+    """Fp8Linear fwd/bwd vs bf16 nn.Linear within fp8 quantization error."""
+    import torch.nn as nn
+
+    from torchx_amd.parallel.fp8 import Fp8Linear
+
+    torch.manual_seed(9)
+    M, K, N = 256, 2048, 4096
+    lin = nn.Linear(K, N, bias=False, dtype=torch.bfloat16).to(dev)
+    f8 = Fp8Linear(K, N, weight=lin.weight)
+    x = torch.randn(M, K, device=dev, dtype=torch.bfloat16,
+                    requires_grad=True)
+    x2 = x.detach().clone().requires_grad_(True)
+
+    y8 = f8(x)
+    y = lin(x2)
+    assert rel_err(y8, y.float()) < 8e-2
+
+    g = torch.randn_like(y)
+    y8.backward(g)
+    y.backward(g)
+    assert rel_err(x.grad, x2.grad.float()) < 1.5e-1
+    # weight grad accumulated on the SHARED parameter by both backwards;
+    # compare halves via fresh run instead
+    lin.weight.grad = None
+    x3 = x.detach().clone().requires_grad_(True)
+    f8(x3).backward(g)
+    dw8 = lin.weight.grad.clone()
+    lin.weight.grad = None
+    x4 = x.detach().clone().requires_grad_(True)
+    lin(x4).backward(g)
+    assert rel_err(dw8, lin.weight.grad.float()) < 1.5e-1
+
+
+def test_fp8_model_step(dev):
+    """Full llama gpu_tiny step with fp8 linears converges."""
+    from torchx_amd.models.llama import LlamaModel, llama_gpu_tiny
+    from torchx_amd.parallel import (
+        FlatAdamW, FlatDDP, FlatParams, convert_to_fp8,
+    )
+
+    cfg = llama_gpu_tiny()
+    model = LlamaModel(cfg, device=dev)
+    convert_to_fp8(model)
+    flat = FlatParams(model, dev)
+    ddp = FlatDDP(flat)
+    opt = FlatAdamW(flat, lr=1e-3)
+    torch.manual_seed(7)
+    tokens = torch.randint(0, cfg.vocab_size, (2, 128), device=dev)
+    targets = torch.roll(tokens, -1, 1)
+    losses = []
+    for _ in range(8):
+        opt.zero_grad()
+        loss = model(tokens, targets)
+        loss.backward()
+        ddp.finish()
+        opt.step()
+        losses.append(float(loss.detach()))
+    assert losses[-1] < losses[0] - 0.5, losses

@@ -94,3 +94,31 @@ class TestTrackerLineage:
         t.add_metadata("r", lr=0.1, steps=10, name="x", flag=True)
         md = t.metadata("r")
         assert md["lr"] == 0.1 and md["steps"] == 10 and md["flag"] is True
+
+
+class TestFp8Conversion:
+    def test_convert_swaps_big_linears_shares_weight(self):
+        import torch.nn as nn
+
+        from torchx_amd.models.llama import LlamaModel, llama_gpu_tiny
+        from torchx_amd.parallel.fp8 import Fp8Linear, convert_to_fp8
+
+        model = LlamaModel(llama_gpu_tiny())
+        before = {id(p) for p in model.parameters()}
+        convert_to_fp8(model)
+        kinds = [type(m).__name__ for m in model.modules()
+                 if isinstance(m, (nn.Linear, Fp8Linear))]
+        assert "Fp8Linear" in kinds
+        after = {id(p) for p in model.parameters()}
+        assert before == after  # weights shared, optimizer-compatible
+
+    def test_cpu_forward_matches_linear(self):
+        import torch
+        import torch.nn as nn
+
+        from torchx_amd.parallel.fp8 import Fp8Linear
+
+        lin = nn.Linear(2048, 1024, bias=False, dtype=torch.bfloat16)
+        f8 = Fp8Linear(2048, 1024, weight=lin.weight)
+        x = torch.randn(4, 2048, dtype=torch.bfloat16)
+        assert torch.equal(f8(x), lin(x))  # CPU path is plain bf16
