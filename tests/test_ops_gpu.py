@@ -264,6 +264,11 @@ def test_fp8_linear_numerics(dev):
                     requires_grad=True)
     x2 = x.detach().clone().requires_grad_(True)
 
+    # delayed scaling warm-up: the first calls calibrate x/g amax
+    for _ in range(2):
+        f8(x.detach().clone().requires_grad_(True)).sum().backward()
+    lin.weight.grad = None
+
     y8 = f8(x)
     y = lin(x2)
     assert rel_err(y8, y.float()) < 8e-2

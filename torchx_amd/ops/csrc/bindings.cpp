@@ -33,6 +33,8 @@ void swiglu_gu_fwd_launch(const void*, void*, long, long, hipStream_t);
 void swiglu_gu_bwd_launch(const void*, const void*, void*, long, long,
                           hipStream_t);
 void mfma_probe_launch(const void*, const void*, void*, hipStream_t);
+void fp8_cast_transpose_launch(const void*, void*, void*, void*,
+                               const void*, long, long, int, hipStream_t);
 }
 
 namespace {
@@ -284,6 +286,27 @@ at::Tensor swiglu_gu_bwd(at::Tensor dout, at::Tensor gu) {
   return dgu;
 }
 
+// ---- fp8 fused cast+transpose ---------------------------------------------
+// x bf16 [R, C] -> (fp8 [R, C], fp8 [C, R], amax_next scalar).
+// scale divides the values (delayed scaling: pass the PREVIOUS amax-derived
+// scale; amax_next feeds the next call).
+std::vector<at::Tensor> fp8_cast_transpose(at::Tensor x, at::Tensor scale,
+                                           bool e5m2) {
+  check_bf16(x, "x");
+  check_f32(scale, "scale");
+  const long C = x.size(-1);
+  const long R = x.numel() / C;
+  TORCH_CHECK(R % 8 == 0 && C % 8 == 0, "fp8 cast: dims must be 8-aligned");
+  auto dt = e5m2 ? at::kFloat8_e5m2 : at::kFloat8_e4m3fn;
+  auto out = at::empty({R, C}, x.options().dtype(dt));
+  auto out_t = at::empty({C, R}, x.options().dtype(dt));
+  auto amax = at::zeros({1}, x.options().dtype(at::kFloat));
+  fp8_cast_transpose_launch(x.data_ptr(), out.data_ptr(), out_t.data_ptr(),
+                            amax.data_ptr(), scale.data_ptr(), R, C,
+                            e5m2 ? 1 : 0, cur_stream());
+  return {out, out_t, amax};
+}
+
 // ---- probe ----------------------------------------------------------------
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
   check_bf16(a, "a");
@@ -311,5 +334,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rope_qkv", &rope_qkv);
   m.def("swiglu_gu_fwd", &swiglu_gu_fwd);
   m.def("swiglu_gu_bwd", &swiglu_gu_bwd);
+  m.def("fp8_cast_transpose", &fp8_cast_transpose);
   m.def("mfma_probe", &mfma_probe);
 }

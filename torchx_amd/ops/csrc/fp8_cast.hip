@@ -1,0 +1,156 @@
+// Fused fp8 cast + transpose for the Fp8Linear path (gfx950).
+//
+// One read of a bf16 [R, C] tensor produces BOTH fp8 orientations —
+// row-major [R, C] and transposed [C, R] — plus the tensor's amax for the
+// NEXT call's scale (TransformerEngine-style delayed scaling).  The naive
+// torch path (amax reduce + quantize + .t().contiguous() + quantize) costs
+// ~3 full sweeps with an uncoalesced transpose and made fp8 SLOWER than
+// bf16 end to end; this kernel is one coalesced read + two 1-byte/elem
+// writes.
+//
+// 64x64 bf16 tiles staged through LDS (padded +8 to break bank conflicts
+// on the transposed reads); fp8 packing via v_cvt_pk_fp8_f32 /
+// v_cvt_pk_bf8_f32 (e4m3 / e5m2).  amax via per-thread max -> wave reduce
+// -> one atomicMax (uint trick: non-negative float bits order like uints).
+#include "common.h"
+
+#define TDIM 64
+#define TPAD 72  // 64 + 8 ushorts
+
+template <bool E5M2>
+__device__ __forceinline__ unsigned int pack4_fp8(const float* v) {
+  unsigned int r = 0;
+  if (E5M2) {
+    r = __builtin_amdgcn_cvt_pk_bf8_f32(v[0], v[1], r, false);
+    r = __builtin_amdgcn_cvt_pk_bf8_f32(v[2], v[3], r, true);
+  } else {
+    r = __builtin_amdgcn_cvt_pk_fp8_f32(v[0], v[1], r, false);
+    r = __builtin_amdgcn_cvt_pk_fp8_f32(v[2], v[3], r, true);
+  }
+  return r;
+}
+
+// 256 threads per block; each block handles one 64x64 tile:
+// load 16 rows/wave vectorized, stash to LDS, write the row-major fp8
+// image from registers, then read LDS transposed and write the [C, R]
+// image.  Guarded for edge tiles.
+template <bool E5M2>
+__global__ void __launch_bounds__(256)
+fp8_cast_transpose_kernel(
+    const unsigned short* __restrict__ x,  // bf16 [R, C]
+    unsigned char* __restrict__ out,       // fp8 [R, C]
+    unsigned char* __restrict__ out_t,     // fp8 [C, R]
+    float* __restrict__ amax_next,         // 1 elem, pre-zeroed
+    const float* __restrict__ scale_ptr,   // divisor (delayed scale)
+    long R, long C) {
+  const float inv_scale = 1.0f / scale_ptr[0];
+  __shared__ unsigned short tile[TDIM * TPAD];
+  __shared__ float red[4];
+
+  const long tiles_c = (C + TDIM - 1) / TDIM;
+  const long tr = (long)blockIdx.x / tiles_c;  // tile row
+  const long tc = (long)blockIdx.x % tiles_c;  // tile col
+  const long r0 = tr * TDIM, c0 = tc * TDIM;
+
+  const int tid = threadIdx.x;
+  float local_amax = 0.f;
+
+  // each thread loads 16 elements: 2 ushort8 chunks of one row
+  // thread -> (row = tid/4, chunk = tid%4) over 64 rows x 8 chunks of 8
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int id = tid + p * 256;       // 512 chunks = 64 rows x 8
+    const int rr = id >> 3, cc = (id & 7) * 8;
+    const long gr = r0 + rr;
+    ushort8 v = (ushort8)0;
+    if (gr < R) {
+      const long gc = c0 + cc;
+      if (gc + 8 <= C) {
+        v = *(const ushort8*)(x + gr * C + gc);
+      } else {
+        for (int j = 0; j < 8 && gc + j < C; ++j) {
+          v[j] = x[gr * C + gc + j];
+        }
+      }
+    }
+    *(ushort8*)(tile + rr * TPAD + cc) = v;
+
+    // row-major fp8 write + amax from the same registers
+    float f[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      f[j] = bf16_to_f32(v[j]);
+      local_amax = fmaxf(local_amax, fabsf(f[j]));
+      f[j] *= inv_scale;
+    }
+    if (gr < R && c0 + cc < C) {
+      unsigned int w0 = pack4_fp8<E5M2>(f);
+      unsigned int w1 = pack4_fp8<E5M2>(f + 4);
+      if (c0 + cc + 8 <= C) {
+        uint2_v w = {w0, w1};
+        *(uint2_v*)(out + gr * C + c0 + cc) = w;
+      } else {
+        unsigned char b[8];
+        *(unsigned int*)b = w0;
+        *(unsigned int*)(b + 4) = w1;
+        for (int j = 0; j < 8 && c0 + cc + j < C; ++j) {
+          out[gr * C + c0 + cc + j] = b[j];
+        }
+      }
+    }
+  }
+
+  // amax: wave -> block -> global (uint atomicMax works for >= 0 floats)
+  local_amax = wave_reduce_max(local_amax);
+  __syncthreads();
+  if ((tid & 63) == 0) red[tid >> 6] = local_amax;
+  __syncthreads();
+  if (tid == 0) {
+    float m = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+    atomicMax((unsigned int*)amax_next, __float_as_uint(m));
+  }
+  __syncthreads();
+
+  // transposed write: thread covers 2 chunks of 8 consecutive ROWS at one
+  // column -> contiguous in the [C, R] output
+  #pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int id = tid + p * 256;
+    const int cc = id >> 3, rr = (id & 7) * 8;  // out row = col cc
+    const long gc = c0 + cc;
+    if (gc >= C) continue;
+    float f[8];
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      f[j] = bf16_to_f32(tile[(rr + j) * TPAD + cc]) * inv_scale;
+    }
+    if (r0 + rr + 8 <= R) {
+      uint2_v w = {pack4_fp8<E5M2>(f), pack4_fp8<E5M2>(f + 4)};
+      *(uint2_v*)(out_t + gc * R + r0 + rr) = w;
+    } else if (r0 + rr < R) {
+      unsigned char b[8];
+      *(unsigned int*)b = pack4_fp8<E5M2>(f);
+      *(unsigned int*)(b + 4) = pack4_fp8<E5M2>(f + 4);
+      for (int j = 0; j < 8 && r0 + rr + j < R; ++j) {
+        out_t[gc * R + r0 + rr + j] = b[j];
+      }
+    }
+  }
+}
+
+extern "C" void fp8_cast_transpose_launch(
+    const void* x, void* out, void* out_t, void* amax_next,
+    const void* scale, long R, long C, int e5m2, hipStream_t stream) {
+  const long tiles = ((R + TDIM - 1) / TDIM) * ((C + TDIM - 1) / TDIM);
+  if (e5m2) {
+    hipLaunchKernelGGL((fp8_cast_transpose_kernel<true>), dim3((int)tiles),
+                       dim3(256), 0, stream, (const unsigned short*)x,
+                       (unsigned char*)out, (unsigned char*)out_t,
+                       (float*)amax_next, (const float*)scale, R, C);
+  } else {
+    hipLaunchKernelGGL((fp8_cast_transpose_kernel<false>), dim3((int)tiles),
+                       dim3(256), 0, stream, (const unsigned short*)x,
+                       (unsigned char*)out, (unsigned char*)out_t,
+                       (float*)amax_next, (const float*)scale, R, C);
+  }
+}

@@ -2,15 +2,18 @@
 
 MI355X's matrix cores run fp8 at ~2x the bf16 rate (measured 2.8 PF/s vs
 1.48 PF/s on the fat MLP shape — tools/probe_fp8.py). ``Fp8Linear`` is a
-drop-in for ``nn.Linear(bias=False)``: per-tensor dynamic scaling, e4m3
-for activations/weights, e5m2 for gradients, all three GEMMs (fwd, dgrad,
-wgrad) in fp8 via ``torch._scaled_mm``. The master weight stays a normal
-bf16 Parameter, so FlatParams/FlatDDP/FlatAdamW work unchanged — fp8 is
-purely a compute-path transform (``convert_to_fp8(model)`` before
-FlatParams).
+drop-in for ``nn.Linear(bias=False)``: all three GEMMs (fwd, dgrad, wgrad)
+run in fp8 via ``torch._scaled_mm`` — e4m3 activations/weights, e5m2
+gradients. Quantization uses the fused HIP cast+transpose kernel
+(ops/csrc/fp8_cast.hip): ONE read of the bf16 tensor yields both fp8
+orientations (``_scaled_mm`` needs a column-major B) plus the amax for the
+next step's scale (TransformerEngine-style delayed scaling — no host
+syncs; the first optimizer step sees a warm-up scale seeded at conversion).
 
-This is an OPT-IN mode (``bench.py --dtype fp8``); the headline benchmark
-stays bf16.
+The master weight stays a normal bf16 Parameter, so FlatParams / FlatDDP /
+FlatAdamW work unchanged — fp8 is purely a compute-path transform
+(``convert_to_fp8(model)`` before FlatParams). Opt-in via
+``bench.py --dtype fp8``; the headline benchmark stays bf16.
 """
 
 from __future__ import annotations
@@ -24,54 +27,49 @@ E4M3_MAX = 448.0
 E5M2_MAX = 57344.0
 
 
-def _amax_scale(t: torch.Tensor, fmax: float) -> torch.Tensor:
-    amax = torch.amax(t.abs().float())
-    return torch.clamp(amax / fmax, min=1e-12)
+def _hip():
+    from torchx_amd import ops
+
+    return ops.hip_ops(required=True)
 
 
-def _quant(t: torch.Tensor, scale: torch.Tensor,
-           dtype: torch.dtype) -> torch.Tensor:
-    return (t.float() / scale).to(dtype)
+def _scale_from_amax(amax: torch.Tensor, fmax: float) -> torch.Tensor:
+    return torch.clamp(amax / fmax, min=1e-10)
 
 
 class _Fp8Matmul(torch.autograd.Function):
-    """y = x @ W^T with all three GEMMs in fp8.
-
-    Forward saves BOTH row orientations of the fp8 operands (the
-    TransformerEngine recipe) so dgrad and wgrad feed ``_scaled_mm``
-    directly — its B operand must be column-major.
-    """
+    """y = x @ W^T, all GEMMs fp8; fused cast+transpose, delayed scaling."""
 
     @staticmethod
-    def forward(ctx, x: torch.Tensor, w: torch.Tensor):
-        # x [M, K] bf16; w [N, K] bf16
-        M, K = x.shape
-        sx = _amax_scale(x, E4M3_MAX)
-        sw = _amax_scale(w, E4M3_MAX)
-        x8 = _quant(x, sx, torch.float8_e4m3fn)            # [M, K] row
-        w8 = _quant(w, sw, torch.float8_e4m3fn)            # [N, K] row
+    def forward(ctx, x: torch.Tensor, w: torch.Tensor, mod: "Fp8Linear"):
+        hip = _hip()
+        sx = _scale_from_amax(mod.x_amax, E4M3_MAX)
+        sw = _scale_from_amax(mod.w_amax, E4M3_MAX)
+        x8, x8t, x_amax = hip.fp8_cast_transpose(x, sx, False)
+        w8, w8t, w_amax = hip.fp8_cast_transpose(w, sw, False)
+        mod.x_amax.copy_(x_amax[0])        # async: feeds the NEXT call
+        mod.w_amax.copy_(w_amax[0])
         y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
                              out_dtype=torch.bfloat16)
-        # transposed fp8 copies for backward
-        x8t = _quant(x.t().contiguous(), sx, torch.float8_e4m3fn)  # [K, M]
-        w8t = _quant(w.t().contiguous(), sw, torch.float8_e4m3fn)  # [K, N]
         ctx.save_for_backward(x8t, w8t, sx, sw)
+        ctx.mod = mod
         return y
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
         x8t, w8t, sx, sw = ctx.saved_tensors
-        dy = dy.contiguous()
-        sg = _amax_scale(dy, E5M2_MAX)
-        dy8 = _quant(dy, sg, torch.float8_e5m2)            # [M, N] row
-        dy8t = _quant(dy.t().contiguous(), sg, torch.float8_e5m2)  # [N, M]
+        mod = ctx.mod
+        hip = _hip()
+        sg = _scale_from_amax(mod.g_amax, E5M2_MAX)
+        dy8, dy8t, g_amax = hip.fp8_cast_transpose(dy.contiguous(), sg, True)
+        mod.g_amax.copy_(g_amax[0])
         # dx [M, K] = dy [M, N] @ W [N, K];  B col-major = w8t.t()
         dx = torch._scaled_mm(dy8, w8t.t(), scale_a=sg, scale_b=sw,
                               out_dtype=torch.bfloat16)
         # dW [N, K] = dy^T [N, M] @ x [M, K];  B col-major = x8t.t()
         dw = torch._scaled_mm(dy8t, x8t.t(), scale_a=sg, scale_b=sx,
                               out_dtype=torch.bfloat16)
-        return dx, dw
+        return dx, dw, None
 
 
 class Fp8Linear(nn.Module):
@@ -84,24 +82,31 @@ class Fp8Linear(nn.Module):
         self.out_features = out_features
         if weight is None:
             weight = nn.Parameter(
-                torch.empty(out_features, in_features,
-                            dtype=torch.bfloat16)
+                torch.empty(out_features, in_features, dtype=torch.bfloat16)
             )
             nn.init.normal_(weight, std=in_features ** -0.5)
         self.weight = weight
+        # delayed-scaling state: amax of the PREVIOUS step's tensors.
+        # w seeded exactly at conversion; x/g warm up over the first steps
+        # (cast saturates to the fp8 max meanwhile).
+        dev = weight.device
+        w_amax = weight.detach().abs().amax().float()
+        self.register_buffer("w_amax", w_amax.clone())
+        self.register_buffer("x_amax", torch.ones((), device=dev))
+        self.register_buffer("g_amax", torch.ones((), device=dev))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         shape = x.shape
         x2 = x.reshape(-1, self.in_features)
         if x.is_cuda:
-            y = _Fp8Matmul.apply(x2, self.weight)
+            y = _Fp8Matmul.apply(x2, self.weight, self)
         else:  # CPU CI path: plain bf16 matmul
             y = x2 @ self.weight.t()
         return y.reshape(*shape[:-1], self.out_features)
 
     def extra_repr(self) -> str:
         return (f"in_features={self.in_features}, "
-                f"out_features={self.out_features}, fp8=e4m3/e5m2")
+                f"out_features={self.out_features}, fp8=e4m3/e5m2(delayed)")
 
 
 def convert_to_fp8(module: nn.Module,
