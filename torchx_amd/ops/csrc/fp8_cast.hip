@@ -17,6 +17,15 @@
 #define TDIM 64
 #define TPAD 72  // 64 + 8 ushorts
 
+// clamp to the format's finite max BEFORE converting: overflow through
+// v_cvt_pk_*fp8* encodes NaN (e4m3fn 0x7f), and one uncalibrated
+// delayed-scaling step would poison the whole run
+template <bool E5M2>
+__device__ __forceinline__ float clamp_fp8(float v) {
+  const float m = E5M2 ? 57344.0f : 448.0f;
+  return fminf(fmaxf(v, -m), m);
+}
+
 template <bool E5M2>
 __device__ __forceinline__ unsigned int pack4_fp8(const float* v) {
   unsigned int r = 0;
@@ -81,7 +90,7 @@ fp8_cast_transpose_kernel(
     for (int j = 0; j < 8; ++j) {
       f[j] = bf16_to_f32(v[j]);
       local_amax = fmaxf(local_amax, fabsf(f[j]));
-      f[j] *= inv_scale;
+      f[j] = clamp_fp8<E5M2>(f[j] * inv_scale);
     }
     if (gr < R && c0 + cc < C) {
       unsigned int w0 = pack4_fp8<E5M2>(f);
@@ -122,7 +131,8 @@ fp8_cast_transpose_kernel(
     float f[8];
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      f[j] = bf16_to_f32(tile[(rr + j) * TPAD + cc]) * inv_scale;
+      f[j] = clamp_fp8<E5M2>(
+          bf16_to_f32(tile[(rr + j) * TPAD + cc]) * inv_scale);
     }
     if (r0 + rr + 8 <= R) {
       uint2_v w = {pack4_fp8<E5M2>(f), pack4_fp8<E5M2>(f + 4)};
