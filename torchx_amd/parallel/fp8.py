@@ -3,11 +3,10 @@
 MI355X's matrix cores run fp8 at ~2x the bf16 rate (measured 2.8 PF/s vs
 1.48 PF/s on the fat MLP shape — tools/probe_fp8.py). ``Fp8Linear`` is a
 drop-in for ``nn.Linear(bias=False)``: all three GEMMs (fwd, dgrad, wgrad)
-— the FORWARD GEMM runs in e4m3 via ``torch._scaled_mm``; the backward
-dgrad/wgrad GEMMs stay bf16 (full-fp8 backward through this hipBLASLt
-build's scaled-mm produced NaNs with a transposed column-major B operand
-— see ROADMAP; the fused cast kernel already produces both orientations
-for when that path is fixed). Quantization uses the fused HIP cast+transpose kernel
+— all three GEMMs (fwd, dgrad, wgrad) run in e4m3 via
+``torch._scaled_mm`` (the cast kernel clamps to the finite fp8 max first:
+hardware overflow encodes NaN, which poisoned delayed scaling until
+calibration). Quantization uses the fused HIP cast+transpose kernel
 (ops/csrc/fp8_cast.hip): ONE read of the bf16 tensor yields both fp8
 orientations (``_scaled_mm`` needs a column-major B) plus the amax for the
 next step's scale (TransformerEngine-style delayed scaling — no host
@@ -41,28 +40,38 @@ def _scale_from_amax(amax: torch.Tensor, fmax: float) -> torch.Tensor:
 
 
 class _Fp8Matmul(torch.autograd.Function):
-    """y = x @ W^T with an e4m3 forward GEMM; bf16 backward."""
+    """y = x @ W^T, all GEMMs e4m3; fused cast+transpose, delayed scaling."""
 
     @staticmethod
     def forward(ctx, x: torch.Tensor, w: torch.Tensor, mod: "Fp8Linear"):
         hip = _hip()
         sx = _scale_from_amax(mod.x_amax, E4M3_MAX)
         sw = _scale_from_amax(mod.w_amax, E4M3_MAX)
-        x8, _x8t, x_amax = hip.fp8_cast_transpose(x, sx, False)
-        w8, _w8t, w_amax = hip.fp8_cast_transpose(w, sw, False)
+        x8, x8t, x_amax = hip.fp8_cast_transpose(x, sx, False)
+        w8, w8t, w_amax = hip.fp8_cast_transpose(w, sw, False)
         mod.x_amax.copy_(x_amax[0])        # async: feeds the NEXT call
         mod.w_amax.copy_(w_amax[0])
         y = torch._scaled_mm(x8, w8.t(), scale_a=sx, scale_b=sw,
                              out_dtype=torch.bfloat16)
-        ctx.save_for_backward(x, w)
+        ctx.save_for_backward(x8t, w8t, sx, sw)
+        ctx.mod = mod
         return y
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
-        x, w = ctx.saved_tensors
-        dy = dy.contiguous()
-        dx = dy @ w            # [M, N] @ [N, K] bf16
-        dw = dy.t() @ x        # [N, M] @ [M, K] bf16
+        x8t, w8t, sx, sw = ctx.saved_tensors
+        mod = ctx.mod
+        hip = _hip()
+        sg = _scale_from_amax(mod.g_amax, E4M3_MAX)
+        dy8, dy8t, g_amax = hip.fp8_cast_transpose(dy.contiguous(), sg,
+                                                   False)
+        mod.g_amax.copy_(g_amax[0])
+        # dx [M, K] = dy [M, N] @ W [N, K];  B col-major = w8t.t()
+        dx = torch._scaled_mm(dy8, w8t.t(), scale_a=sg, scale_b=sw,
+                              out_dtype=torch.bfloat16)
+        # dW [N, K] = dy^T [N, M] @ x [M, K];  B col-major = x8t.t()
+        dw = torch._scaled_mm(dy8t, x8t.t(), scale_a=sg, scale_b=sx,
+                              out_dtype=torch.bfloat16)
         return dx, dw, None
 
 
