@@ -87,6 +87,9 @@ def ddp(
     """
     if (script is None) == (m is None):
         raise ValueError("exactly one of --script / -m must be set")
+    # allow the conventional `--` separator before script args
+    if script_args and script_args[0] == "--":
+        script_args = script_args[1:]
 
     min_n, max_n, nproc, nnodes_spec = parse_nnodes(j)
 
