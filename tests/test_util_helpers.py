@@ -122,3 +122,52 @@ class TestFp8Conversion:
         f8 = Fp8Linear(2048, 1024, weight=lin.weight)
         x = torch.randn(4, 2048, dtype=torch.bfloat16)
         assert torch.equal(f8(x), lin(x))  # CPU path is plain bf16
+
+
+class TestCopyApp:
+    def test_copy_main_fsspec(self, tmp_path):
+        from torchx_amd.apps import copy_main
+
+        src = tmp_path / "a.txt"
+        src.write_text("payload-123")
+        dst = tmp_path / "out" / "b.txt"
+        import sys
+
+        old = sys.argv
+        sys.argv = ["copy", "--src", str(src), "--dst", str(dst)]
+        try:
+            rc = copy_main.main()
+        finally:
+            sys.argv = old
+        assert rc in (0, None)
+        assert dst.read_text() == "payload-123"
+
+
+class TestTmpDirWorkspace:
+    def test_tmpdir_mixin_copies_and_sets_image(self, tmp_path):
+        from torchx_amd.specs import AppDef, Role
+        from torchx_amd.workspace.dir_workspace import TmpDirWorkspaceMixin
+
+        class WS(TmpDirWorkspaceMixin):
+            pass
+
+        (tmp_path / "code.py").write_text("x = 1")
+        role = Role(name="r", image="ignored", entrypoint="python")
+        app = AppDef(name="a", roles=[role])
+        WS().build_workspaces(app, str(tmp_path), {})
+        import os
+
+        assert os.path.isfile(os.path.join(role.image, "code.py"))
+
+
+class TestAppStatusAggregation:
+    def test_state_precedence(self):
+        from torchx_amd.schedulers.slurm_scheduler import _aggregate
+        from torchx_amd.specs import AppState
+
+        assert _aggregate([AppState.SUCCEEDED, AppState.RUNNING]) == \
+            AppState.RUNNING
+        assert _aggregate([AppState.SUCCEEDED, AppState.FAILED]) == \
+            AppState.FAILED
+        assert _aggregate([AppState.SUCCEEDED, AppState.SUCCEEDED]) == \
+            AppState.SUCCEEDED
