@@ -5,18 +5,30 @@
 // Layouts are BSHD: q [B,S,Hq,128], k/v [B,S,Hkv,128], all bf16; lse/delta
 // [B,Hq,S] f32.
 //
-// Design notes (see /opt/skills/guides/cdna_hip_programming.md Appendix B):
+// Design (the measured ladder is in profiles/README.md; idioms follow the
+// CDNA4 guide's attention appendix):
 //  * "swapped QK^T": S^T = mfma(A=K, B=Q) so each lane owns ONE q row's
-//    scores (C-layout col = lane&31) -> softmax is in-register (16 regs +
-//    one shfl_xor(32) partner exchange), no cross-lane LDS reduction.
-//  * P (f32, C-layout) -> bf16 MFMA fragments via pack + lane-half exchange
-//    (the guide's cvt_pk + permlane32_swap idiom; v1 uses shfl_xor).
-//  * PV uses A=V^T (from a transposed LDS image) so the O accumulator stays
-//    in the lane-owns-q-column layout: per-row rescale/divide are per-lane
-//    scalars.
-//  * Backward is FA2-style: preprocess delta = rowsum(dO*O); dQ kernel
-//    (grid over q tiles, loop kv); dK/dV kernel (grid over kv tiles of 128
-//    rows, loop q tiles x GQA group) -> no atomics anywhere.
+//    scores -> softmax fully in-register (exp2 domain, v_cvt_pk_bf16_f32
+//    fragment packing, defer-max rescale skip), one shfl_xor(32) partner
+//    exchange, no cross-lane LDS reduction.
+//  * KV tiles of 64 rows; K/V arrive by async global_load_lds into
+//    XOR-swizzled natural images; the transposed operand images (V^T /
+//    Q^T / dO^T / K^T) use a granule-ROTATED placement (rot(d) =
+//    (d>>1 ^ d>>4) & 7: reads conflict-free, staging writes ~4-way — a
+//    plain layout measured 36% of wave cycles in LDS bank conflicts) and
+//    are built WAVE-LOCALLY from the just-landed naturals (each wave
+//    transposes the rows it staged, so its own vmcnt(0) orders the read
+//    and ONE barrier per tile publishes everything).
+//  * T1 XCD swizzle: blocks streaming the same (batch, kv-head) tensors
+//    land on one XCD so the stream is L2-resident (the kernels are
+//    otherwise HBM-bound at ~128 FLOP/B).
+//  * Forward block size is S-dependent: 4 waves below S=8192, 8 waves at
+//    or above (staging amortizes over 2x compute once causal wave-skew
+//    is relatively small).
+//  * Backward is FA2-style without atomics: preprocess delta =
+//    rowsum(dO*O); a dQ kernel (grid over q tiles); ONE combined 8-wave
+//    dK+dV kernel where waves w and w+4 own the same 32 kv rows (dV / dK
+//    roles) so Q/dO are staged once for both outputs.
 //
 // MFMA fragment maps used (verified against rocm CK xdlops_gemm.hpp and the
 // CDNA4 guide; C/D map from the guide):
