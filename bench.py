@@ -97,6 +97,17 @@ def main() -> int:
         # expert count (BASELINE config 5: EP all-to-all over xGMI);
         # expert params are EP-sharded -> excluded from the DP all-reduce
         ep_size = world if (world > 1 and cfg.num_experts % world == 0) else 1
+        if use_gpu and args.model == "mixtral8x7b":
+            # optimizer states are 16 B/param (bf16 p+g, fp32 p32/m/v):
+            # 47B params need ~750 GB total -> at least EP4 on 288 GB GPUs
+            params_b = 47e9 / max(ep_size, 1) + 2e9
+            need_gb = params_b * 16 / 1e9
+            if need_gb > 250:
+                print(f"mixtral8x7b needs ~{need_gb:.0f} GB/GPU for "
+                      f"optimizer state at ep_size={ep_size}; run with >=4 "
+                      f"GPUs (EP-sharded) or use --model mixtral_small",
+                      file=sys.stderr)
+                return 2
         model = MixtralModel(cfg, device=device, ep_group=None,
                              ep_size=ep_size, ep_rank=rank % ep_size)
 
