@@ -89,13 +89,15 @@ def _entry_point_modules() -> Dict[str, str]:
         from importlib.metadata import entry_points
 
         eps = entry_points()
-        group = (
-            eps.select(group="torchx_amd.components")
-            if hasattr(eps, "select")
-            else eps.get("torchx_amd.components", [])
-        )
-        for ep in group:
-            out[ep.name] = ep.value
+        # legacy reference group kept working for drop-in compatibility
+        for gname in ("torchx_amd.components", "torchx.components"):
+            group = (
+                eps.select(group=gname)
+                if hasattr(eps, "select")
+                else eps.get(gname, [])
+            )
+            for ep in group:
+                out.setdefault(ep.name, ep.value)
     except Exception:  # noqa: BLE001
         pass
     return out
