@@ -339,6 +339,9 @@ class LocalScheduler(Scheduler[PopenRequest]):
             start_new_session=True,
             cwd=p.cwd,
         )
+        # the child owns the fds now; drop the parent-side objects
+        stdout_f.close()
+        stderr_f.close()
         tee = Tee(combined_f, p.stdout, p.stderr)
         return _Replica(role, idx, proc, p, tee)
 

@@ -48,5 +48,6 @@ class Tee:
         with self._lock:
             try:
                 self._dst.flush()
+                self._dst.close()
             except ValueError:
                 pass
