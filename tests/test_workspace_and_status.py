@@ -170,3 +170,14 @@ class TestRunopts:
         # unknown keys pass through (reference parity: resolve starts
         # from {**cfg}; schedulers may accept extras)
         assert o.resolve({"extra": 1})["extra"] == 1
+
+
+class TestRunoptsBoolVocabulary:
+    def test_ini_bool_vocabulary(self):
+        # reference parity: INI bool vocabulary (specs/api.py:1079-1325)
+        o = runopts()
+        o.add("b", type_=bool, default=False, help="h")
+        for v in ("True", "true", "1", "yes"):
+            assert o.cfg_from_str(f"b={v}")["b"] is True, v
+        for v in ("False", "false", "0", "no"):
+            assert o.cfg_from_str(f"b={v}")["b"] is False, v
