@@ -20,6 +20,7 @@ FlatAdamW work unchanged — fp8 is purely a compute-path transform
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -27,6 +28,12 @@ import torch.nn as nn
 
 E4M3_MAX = 448.0
 E5M2_MAX = 57344.0
+
+# experiment knob for round 2: e5m2 gradients (wider range, less mantissa);
+# the mixed e5m2 x e4m3 scaled-mm path needs re-validation now that the
+# overflow-NaN poisoning is fixed
+_E5M2_GRADS = os.environ.get("TORCHX_AMD_FP8_E5M2_GRADS", "0") == "1"
+_GRAD_MAX = E5M2_MAX if _E5M2_GRADS else E4M3_MAX
 
 
 def _hip():
@@ -62,9 +69,9 @@ class _Fp8Matmul(torch.autograd.Function):
         x8t, w8t, sx, sw = ctx.saved_tensors
         mod = ctx.mod
         hip = _hip()
-        sg = _scale_from_amax(mod.g_amax, E4M3_MAX)
+        sg = _scale_from_amax(mod.g_amax, _GRAD_MAX)
         dy8, dy8t, g_amax = hip.fp8_cast_transpose(dy.contiguous(), sg,
-                                                   False)
+                                                   _E5M2_GRADS)
         mod.g_amax.copy_(g_amax[0])
         # dx [M, K] = dy [M, N] @ W [N, K];  B col-major = w8t.t()
         dx = torch._scaled_mm(dy8, w8t.t(), scale_a=sg, scale_b=sw,
