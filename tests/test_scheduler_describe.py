@@ -66,6 +66,105 @@ class TestSlurmDescribe:
                    return_value=json.dumps(payload).encode()):
             assert _slurm().describe("1").state == AppState.CANCELLED
 
+    def test_squeue_hetjob_multi_role(self):
+        # hetjob: trainer + reader roles as het components of one job id
+        payload = {"jobs": [
+            {"job_id": 300, "name": "trainer-0",
+             "job_state": ["RUNNING"], "nodes": "n1"},
+            {"job_id": 300, "het_job_offset": 1, "name": "trainer-1",
+             "job_state": ["RUNNING"], "nodes": "n2"},
+            {"job_id": 300, "het_job_offset": 2, "name": "reader-0",
+             "job_state": ["RUNNING"], "nodes": "n3"},
+        ]}
+        with patch("subprocess.check_output",
+                   return_value=json.dumps(payload).encode()):
+            resp = _slurm().describe("300")
+        assert resp.state == AppState.RUNNING
+        by_role = {r.role: r for r in resp.roles_statuses}
+        assert set(by_role) == {"trainer", "reader"}
+        assert len(by_role["trainer"].replicas) == 2
+        assert by_role["reader"].replicas[0].hostname == "n3"
+
+    def test_squeue_terminal_state_vocabulary(self):
+        # the long tail of slurm terminal states maps to FAILED
+        for slurm_state in ("TIMEOUT", "NODE_FAIL", "OUT_OF_MEMORY",
+                            "PREEMPTED", "BOOT_FAIL", "DEADLINE"):
+            payload = {"jobs": [
+                {"job_id": 1, "name": "w-0", "job_state": [slurm_state]},
+            ]}
+            with patch("subprocess.check_output",
+                       return_value=json.dumps(payload).encode()):
+                assert _slurm().describe("1").state == AppState.FAILED, \
+                    slurm_state
+
+    def test_squeue_requeued_and_suspended_are_pending(self):
+        for slurm_state in ("REQUEUED", "SUSPENDED"):
+            payload = {"jobs": [
+                {"job_id": 1, "name": "w-0", "job_state": [slurm_state]},
+            ]}
+            with patch("subprocess.check_output",
+                       return_value=json.dumps(payload).encode()):
+                assert _slurm().describe("1").state == AppState.PENDING
+
+    def test_sacct_cancelled_by_user_suffix(self):
+        # sacct renders "CANCELLED by <uid>"; only the first token counts
+        sacct = (
+            "JobID|JobName|State\n"
+            "400|trainer-0|CANCELLED by 1000\n"
+        )
+
+        def fake_check_output(cmd, timeout=None):
+            if cmd[0] == "squeue":
+                raise FileNotFoundError
+            return sacct.encode()
+
+        with patch("subprocess.check_output", side_effect=fake_check_output):
+            assert _slurm().describe("400").state == AppState.CANCELLED
+
+    def test_sacct_skips_job_steps_and_other_jobs(self):
+        # .batch/.extern steps and unrelated job ids must not contribute
+        sacct = (
+            "JobID|JobName|State\n"
+            "500|trainer-0|COMPLETED\n"
+            "500.batch|batch|COMPLETED\n"
+            "500.extern|extern|COMPLETED\n"
+            "5001|other-0|FAILED\n"
+        )
+
+        def fake_check_output(cmd, timeout=None):
+            if cmd[0] == "squeue":
+                raise FileNotFoundError
+            return sacct.encode()
+
+        with patch("subprocess.check_output", side_effect=fake_check_output):
+            resp = _slurm().describe("500")
+        assert resp.state == AppState.SUCCEEDED
+        assert len(resp.roles_statuses) == 1
+        assert len(resp.roles_statuses[0].replicas) == 1
+
+    def test_unknown_job_everywhere_returns_none(self):
+        def fake_check_output(cmd, timeout=None):
+            if cmd[0] == "squeue":
+                return json.dumps({"jobs": []}).encode()
+            return b"JobID|JobName|State\n"
+
+        with patch("subprocess.check_output", side_effect=fake_check_output):
+            assert _slurm().describe("999") is None
+
+    def test_list_groups_het_components(self):
+        payload = {"jobs": [
+            {"job_id": 700, "het_job_id": 700, "het_job_offset": 0,
+             "name": "w-0", "job_state": ["RUNNING"]},
+            {"job_id": 701, "het_job_id": 700, "het_job_offset": 1,
+             "name": "w-1", "job_state": ["RUNNING"]},
+            {"job_id": 800, "name": "x-0", "job_state": "PENDING"},
+        ]}
+        with patch("subprocess.check_output",
+                   return_value=json.dumps(payload).encode()):
+            apps = _slurm().list()
+        got = {a.app_id: a.state for a in apps}
+        assert got == {"700": AppState.RUNNING, "800": AppState.PENDING}
+
 
 class TestK8sDescribe:
     def _sched(self, status):
@@ -95,6 +194,46 @@ class TestK8sDescribe:
     def test_failed(self):
         s = self._sched({"state": {"phase": "Failed"}})
         assert s.describe("default:app-x").state == AppState.FAILED
+
+    def test_volcano_phase_vocabulary(self):
+        # full Volcano phase map, incl. the transitional states
+        cases = {
+            "Pending": AppState.PENDING,
+            "Inqueue": AppState.PENDING,
+            "Aborting": AppState.RUNNING,
+            "Restarting": AppState.RUNNING,
+            "Completing": AppState.RUNNING,
+            "Terminating": AppState.RUNNING,
+            "Aborted": AppState.CANCELLED,
+            "Terminated": AppState.FAILED,
+            "SomethingNew": AppState.UNKNOWN,
+        }
+        for phase, expected in cases.items():
+            s = self._sched({"state": {"phase": phase}})
+            assert s.describe("default:app-x").state == expected, phase
+
+    def test_task_status_counts_become_replicas(self):
+        s = self._sched({
+            "state": {"phase": "Running"},
+            "taskStatusCount": {
+                "trainer-0": {"phase": {"Running": 1}},
+                "trainer-1": {"phase": {"Pending": 1}},
+                "reader-0": {"phase": {"Succeeded": 1}},
+            },
+        })
+        resp = s.describe("default:app-x")
+        by_role = {r.role: r for r in resp.roles_statuses}
+        assert set(by_role) == {"trainer", "reader"}
+        trainer_states = {r.id: r.state for r in by_role["trainer"].replicas}
+        assert trainer_states == {0: AppState.RUNNING, 1: AppState.PENDING}
+        assert by_role["reader"].replicas[0].state == AppState.SUCCEEDED
+
+    def test_no_status_block_is_unknown(self):
+        # job created but volcano has not stamped status yet
+        s = self._sched({})
+        resp = s.describe("default:app-x")
+        assert resp.state == AppState.UNKNOWN
+        assert resp.roles_statuses == []
 
     def test_missing_job_returns_none(self):
         from torchx_amd.schedulers.kubernetes_scheduler import (
