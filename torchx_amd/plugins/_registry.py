@@ -77,6 +77,9 @@ class PluginRegistry:
                         )
         except ImportError:
             pass
+        except Exception as e:  # noqa: BLE001 — a broken plugin ROOT
+            # (reference's broken_root fixture) must not crash discovery
+            self._errors.append(f"torchx_amd_plugins: {e}")
         # entry points
         try:
             from importlib.metadata import entry_points
