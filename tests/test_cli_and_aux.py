@@ -237,3 +237,58 @@ def test_cli_tracker_cmds(tmp_path, monkeypatch):
     monkeypatch.setenv("TORCHXCONFIG", str(cfg))
     assert cli_main(["tracker", "list", "jobs"]) == 0
     assert cli_main(["tracker", "list", "metadata", "--run_id", "run7"]) == 0
+
+
+def test_cli_entry_point_subcommands(monkeypatch):
+    # custom subcommand via the torchx_amd.cli.cmds entry-point group;
+    # an entry point named like a builtin ("status") overrides it
+    import importlib.metadata as md
+
+    from torchx_amd.cli import main as cli_main
+
+    calls = []
+
+    class HelloCmd:
+        """say hello"""
+
+        def add_arguments(self, parser):
+            parser.add_argument("--who", default="world")
+
+        def run(self, args):
+            calls.append(("hello", args.who))
+            return 0
+
+    class StatusOverride:
+        def add_arguments(self, parser):
+            parser.add_argument("extra")
+
+        def run(self, args):
+            calls.append(("status", args.extra))
+            return 0
+
+    class FakeEp:
+        def __init__(self, name, obj):
+            self.name = name
+            self._obj = obj
+
+        def load(self):
+            return self._obj
+
+    class BrokenEp:
+        name = "broken"
+
+        def load(self):
+            raise RuntimeError("cli-plugin-boom")
+
+    class FakeEps:
+        def select(self, group):
+            if group == "torchx_amd.cli.cmds":
+                return [FakeEp("hello", HelloCmd),
+                        FakeEp("status", StatusOverride),
+                        BrokenEp()]
+            return []
+
+    monkeypatch.setattr(md, "entry_points", lambda: FakeEps())
+    assert cli_main.main(["hello", "--who", "mi355x"]) == 0
+    assert cli_main.main(["status", "anything"]) == 0
+    assert calls == [("hello", "mi355x"), ("status", "anything")]

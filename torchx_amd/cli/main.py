@@ -4,6 +4,7 @@ cancel, configure, delete, describe, list, log, runopts, status, tracker)."""
 from __future__ import annotations
 
 import argparse
+import inspect
 import json
 import logging
 import os
@@ -352,7 +353,43 @@ def create_parser() -> argparse.ArgumentParser:
     p.add_argument("--run_id", type=str, default=None)
     p.set_defaults(func=cmd_tracker)
 
+    _add_entry_point_cmds(sub)
+
     return parser
+
+
+def _add_entry_point_cmds(sub: argparse._SubParsersAction) -> None:
+    """Custom/override subcommands from entry points (reference parity:
+    torchx/cli/main.py:64 group ``torchx.cli.cmds``). Each entry point
+    loads to a SubCommand-style object with ``add_arguments(parser)`` and
+    ``run(args)``; an entry point named like a builtin replaces it."""
+    try:
+        from importlib.metadata import entry_points
+
+        eps = entry_points()
+    except Exception:  # noqa: BLE001
+        return
+    for group in ("torchx_amd.cli.cmds", "torchx.cli.cmds"):
+        found = (eps.select(group=group) if hasattr(eps, "select")
+                 else eps.get(group, []))
+        for ep in found:
+            try:
+                obj = ep.load()
+                cmd = obj() if isinstance(obj, type) else obj
+                # override: drop a builtin parser of the same name
+                sub._name_parser_map.pop(ep.name, None)
+                p = sub.add_parser(
+                    ep.name,
+                    help=(inspect.getdoc(cmd) or "").split("\n")[0] or None,
+                )
+                add_args = getattr(cmd, "add_arguments", None)
+                if add_args:
+                    add_args(p)
+                p.set_defaults(func=cmd.run)
+            except Exception as e:  # noqa: BLE001 — a broken CLI plugin
+                # must not take down the whole CLI
+                logger.warning("skipping CLI plugin %s:%s: %s",
+                               group, ep.name, e)
 
 
 def main(argv: Optional[List[str]] = None) -> int:
