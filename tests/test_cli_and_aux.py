@@ -223,7 +223,37 @@ def test_cli_configure_dumps_template(tmp_path, monkeypatch):
     assert cli_main(["configure", "-s", "local_cwd"]) == 0
     cfg = tmp_path / ".torchxconfig"
     assert cfg.exists()
-    assert "[local_cwd]" in cfg.read_text()
+    text = cfg.read_text()
+    assert "[local_cwd]" in text
+    # generated from the scheduler's ACTUAL runopts, not a static template
+    assert "log_dir" in text
+    assert "auto_set_hip_visible_devices" in text
+    assert "[kubernetes]" not in text  # -s filter respected
+
+
+def test_configure_all_schedulers_marks_required(tmp_path):
+    from torchx_amd.runner import config as torchx_config
+
+    path = str(tmp_path / "cfg.ini")
+    torchx_config.dump(path)
+    text = open(path).read()
+    # every registered scheduler gets a section
+    for section in ("[local_cwd]", "[local_docker]", "[slurm]",
+                    "[kubernetes]"):
+        assert section in text, section
+    # required opts surface as uncommented #FIXME lines
+    assert "queue = #FIXME" in text
+    # optional opts are commented with their default
+    assert "# namespace = default" in text
+
+
+def test_configure_unknown_scheduler_raises(tmp_path):
+    import pytest
+
+    from torchx_amd.runner import config as torchx_config
+
+    with pytest.raises(ValueError):
+        torchx_config.dump(str(tmp_path / "x.ini"), schedulers=["nope"])
 
 
 def test_cli_tracker_cmds(tmp_path, monkeypatch):

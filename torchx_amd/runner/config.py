@@ -20,20 +20,6 @@ from typing import Any, Dict, List, Mapping, Optional
 CONFIG_FILE = ".torchxconfig"
 ENV_TORCHXCONFIG = "TORCHXCONFIG"
 
-_TEMPLATE = """#
-# torchx_amd configuration (generated by `torchx configure`)
-# Uncomment and fill in the #FIXME values.
-#
-[local_cwd]
-# log_dir = #FIXME
-# auto_set_hip_visible_devices = True
-
-# [component:dist.ddp]
-# j = 1x8
-# h = mi355x.8gpu
-"""
-
-
 def find_configs(dirs: Optional[List[str]] = None) -> List[str]:
     env = os.environ.get(ENV_TORCHXCONFIG)
     if env:
@@ -93,10 +79,57 @@ def apply(scheduler: str, cfg: Dict[str, Any],
     load(scheduler, cfg, dirs)
 
 
-def dump(path: str, schedulers: Optional[List[str]] = None) -> None:
-    """Write a template .torchxconfig (``torchx configure``)."""
+def dump(path: str, schedulers: Optional[List[str]] = None,
+         required_only: bool = False) -> None:
+    """Write a .torchxconfig template generated from each scheduler's
+    actual runopts (``torchx configure``; reference config.py:222 —
+    optional opts pre-filled with defaults, required ones as #FIXME).
+    Orchestrator plugins registered under the
+    ``torchx_amd.schedulers.orchestrator`` entry-point group are
+    included alongside the built-in schedulers."""
+    from torchx_amd.schedulers import get_scheduler_factories
+
+    factories = dict(get_scheduler_factories())
+    try:
+        from importlib.metadata import entry_points
+
+        eps = entry_points()
+        found = (eps.select(group="torchx_amd.schedulers.orchestrator")
+                 if hasattr(eps, "select")
+                 else eps.get("torchx_amd.schedulers.orchestrator", []))
+        for ep in found:
+            factories.setdefault(ep.name, ep.load())
+    except Exception:  # noqa: BLE001
+        pass
+    explicit = schedulers is not None and len(schedulers) > 0
+    names = schedulers if explicit else list(factories)
+    lines = ["#", "# torchx_amd configuration (torchx configure)",
+             "# Fill in the #FIXME values; delete what you don't need.",
+             "#"]
+    for name in names:
+        factory = factories.get(name)
+        if factory is None:
+            raise ValueError(
+                f"unknown scheduler {name!r}; known: {sorted(factories)}"
+            )
+        try:
+            opts = factory("_").run_opts()
+        except Exception:  # noqa: BLE001 — a broken (plugin) scheduler
+            # must not break `torchx configure` for the others
+            if explicit:
+                raise
+            continue
+        lines.append(f"[{name}]")
+        for key, opt in opts:
+            if opt.required:
+                lines.append(f"{key} = #FIXME ({opt.opt_type.__name__}) "
+                             f"{opt.help}")
+            elif not required_only:
+                default = "" if opt.default is None else opt.default
+                lines.append(f"# {key} = {default}")
+        lines.append("")
     with open(path, "w") as f:
-        f.write(_TEMPLATE)
+        f.write("\n".join(lines) + "\n")
 
 
 def get_configured_trackers(
