@@ -298,3 +298,41 @@ def test_integ_matrix_all_cheap_components(tmp_path):
             status = _wait(runner, handle, timeout=120)
             assert status.state == AppState.SUCCEEDED, (name, status)
     assert (tmp_path / "touched.txt").exists()
+
+
+def test_multi_role_ps_style_app(tmp_path):
+    # parameter-server-style AppDef: three heterogeneous roles in one app
+    # (reference parity: cli/test/container ps_main/train_main/reader_main)
+    from torchx_amd.specs import AppDef, Resource, Role, macros
+
+    def role(name, num_replicas, msg):
+        return Role(
+            name=name, image="/", entrypoint=sys.executable,
+            args=["-c",
+                  f"print('{msg}', '{macros.replica_id}')"],
+            num_replicas=num_replicas,
+            resource=Resource(cpu=1, gpu=0, memMB=256),
+        )
+
+    app = AppDef(name="psjob", roles=[
+        role("ps", 1, "ps-up"),
+        role("trainer", 2, "train-step"),
+        role("reader", 2, "read-batch"),
+    ])
+    with get_runner("test") as runner:
+        handle = runner.run(
+            app, scheduler="local_cwd",
+            cfg={"log_dir": str(tmp_path),
+                 "auto_set_hip_visible_devices": False},
+        )
+        status = _wait(runner, handle)
+        assert status.state == AppState.SUCCEEDED
+        # per-role status carries every replica of every role
+        by_role = {r.role: r for r in status.roles}
+        assert set(by_role) == {"ps", "trainer", "reader"}
+        assert len(by_role["trainer"].replicas) == 2
+        # logs are addressable per role/replica; ${replica_id} substituted
+        trainer1 = list(runner.log_lines(handle, "trainer", k=1))
+        assert any("train-step 1" in ln for ln in trainer1)
+        reader0 = list(runner.log_lines(handle, "reader", k=0))
+        assert any("read-batch 0" in ln for ln in reader0)
