@@ -322,3 +322,15 @@ def test_cli_entry_point_subcommands(monkeypatch):
     assert cli_main.main(["hello", "--who", "mi355x"]) == 0
     assert cli_main.main(["status", "anything"]) == 0
     assert calls == [("hello", "mi355x"), ("status", "anything")]
+
+
+def test_cli_run_scheduler_from_config(tmp_path, monkeypatch, capsys):
+    # [cli:run] scheduler= supplies the default; -s still wins
+    monkeypatch.chdir(tmp_path)
+    monkeypatch.delenv("TORCHXCONFIG", raising=False)
+    (tmp_path / ".torchxconfig").write_text(
+        "[cli:run]\nscheduler = slurm\n"
+    )
+    assert cli_main(["run", "--dryrun", "utils.echo", "--msg", "hi"]) == 0
+    out = capsys.readouterr().out
+    assert "sbatch" in out or "srun" in out  # slurm request generated

@@ -78,7 +78,11 @@ def cmd_run(args: argparse.Namespace) -> int:
         )
         args.dryrun = bool(spec.get("dryrun", args.dryrun))
     component, comp_args = _parse_run_args(args.component_name_and_args)
-    scheduler = args.scheduler
+    # precedence: -s flag > [cli:run] scheduler= > built-in default
+    # (reference parity: cli/argparse_util.py torchxconfig-backed defaults)
+    scheduler = (args.scheduler
+                 or torchx_config.get_config("cli", "run", "scheduler")
+                 or "local_cwd")
     runner = get_runner(component_defaults=_component_defaults())
     cfg = runner.scheduler_run_opts(scheduler).cfg_from_str(args.scheduler_args)
     if args.dryrun:
@@ -294,7 +298,9 @@ def create_parser() -> argparse.ArgumentParser:
     default_sched = "local_cwd"
 
     p = sub.add_parser("run", help="run a component as a job")
-    p.add_argument("-s", "--scheduler", type=str, default=default_sched)
+    p.add_argument("-s", "--scheduler", type=str, default=None,
+                   help=f"scheduler (default: [cli:run] scheduler= from "
+                        f".torchxconfig, else {default_sched})")
     p.add_argument("-cfg", "--scheduler_args", type=str, default="",
                    help="scheduler runopts, e.g. k1=v1,k2=v2")
     p.add_argument("--dryrun", action="store_true")
