@@ -95,3 +95,49 @@ def test_structured_error_surfaces_in_status(tmp_path):
         status = runner.wait(handle, wait_interval=0.3)
     assert status.state == AppState.FAILED
     assert "intentional-kaboom" in status.format()
+
+
+def test_print_log_lines_merges_with_prefixes():
+    import io
+
+    from torchx_amd.utils.log_tee import print_log_lines
+
+    def lines(role, replica):
+        return [f"{role}-{replica} line{j}" for j in range(3)]
+
+    buf = io.StringIO()
+    print_log_lines([("trainer", 0), ("trainer", 1), ("ps", 0)],
+                    lines, stream=buf, colored=False)
+    out = buf.getvalue().splitlines()
+    assert len(out) == 9
+    # every line carries its role/replica prefix; streams fully drained
+    assert sum(1 for ln in out if ln.startswith("trainer/0 ")) == 3
+    assert sum(1 for ln in out if ln.startswith("trainer/1 ")) == 3
+    assert sum(1 for ln in out if ln.startswith("ps/0 ")) == 3
+    assert "trainer/0 trainer-0 line2" in out
+
+
+def test_print_log_lines_colored_prefix_stable():
+    import io
+
+    from torchx_amd.utils.log_tee import print_log_lines
+
+    buf = io.StringIO()
+    print_log_lines([("w", 0)], lambda r, k: ["x"], stream=buf, colored=True)
+    line = buf.getvalue()
+    assert "\033[" in line and "w/0" in line
+
+
+def test_print_log_lines_raises_puller_error():
+    import io
+
+    import pytest as _pytest
+
+    from torchx_amd.utils.log_tee import print_log_lines
+
+    def boom(role, replica):
+        raise RuntimeError("log source gone")
+
+    with _pytest.raises(RuntimeError, match="log source gone"):
+        print_log_lines([("w", 0)], boom, stream=io.StringIO(),
+                        colored=False)
