@@ -256,3 +256,30 @@ class TestK8sDescribe:
         except Exception:
             resp = None
         assert resp is None
+
+
+class TestSlurmLogIter:
+    def test_reads_role_replica_file_with_regex(self, tmp_path, monkeypatch):
+        s = _slurm()
+        # job-dir registry maps the app to its sbatch dir
+        monkeypatch.setattr(s, "_get_job_dirs",
+                            lambda: {"42": str(tmp_path)})
+        (tmp_path / "slurm-42-trainer-1.out").write_text(
+            "step 1 loss 2.0\nnoise\nstep 2 loss 1.5\n"
+        )
+        (tmp_path / "slurm-42-trainer-1.err").write_text("warn: x\n")
+        lines = list(s.log_iter("42", "trainer", 1, regex=r"step \d"))
+        assert lines == ["step 1 loss 2.0", "step 2 loss 1.5"]
+        from torchx_amd.schedulers.api import Stream
+
+        err = list(s.log_iter("42", "trainer", 1, streams=Stream.STDERR))
+        assert err == ["warn: x"]
+
+    def test_missing_log_file_raises(self, tmp_path, monkeypatch):
+        import pytest
+
+        s = _slurm()
+        monkeypatch.setattr(s, "_get_job_dirs",
+                            lambda: {"42": str(tmp_path)})
+        with pytest.raises(FileNotFoundError):
+            list(s.log_iter("42", "trainer", 0))
