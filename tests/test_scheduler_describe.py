@@ -283,3 +283,21 @@ class TestSlurmLogIter:
                             lambda: {"42": str(tmp_path)})
         with pytest.raises(FileNotFoundError):
             list(s.log_iter("42", "trainer", 0))
+
+    def test_log_iter_reads_pod_log(self):
+        from torchx_amd.schedulers.kubernetes_scheduler import (
+            KubernetesScheduler,
+        )
+
+        s = KubernetesScheduler("t")
+        core = MagicMock()
+        core.read_namespaced_pod_log.return_value = (
+            "step 1\nnoise\nstep 2\n"
+        )
+        s._core_api = lambda: core
+        lines = list(s.log_iter("ns1:app-x", "trainer", 3, regex=r"step"))
+        assert lines == ["step 1", "step 2"]
+        kw = core.read_namespaced_pod_log.call_args.kwargs
+        assert kw["namespace"] == "ns1"
+        # volcano pod naming: <job>-<task>-<replica>-0
+        assert kw["name"] == "app-x-trainer-3-0"
