@@ -116,3 +116,18 @@ class TestVersionAndPipelines:
         # intentionally empty in-core; adapters ship as plugins
         mod = importlib.import_module("torchx_amd.pipelines")
         assert mod is pipelines
+
+
+class TestIds:
+    def test_make_unique_sanitizes_and_suffixes(self):
+        from torchx_amd.schedulers.ids import make_unique, random_id
+
+        uid = make_unique("my app/v2")
+        base, _, suffix = uid.rpartition("-")
+        assert base == "my-app-v2"
+        assert len(suffix) == 8 and suffix.isalnum()
+        # ids are unique across calls
+        assert make_unique("x") != make_unique("x")
+        # degenerate names still produce a valid id
+        assert make_unique("///").startswith("app-")
+        assert len(random_id(4)) == 4
