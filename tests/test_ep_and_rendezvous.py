@@ -69,6 +69,88 @@ def test_rendezvous_single_node():
     rdzv._store = None  # drop the TCPStore
 
 
+def test_rendezvous_late_joiner_scale_up():
+    # a node joining just after round close must not
+    # crash. With room below max_nodes it signals a scale-up
+    # re-rendezvous and both nodes re-form at the next round.
+    import threading
+
+    from torchx_amd.agent.rendezvous import C10dRendezvous, free_port
+
+    port = free_port()
+    a = C10dRendezvous(f"127.0.0.1:{port}", "unit-su", 1, 2,
+                       timeout=30.0, last_call_timeout=0.2)
+    b = C10dRendezvous(f"127.0.0.1:{port}", "unit-su", 1, 2,
+                       timeout=30.0, last_call_timeout=0.2)
+
+    results = {}
+
+    def agent_a():
+        r0 = a.join(0)
+        results["a0"] = r0
+        # simulate the agent main loop noticing the restart signal
+        import time
+        deadline = time.time() + 30
+        while a.restart_round() <= 0:
+            if time.time() > deadline:
+                return
+            time.sleep(0.05)
+        results["a1"] = a.join(a.restart_round())
+
+    t = threading.Thread(target=agent_a, daemon=True)
+    t.start()
+    # wait until round 0 is closed (world=1), then join late
+    import time
+    deadline = time.time() + 30
+    while "a0" not in results:
+        if time.time() > deadline:
+            raise AssertionError("agent A never joined round 0")
+        time.sleep(0.05)
+    assert results["a0"].num_nodes == 1
+
+    rb = b.join(0)  # late: round 0 already closed
+    t.join(timeout=30)
+    assert rb.round >= 1
+    assert rb.num_nodes == 2
+    assert "a1" in results
+    assert results["a1"].round == rb.round
+    assert {results["a1"].node_rank, rb.node_rank} == {0, 1}
+    a._store = None
+    b._store = None
+
+
+def test_rendezvous_late_joiner_standby_replaces_failed_node():
+    # Gang full (max_nodes reached): the late joiner stands by; when a
+    # failure bumps the restart counter it joins the next round alone
+    # (replacement-node case).
+    import threading
+    import time
+
+    from torchx_amd.agent.rendezvous import C10dRendezvous, free_port
+
+    port = free_port()
+    a = C10dRendezvous(f"127.0.0.1:{port}", "unit-sb", 1, 1,
+                       timeout=30.0, last_call_timeout=0.1)
+    b = C10dRendezvous(f"127.0.0.1:{port}", "unit-sb", 1, 1,
+                       timeout=30.0, last_call_timeout=0.1)
+    ra = a.join(0)
+    assert ra.num_nodes == 1
+
+    def fail_later():
+        time.sleep(0.5)
+        a.signal_restart(0)  # simulates worker failure on node A
+
+    t = threading.Thread(target=fail_later, daemon=True)
+    t.start()
+    rb = b.join(0)  # stands by until the restart opens round 1
+    t.join(timeout=10)
+    assert rb.round == 1
+    assert rb.num_nodes == 1
+    assert rb.node_rank == 0
+    a._store = None
+    b._store = None
+
+
 def test_free_port_is_bindable():
     import socket
 

@@ -142,6 +142,24 @@ class TestRunnerSemantics:
         assert "TORCHX_JOB_ID" in role.env
         assert role.env.get("TORCHX_INTERNAL_SESSION_ID")
 
+    def test_job_id_reaches_materialized_request(self):
+        # regression: TORCHX_JOB_ID must be injected
+        # BEFORE submit_dryrun so the ${app_id} macro is substituted into
+        # the materialized PopenRequest replica env — otherwise every real
+        # job sees <unset_run_id> in AppRun.run_from_env().
+        app = AppDef(name="a", roles=[_role(num_replicas=2)])
+        with get_runner("sess-y") as runner:
+            info = runner.dryrun(app, "local_cwd",
+                                 cfg={"auto_set_hip_visible_devices": False})
+        for params in info.request.role_params.values():
+            for p in params:
+                job_id = p.env.get("TORCHX_JOB_ID")
+                assert job_id is not None
+                assert "${app_id}" not in job_id
+                # handle embeds the scheduler-assigned app id
+                assert job_id.startswith("local_cwd://sess-y/a-")
+                assert p.env.get("TORCHX_INTERNAL_SESSION_ID")
+
     def test_run_rejects_unknown_scheduler(self):
         with get_runner("t") as runner:
             with pytest.raises(Exception):

@@ -246,6 +246,9 @@ def main(argv: Optional[List[str]] = None) -> int:
                 round_, restart_count, args.max_restarts, args.nproc_per_node,
             )
             result = rdzv.join(round_)
+            # a late joiner may have stood by / scale-up-signalled into a
+            # later round — result.round is authoritative
+            round_ = result.round
             log.info(
                 "joined as node %d/%d; master %s:%d",
                 result.node_rank, result.num_nodes, result.master_addr,

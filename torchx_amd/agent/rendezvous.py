@@ -112,6 +112,58 @@ class C10dRendezvous:
 
     # -- join ---------------------------------------------------------------
     def join(self, round_: int) -> RendezvousResult:
+        """Join round ``round_`` (or a later one).
+
+        A node that arrives after a round has closed does NOT error out
+        (torchelastic semantics — the reference delegates this to
+        torchelastic via torchx/components/dist.py:262): if the closed
+        round has room below max_nodes the late joiner signals a
+        re-rendezvous (elastic scale-up) and everyone re-forms at the next
+        round; if the gang is already full it stands by until the next
+        restart round opens. The returned ``RendezvousResult.round`` is
+        therefore authoritative and may be greater than ``round_``.
+        """
+        deadline = time.time() + self.timeout
+        while True:
+            result = self._try_join(round_)
+            if result is not None:
+                return result
+            # round closed before we joined
+            world = self._round_world(round_)
+            if world is not None and world < self.max_nodes:
+                # scale-up: ask the running gang to re-rendezvous with us
+                log.info(
+                    "late join for round %d (world %d < max %d); "
+                    "signalling scale-up re-rendezvous", round_, world,
+                    self.max_nodes,
+                )
+                self.signal_restart(round_)
+            round_ = self._wait_for_round_after(round_, deadline)
+
+    def _round_world(self, round_: int) -> Optional[int]:
+        store = self.store()
+        prefix = f"rdzv/{self.run_id}/{round_}/"
+        try:
+            if store.check([prefix + "closed"]):
+                return int(json.loads(store.get(prefix + "closed"))["world"])
+        except Exception:  # noqa: BLE001
+            pass
+        return None
+
+    def _wait_for_round_after(self, round_: int, deadline: float) -> int:
+        """Standby: poll the restart counter until a round > round_ opens."""
+        while True:
+            cur = self.restart_round()
+            if cur > round_:
+                return cur
+            if time.time() > deadline:
+                raise TimeoutError(
+                    f"timed out standing by for a rendezvous round after "
+                    f"{round_}"
+                )
+            time.sleep(0.5)
+
+    def _try_join(self, round_: int) -> Optional[RendezvousResult]:
         store = self.store()
         prefix = f"rdzv/{self.run_id}/{round_}/"
         seq = int(store.add(prefix + "count", 1)) - 1
@@ -149,10 +201,8 @@ class C10dRendezvous:
         blob = store.get(prefix + "closed")
         info = json.loads(blob)
         if seq >= info["world"]:
-            raise RuntimeError(
-                f"node joined too late for round {round_} "
-                f"(seq {seq} >= world {info['world']}); rejoin next round"
-            )
+            # joined after the leader closed the round — caller retries
+            return None
         return RendezvousResult(
             round=round_,
             node_rank=seq,

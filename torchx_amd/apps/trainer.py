@@ -77,7 +77,9 @@ def load_checkpoint(path: str, model, opt) -> int:
 
     fs, p = fsspec.core.url_to_fs(path)
     with fs.open(p, "rb") as f:
-        sd = torch.load(f, map_location="cpu", weights_only=False)
+        # weights_only: the payload is plain tensors/ints; never unpickle
+        # arbitrary objects from an fsspec URI (shared/writable dirs)
+        sd = torch.load(f, map_location="cpu", weights_only=True)
     model.load_state_dict(sd["model"])
     opt.load_state_dict(sd["opt"])
     return int(sd["step"])

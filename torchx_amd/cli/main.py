@@ -55,6 +55,7 @@ def _component_defaults() -> Dict[str, Dict[str, str]]:
 
 
 def cmd_run(args: argparse.Namespace) -> int:
+    stdin_scheduler_args: Optional[Dict[str, object]] = None
     if getattr(args, "stdin", False):
         # JSON run spec from stdin: {"component": ..., "component_args":
         # [...], "scheduler": ..., "scheduler_args": {...}, "dryrun": bool}
@@ -73,9 +74,9 @@ def cmd_run(args: argparse.Namespace) -> int:
                                    spec.get("component_args", [])]
         )
         args.scheduler = spec.get("scheduler", args.scheduler)
-        args.scheduler_args = ",".join(
-            f"{k}={v}" for k, v in spec.get("scheduler_args", {}).items()
-        )
+        # keep the JSON dict as-is — flattening to "k=v,k=v" would corrupt
+        # values containing ',' or '=' (mounts, dict-typed opts)
+        stdin_scheduler_args = dict(spec.get("scheduler_args", {}))
         args.dryrun = bool(spec.get("dryrun", args.dryrun))
     component, comp_args = _parse_run_args(args.component_name_and_args)
     # precedence: -s flag > [cli:run] scheduler= > built-in default
@@ -84,7 +85,12 @@ def cmd_run(args: argparse.Namespace) -> int:
                  or torchx_config.get_config("cli", "run", "scheduler")
                  or "local_cwd")
     runner = get_runner(component_defaults=_component_defaults())
-    cfg = runner.scheduler_run_opts(scheduler).cfg_from_str(args.scheduler_args)
+    opts = runner.scheduler_run_opts(scheduler)
+    cfg = opts.cfg_from_str(args.scheduler_args)
+    if stdin_scheduler_args:
+        for k, v in stdin_scheduler_args.items():
+            opt = opts.get(k)
+            cfg[k] = opt.cast(v) if opt is not None else v
     if args.dryrun:
         info = runner.dryrun_component(
             component, comp_args, scheduler, cfg=cfg,

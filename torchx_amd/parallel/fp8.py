@@ -109,7 +109,16 @@ class Fp8Linear(nn.Module):
         shape = x.shape
         x2 = x.reshape(-1, self.in_features)
         if x.is_cuda:
+            # the fused cast+transpose kernel stores 8-wide vectors: pad the
+            # token dim up to a multiple of 8 (zero rows quantize to zero and
+            # the slice below keeps their grads out of the graph)
+            rows = x2.shape[0]
+            pad = (-rows) % 8
+            if pad:
+                x2 = torch.nn.functional.pad(x2, (0, 0, 0, pad))
             y = _Fp8Matmul.apply(x2, self.weight, self)
+            if pad:
+                y = y[:rows]
         else:  # CPU CI path: plain bf16 matmul
             y = x2 @ self.weight.t()
         return y.reshape(*shape[:-1], self.out_features)
@@ -119,20 +128,44 @@ class Fp8Linear(nn.Module):
                 f"out_features={self.out_features}, fp8=e4m3(delayed)")
 
 
+DEFAULT_FP8_EXCLUDE = ("lm_head", "output", "head", "embed")
+
+
 def convert_to_fp8(module: nn.Module,
-                   min_features: int = 1024) -> nn.Module:
+                   min_features: int = 1024,
+                   exclude: tuple = DEFAULT_FP8_EXCLUDE) -> nn.Module:
     """Replace every bias-free ``nn.Linear`` whose dims are fp8-friendly
     (divisible by 16, at least ``min_features``) with an :class:`Fp8Linear`
     SHARING the same weight Parameter. Call BEFORE FlatParams so the
-    flat-buffer views attach to the shared weights."""
-    for name, child in list(module.named_children()):
-        if (isinstance(child, nn.Linear) and child.bias is None
-                and child.in_features % 16 == 0
-                and child.out_features % 16 == 0
-                and child.in_features >= min_features):
-            repl = Fp8Linear(child.in_features, child.out_features,
-                             weight=child.weight)
-            setattr(module, name, repl)
-        else:
-            convert_to_fp8(child, min_features)
+    flat-buffer views attach to the shared weights.
+
+    Modules whose name contains any ``exclude`` substring stay in high
+    precision — standard fp8 recipes (TransformerEngine) keep the output
+    projection out of e4m3: quantized logits feeding cross-entropy degrade
+    loss quality. Weights tied to an excluded module (e.g. lm_head sharing
+    the embedding parameter) are also left alone.
+    """
+    # collect parameters owned by excluded modules so tied weights skip too
+    excluded_params = set()
+    for qname, sub in module.named_modules():
+        leaf = qname.rsplit(".", 1)[-1]
+        if any(pat in leaf for pat in exclude):
+            for p in sub.parameters(recurse=True):
+                excluded_params.add(id(p))
+
+    def _convert(mod: nn.Module) -> None:
+        for name, child in list(mod.named_children()):
+            if (isinstance(child, nn.Linear) and child.bias is None
+                    and child.in_features % 16 == 0
+                    and child.out_features % 16 == 0
+                    and child.in_features >= min_features
+                    and not any(pat in name for pat in exclude)
+                    and id(child.weight) not in excluded_params):
+                repl = Fp8Linear(child.in_features, child.out_features,
+                                 weight=child.weight)
+                setattr(mod, name, repl)
+            else:
+                _convert(child)
+
+    _convert(module)
     return module

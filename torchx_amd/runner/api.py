@@ -193,6 +193,15 @@ class Runner:
                             key = f"TORCHX_TRACKER_{tname.upper()}_CONFIG"
                             role.env.setdefault(key, tcfg)
 
+            # TORCHX_JOB_ID must be set BEFORE submit_dryrun materializes the
+            # request (reference api.py:400-459): the ${app_id} macro is
+            # substituted per replica by the scheduler at materialize time.
+            for role in app.roles:
+                role.env.setdefault(
+                    "TORCHX_JOB_ID",
+                    make_app_handle(scheduler, self._name, "${app_id}"),
+                )
+
             # workspace build (mutates role.image for build-based scheds)
             if workspace:
                 from torchx_amd.workspace.api import WorkspaceMixin
@@ -200,16 +209,7 @@ class Runner:
                 if isinstance(sched, WorkspaceMixin):
                     sched.build_workspaces(app, workspace, cfg)
 
-            info = sched.submit_dryrun(app, cfg)
-
-            # TORCHX_JOB_ID is the handle; known only post-schedule for some
-            # schedulers — inject the template (scheduler fills app_id macro)
-            for role in app.roles:
-                role.env.setdefault(
-                    "TORCHX_JOB_ID",
-                    make_app_handle(scheduler, self._name, "${app_id}"),
-                )
-            return info
+            return sched.submit_dryrun(app, cfg)
 
     def schedule(self, dryrun_info: AppDryRunInfo) -> AppHandle:
         scheduler = dryrun_info._scheduler

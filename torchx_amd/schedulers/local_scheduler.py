@@ -271,6 +271,15 @@ class LocalScheduler(Scheduler[PopenRequest]):
                 replica = values.apply(role)
                 replica_log = os.path.join(log_dir, role.name, str(replica_id))
                 env = dict(replica.env)
+                # PATH precedence (parity: local_scheduler.py:977-990):
+                # prepend_cwd puts binaries in cwd ahead of PATH, default
+                # appends cwd so PATH binaries win
+                base_path = env.get("PATH") or os.environ.get("PATH", "")
+                cwd = os.getcwd()
+                if cfg.get("prepend_cwd"):
+                    env["PATH"] = os.pathsep.join(p for p in (cwd, base_path) if p)
+                else:
+                    env["PATH"] = os.pathsep.join(p for p in (base_path, cwd) if p)
                 env.setdefault("TORCHX_RANK0_HOST", "localhost")
                 env.setdefault("TORCHX_IMAGE", role.image)
                 env.setdefault("PYTHONUNBUFFERED", "1")
@@ -326,11 +335,13 @@ class LocalScheduler(Scheduler[PopenRequest]):
         args = list(p.args)
         if not args or not args[0]:
             raise ValueError(f"role {role} replica {idx}: empty entrypoint")
-        # resolve entrypoint relative to cwd if present (local_cwd semantics)
-        if not os.path.isabs(args[0]):
-            cand = os.path.join(os.getcwd(), args[0])
-            if os.path.isfile(cand) and os.access(cand, os.X_OK):
-                args[0] = cand
+        # bare entrypoint names resolve through the replica PATH (which has
+        # cwd appended/prepended per the prepend_cwd opt); explicit relative
+        # paths resolve against cwd as Popen does natively
+        if os.sep not in args[0] and not os.path.isabs(args[0]):
+            resolved = shutil.which(args[0], path=env.get("PATH"))
+            if resolved:
+                args[0] = resolved
         proc = subprocess.Popen(
             args,
             env=env,
@@ -435,30 +446,32 @@ class LogIterator:
         self._path = path
         self._tail = should_tail
         self._pos = 0
-        self._buf = ""
+        self._buf = b""
 
     def __iter__(self):
+        # read in BINARY mode and track byte offsets: seeking a text-mode
+        # file by character count corrupts the tail on multibyte UTF-8
         while True:
             try:
-                with open(self._path, "r", errors="replace") as f:
+                with open(self._path, "rb") as f:
                     f.seek(self._pos)
                     chunk = f.read(65536)
             except OSError:
-                chunk = ""
+                chunk = b""
             if chunk:
                 self._pos += len(chunk)
                 self._buf += chunk
                 while True:
-                    nl = self._buf.find("\n")
+                    nl = self._buf.find(b"\n")
                     if nl < 0:
                         break
-                    yield self._buf[:nl]
+                    yield self._buf[:nl].decode("utf-8", errors="replace")
                     self._buf = self._buf[nl + 1:]
             else:
                 finished = is_terminal(self._app.poll())
                 if finished or not self._tail:
                     if self._buf:
-                        yield self._buf
+                        yield self._buf.decode("utf-8", errors="replace")
                     return
                 time.sleep(0.1)
 

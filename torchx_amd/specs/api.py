@@ -170,15 +170,34 @@ Mount = Union[BindMount, VolumeMount, DeviceMount]
 
 @dataclass
 class Workspace:
-    """A local project directory to overlay onto the role image at build time."""
+    """Local project directories to overlay onto the role image at build
+    time. ``projects`` maps a local dir -> destination path inside the
+    image root ("" = image root). Multiple projects are merged (later
+    entries win on conflicts) before the build (parity:
+    torchx/workspace/api.py:97-179)."""
 
     projects: Dict[str, str] = field(default_factory=dict)
 
     @staticmethod
     def from_str(workspace: Optional[str]) -> "Workspace":
+        """``"dir"`` -> single unmapped project; ``"a:dst1,b:dst2"`` ->
+        mapped multi-project."""
         if not workspace:
             return Workspace()
-        return Workspace(projects={workspace: ""})
+        if isinstance(workspace, Workspace):
+            return workspace
+        projects: Dict[str, str] = {}
+        for part in str(workspace).split(","):
+            part = part.strip()
+            if not part:
+                continue
+            src, _, dst = part.partition(":")
+            projects[src] = dst
+        return Workspace(projects=projects)
+
+    def is_unmapped_single_project(self) -> bool:
+        return (len(self.projects) == 1
+                and not next(iter(self.projects.values())))
 
     def __bool__(self) -> bool:
         return bool(self.projects)

@@ -7,9 +7,12 @@ from __future__ import annotations
 import fnmatch
 import os
 import posixpath
-from typing import Any, Iterable, List, Mapping, Tuple
+import tempfile
+from typing import Any, Iterable, List, Mapping, Tuple, Union
 
-from torchx_amd.specs import AppDef, Role
+from torchx_amd.specs import AppDef, Role, Workspace
+
+WorkspaceLike = Union[str, "Workspace"]
 
 TORCHX_IGNORE = ".torchxignore"
 
@@ -57,6 +60,20 @@ def walk_workspace(workspace: str) -> Iterable[Tuple[str, str]]:
                 yield os.path.join(root, f), rel
 
 
+def merge_workspace(ws: "Workspace", outdir: str) -> None:
+    """Copy every project of ``ws`` into ``outdir`` under its destination
+    path, honoring each project's .torchxignore; later projects win on
+    file conflicts (parity: torchx/workspace/api.py:149-154)."""
+    import shutil
+
+    for src, dst in ws.projects.items():
+        base = os.path.join(outdir, dst) if dst else outdir
+        for abs_path, rel in walk_workspace(src):
+            target = os.path.join(base, rel)
+            os.makedirs(os.path.dirname(target) or base, exist_ok=True)
+            shutil.copy2(abs_path, target)
+
+
 class WorkspaceMixin:
     """Schedulers mix this in to support workspace patching."""
 
@@ -65,19 +82,45 @@ class WorkspaceMixin:
 
         return runopts()
 
-    def build_workspaces(self, app: AppDef, workspace: str,
+    def build_workspaces(self, app: AppDef, workspace: "WorkspaceLike",
                          cfg: Mapping[str, Any]) -> None:
-        images: dict = {}
-        for role in app.roles:
-            key = (role.image, workspace)
-            if key not in images:
-                images[key] = self.build_workspace_and_update_role(
-                    role, workspace, cfg
+        """Build each role's workspace and update ``role.image`` in place.
+
+        ``workspace`` may be a plain dir, a ``"src:dst,src2:dst2"`` spec
+        string, or a :class:`~torchx_amd.specs.Workspace`. Multi-project
+        workspaces are merged into a tmpdir first; builds are cached per
+        (image, workspace) so roles sharing both build once.
+        """
+        from torchx_amd.specs import Workspace
+
+        ws = Workspace.from_str(workspace)
+        if not ws:
+            return
+        build_cache: dict = {}
+        key_ws = tuple(sorted(ws.projects.items()))
+        merged_dir: str = ""
+        tmp = None
+        try:
+            for role in app.roles:
+                key = (role.image, key_ws)
+                if key in build_cache:
+                    role.image = build_cache[key]
+                    continue
+                if ws.is_unmapped_single_project():
+                    build_dir = next(iter(ws.projects))
+                else:
+                    if not merged_dir:
+                        tmp = tempfile.TemporaryDirectory(
+                            suffix="_torchx_workspace")
+                        merged_dir = tmp.name
+                        merge_workspace(ws, merged_dir)
+                    build_dir = merged_dir
+                build_cache[key] = self.build_workspace_and_update_role(
+                    role, build_dir, cfg
                 )
-            else:
-                role.image = images[key]
-        for key, img in images.items():
-            pass
+        finally:
+            if tmp is not None:
+                tmp.cleanup()
 
     def build_workspace_and_update_role(self, role: Role, workspace: str,
                                         cfg: Mapping[str, Any]) -> str:
