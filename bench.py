@@ -36,6 +36,11 @@ def main() -> int:
                    choices=["bf16", "fp8"],
                    help="fp8: OCP e4m3/e5m2 GEMMs via hipBLASLt scaled-mm "
                         "(opt-in; the headline metric is bf16)")
+    p.add_argument("--launch-latency", type=str, default="auto",
+                   choices=["auto", "on", "off"],
+                   help="also measure launch->first-step latency through the "
+                        "full dist.ddp launcher path (the other half of the "
+                        "BASELINE metric); auto = on for 1-GPU runs")
     args = p.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -175,6 +180,26 @@ def main() -> int:
     tokens_per_step = B * S * world
     tok_s = tokens_per_step / (elapsed / args.steps)
 
+    # launch->first-step latency (the other half of the BASELINE metric):
+    # the full dist.ddp launcher path — materialize -> Popen -> agent ->
+    # c10d rendezvous -> RCCL init -> first fwd+bwd+step (1x1, gpu_tiny)
+    launch_latency = None
+    measure_latency = (args.launch_latency == "on"
+                       or (args.launch_latency == "auto"
+                           and world == 1 and use_gpu))
+    if measure_latency and rank == 0:
+        try:
+            # free the bench model's HBM first so the spawned trainer and
+            # this process never contend for memory
+            del model, flat, ddp, opt, tokens, targets
+            if use_gpu:
+                torch.cuda.empty_cache()
+            from torchx_amd.utils.launch_latency import measure_launch_latency
+
+            launch_latency = measure_launch_latency(nproc=1, timeout=180.0)
+        except Exception as e:  # noqa: BLE001 — latency is auxiliary
+            print(f"launch-latency measurement failed: {e}", file=sys.stderr)
+
     if rank == 0:
         result = {
             "metric": "tokens_per_second",
@@ -197,6 +222,9 @@ def main() -> int:
                 "final_loss": loss,
             },
         }
+        if launch_latency is not None:
+            result["launch_to_first_step_s"] = launch_latency["value"]
+            result["launch_submit_s"] = launch_latency["submit_seconds"]
         print(json.dumps(result), flush=True)
 
     if distributed:
