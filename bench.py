@@ -49,8 +49,10 @@ def main() -> int:
 
     use_gpu = torch.cuda.is_available() and args.device != "cpu"
     if use_gpu:
-        torch.cuda.set_device(local_rank)
-        device = torch.device("cuda", local_rank)
+        # modulo: >1 ranks may share a device (single-GPU RCCL de-risk tests)
+        dev_idx = local_rank % torch.cuda.device_count()
+        torch.cuda.set_device(dev_idx)
+        device = torch.device("cuda", dev_idx)
         backend = "nccl"  # RCCL on ROCm
         # tuned hipBLASLt algorithm table (tools/gemm_tune.py), read-only
         tuned = os.path.join(os.path.dirname(os.path.abspath(__file__)),

@@ -1,0 +1,94 @@
+#!/usr/bin/env python3
+"""Per-shape GEMM efficiency probe for the Llama-3-8B step shapes.
+
+The r2a profile shows the three GEMM passes running at very different
+rates (fwd NT 1.04 PF/s, dgrad NN 1.34, wgrad split-K TN 1.52) — this
+isolates each (shape, layout) pair to see whether NT is intrinsically slow
+on these shapes (=> store weights transposed) or only slow in-loop
+(=> cache/clock effect).
+
+Layouts measured per shape (M tokens, N out, K in), weight W [N,K]:
+  fwd-NT    x[M,K] @ W^T          (what nn.Linear does today)
+  fwd-NN    x[M,K] @ Wt           (pre-transposed weight Wt=[K,N])
+  dgrad-NN  dy[M,N] @ W
+  dgrad-NT  dy[M,N] @ Wt^T
+  wgrad-TN  dy^T[N,M] @ x         (dW [N,K])
+  wgrad-TN' x^T[K,M] @ dy         (dWt [K,N])
+"""
+
+import argparse
+import json
+import sys
+import time
+
+import torch
+
+
+def bench_mm(fn, iters=20, warmup=5):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--m", type=int, default=16384)
+    p.add_argument("--iters", type=int, default=20)
+    args = p.parse_args()
+    M = args.m
+
+    shapes = [
+        ("qkv", 6144, 4096),
+        ("wo", 4096, 4096),
+        ("gu", 28672, 4096),
+        ("down", 4096, 14336),
+        ("lm_head", 128256, 4096),
+    ]
+
+    dev = torch.device("cuda:0")
+    results = []
+    for name, N, K in shapes:
+        x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+        wt = w.t().contiguous()  # [K, N]
+        dy = torch.randn(M, N, device=dev, dtype=torch.bfloat16)
+        flops = 2.0 * M * N * K
+
+        cases = {
+            "fwd-NT": lambda: x @ w.t(),
+            "fwd-NN": lambda: x @ wt,
+            "dgrad-NN": lambda: dy @ w,
+            "dgrad-NT": lambda: dy @ wt.t(),
+            "wgrad-TN": lambda: dy.t() @ x,
+            "wgrad-TNp": lambda: x.t() @ dy,
+        }
+        row = {"shape": name, "M": M, "N": N, "K": K}
+        for cname, fn in cases.items():
+            dt = bench_mm(fn, iters=args.iters)
+            row[cname] = round(flops / dt / 1e12, 1)  # TF/s
+        results.append(row)
+        print(json.dumps(row), flush=True)
+        del x, w, wt, dy
+        torch.cuda.empty_cache()
+
+    # totals at current vs best-layout assignment
+    tot_cur = tot_best = 0.0
+    for r in results:
+        fl = 2.0 * r["M"] * r["N"] * r["K"] / 1e12
+        tot_cur += fl / r["fwd-NT"] + fl / r["dgrad-NN"] + fl / r["wgrad-TN"]
+        best_fwd = max(r["fwd-NT"], r["fwd-NN"])
+        best_dgrad = max(r["dgrad-NN"], r["dgrad-NT"])
+        best_wgrad = max(r["wgrad-TN"], r["wgrad-TNp"])
+        tot_best += fl / best_fwd + fl / best_dgrad + fl / best_wgrad
+    print(json.dumps({"per_layer_ms_current": tot_cur * 1e3,
+                      "per_layer_ms_best": tot_best * 1e3}))
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

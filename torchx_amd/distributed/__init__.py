@@ -40,7 +40,9 @@ def local_rank() -> int:
 
 def local_device() -> torch.device:
     if torch.cuda.is_available():
-        return torch.device("cuda", local_rank())
+        # modulo: more local ranks than devices = ranks sharing a GPU
+        # (single-GPU multi-rank RCCL tests); 1:1 in production
+        return torch.device("cuda", local_rank() % torch.cuda.device_count())
     return torch.device("cpu")
 
 

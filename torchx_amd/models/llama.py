@@ -148,10 +148,13 @@ class LlamaModel(nn.Module):
                 targets: Optional[torch.Tensor] = None) -> torch.Tensor:
         """tokens [B, S] -> loss (if targets given) else logits."""
         x = self.forward_hidden(tokens)
-        logits = self.lm_head(x)
         if targets is None:
-            return logits
-        return ops.cross_entropy(logits, targets)
+            return self.lm_head(x)
+        # fused chunked lm_head GEMM + CE: the [T, vocab] logits are never
+        # fully materialized or saved for backward (ops.fused_linear_cross_entropy)
+        return ops.fused_linear_cross_entropy(
+            x, self.lm_head.weight, targets
+        )
 
     def num_params(self) -> int:
         return sum(p.numel() for p in self.parameters())

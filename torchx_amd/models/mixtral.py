@@ -249,10 +249,11 @@ class MixtralModel(nn.Module):
         for blk in self.blocks:
             x = blk(x, self.rope_cos, self.rope_sin)
         x = ops.rmsnorm(x, self.final_norm, self.cfg.rms_eps)
-        logits = self.lm_head(x)
         if targets is None:
-            return logits
-        return ops.cross_entropy(logits, targets)
+            return self.lm_head(x)
+        return ops.fused_linear_cross_entropy(
+            x, self.lm_head.weight, targets
+        )
 
     def num_params(self) -> int:
         return sum(p.numel() for p in self.parameters())

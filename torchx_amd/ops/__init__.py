@@ -186,6 +186,67 @@ def cross_entropy(logits: torch.Tensor, targets: torch.Tensor) -> torch.Tensor:
 
 
 # ---------------------------------------------------------------------------
+# Fused lm_head GEMM + cross entropy (chunked — the [T, V] logits are never
+# fully materialized or saved).  The CE loss is the last op of the step, so
+# its input grads are known in forward up to the scalar gout: each token
+# chunk runs  logits GEMM -> ce_fwd -> ce_bwd -> dx GEMM -> dW accumulation
+# back-to-back while the chunk's logits are hot, then the chunk buffers are
+# recycled.  dW accumulates in fp32 (bf16 += across chunks would drift).
+# Cuts the saved-tensor footprint by 2*T*V bf16 (8.4 GB at the bench shape)
+# and keeps the GEMM sizes fat (chunk=4096 rows).
+# ---------------------------------------------------------------------------
+
+
+class _FusedLinearCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x2: torch.Tensor, w: torch.Tensor,
+                targets: torch.Tensor, chunk: int):
+        hip = hip_ops()
+        T, H = x2.shape
+        dx = torch.empty_like(x2)
+        dw32 = torch.zeros_like(w, dtype=torch.float32)
+        loss_sum = torch.zeros((), device=x2.device, dtype=torch.float32)
+        inv_t = 1.0 / T
+        for s in range(0, T, chunk):
+            e = min(T, s + chunk)
+            x_c = x2[s:e]
+            t_c = targets[s:e].contiguous()
+            logits_c = x_c @ w.t()
+            loss_c, lse_c = hip.ce_fwd(logits_c, t_c)
+            loss_sum += loss_c.sum()
+            gscale = torch.full((e - s,), inv_t, device=x2.device,
+                                dtype=torch.float32)
+            dlog_c = hip.ce_bwd(logits_c, t_c, lse_c, gscale)
+            torch.matmul(dlog_c, w, out=dx[s:e])
+            dw32 += dlog_c.t() @ x_c
+        ctx.save_for_backward(dx, dw32)
+        return loss_sum * inv_t
+
+    @staticmethod
+    def backward(ctx, gout: torch.Tensor):
+        dx, dw32 = ctx.saved_tensors
+        g = gout.to(torch.float32)
+        # single fused scale+cast pass for dW; dx scales in place
+        dw = (dw32 * g).to(dx.dtype)
+        dx = dx * gout.to(dx.dtype)
+        return dx, dw, None, None
+
+
+def fused_linear_cross_entropy(
+    x: torch.Tensor, w: torch.Tensor, targets: torch.Tensor,
+    chunk: int = 4096,
+) -> torch.Tensor:
+    """mean CE of ``x @ w.T`` against ``targets`` without materializing the
+    full logits. x [..., H] bf16, w [V, H] bf16, targets [...] int64."""
+    H = x.shape[-1]
+    x2 = x.reshape(-1, H)
+    t = targets.reshape(-1)
+    if _on_gpu(x):
+        return _FusedLinearCE.apply(x2.contiguous(), w, t, chunk)
+    return reference.cross_entropy(x2 @ w.t(), t)
+
+
+# ---------------------------------------------------------------------------
 # Flash attention (causal, GQA), BSHD layout
 # ---------------------------------------------------------------------------
 
