@@ -51,87 +51,91 @@ rmsnorm_fwd_kernel(
 }
 
 // dx = r*(dy*w) - x * (r^3/H) * sum(dy*w*x);  dw = sum_rows(dy * x * r).
-// Each block walks ROWS_PER_BLOCK rows, keeping its dw partial in registers,
-// then does ONE atomicAdd per element at the end — far fewer contending
-// atomics than a per-row scheme (the naive version spent 14% of a full
-// Llama-3-8B step serializing 67M atomics on 4096 addresses).  8 rows per
-// block: 32 starved the chip (rows/32 = 512 blocks = 2 per CU, and each
-// block walks its rows SERIALLY behind a per-row block reduce — measured
-// 25% of HBM roofline; 8 rows -> 2048 blocks keeps the atomic count low
-// while filling the CUs).
-#define RMS_ROWS_PER_BLOCK 8
+//
+// Two kernels (r2 rework): the fused 8-serial-rows-per-block version
+// measured ~10x off the HBM roofline (each block crawled its rows behind a
+// per-row 4-wave LDS reduce, so the chip sat on reduce barriers).  Split:
+//   dx: one block per row, exactly the fwd kernel's shape (which runs at
+//       roofline) — no dw bookkeeping, no serial row walk.
+//   dw: column reduction over row chunks, fp32 register accumulators, one
+//       atomicAdd per element per chunk.  Re-reads dy/x (+33 us at the
+//       bench shape) but every CU stays busy.
 
 template <int ITERS>
 __global__ void __launch_bounds__(256)
-rmsnorm_bwd_kernel(
+rmsnorm_bwd_dx_kernel(
     const unsigned short* __restrict__ dy,
     const unsigned short* __restrict__ x,
     const unsigned short* __restrict__ w,
     const float* __restrict__ invrms,
     unsigned short* __restrict__ dx,
-    float* __restrict__ dw,  // [H] f32, pre-zeroed
-    long rows, int H) {
+    int H) {
   __shared__ float red[4];
-  float wv[ITERS][8], dwacc[ITERS][8];
+  const long row = blockIdx.x;
+  const unsigned short* xr = x + row * (long)H;
+  const unsigned short* dyr = dy + row * (long)H;
+  unsigned short* dxr = dx + row * (long)H;
+  const float r = invrms[row];
+
+  float xs[ITERS][8], dyw[ITERS][8];
+  float t = 0.f;
   #pragma unroll
   for (int it = 0; it < ITERS; ++it) {
     int i = (it * 256 + threadIdx.x) * 8;
-    ushort8 v = *(const ushort8*)(w + i);
+    ushort8 xv = *(const ushort8*)(xr + i);
+    ushort8 dv = *(const ushort8*)(dyr + i);
+    ushort8 wv = *(const ushort8*)(w + i);
     #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      wv[it][j] = bf16_to_f32(v[j]);
-      dwacc[it][j] = 0.f;
+      float xf = bf16_to_f32(xv[j]);
+      float dwf = bf16_to_f32(dv[j]) * bf16_to_f32(wv[j]);
+      xs[it][j] = xf;
+      dyw[it][j] = dwf;
+      t = fmaf(dwf, xf, t);
     }
   }
-
-  const long row0 = (long)blockIdx.x * RMS_ROWS_PER_BLOCK;
-  const long row1 = min(row0 + RMS_ROWS_PER_BLOCK, rows);
-  for (long row = row0; row < row1; ++row) {
-    const unsigned short* xr = x + row * (long)H;
-    const unsigned short* dyr = dy + row * (long)H;
-    unsigned short* dxr = dx + row * (long)H;
-    const float r = invrms[row];
-
-    float xs[ITERS][8], dyw[ITERS][8];
-    float t = 0.f;
-    #pragma unroll
-    for (int it = 0; it < ITERS; ++it) {
-      int i = (it * 256 + threadIdx.x) * 8;
-      ushort8 xv = *(const ushort8*)(xr + i);
-      ushort8 dv = *(const ushort8*)(dyr + i);
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float xf = bf16_to_f32(xv[j]);
-        float df = bf16_to_f32(dv[j]);
-        xs[it][j] = xf;
-        dyw[it][j] = df * wv[it][j];
-        dwacc[it][j] = fmaf(df * xf, r, dwacc[it][j]);
-        t = fmaf(df * wv[it][j], xf, t);
-      }
-    }
-    float ts = block_reduce<4>(t, red,
-        [] __device__ (float a, float b) { return a + b; });
-    const float kk = ts * r * r * r / (float)H;
-
-    #pragma unroll
-    for (int it = 0; it < ITERS; ++it) {
-      int i = (it * 256 + threadIdx.x) * 8;
-      ushort8 ov;
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        ov[j] = f32_to_bf16(fmaf(dyw[it][j], r, -xs[it][j] * kk));
-      }
-      *(ushort8*)(dxr + i) = ov;
-    }
-    __syncthreads();  // red[] reuse across rows
-  }
+  float ts = block_reduce<4>(t, red,
+      [] __device__ (float a, float b) { return a + b; });
+  const float kk = ts * r * r * r / (float)H;
 
   #pragma unroll
   for (int it = 0; it < ITERS; ++it) {
     int i = (it * 256 + threadIdx.x) * 8;
+    ushort8 ov;
     #pragma unroll
-    for (int j = 0; j < 8; ++j) atomicAdd(dw + i + j, dwacc[it][j]);
+    for (int j = 0; j < 8; ++j) {
+      ov[j] = f32_to_bf16(fmaf(dyw[it][j], r, -xs[it][j] * kk));
+    }
+    *(ushort8*)(dxr + i) = ov;
   }
+}
+
+// dw column reduction: grid (row_chunks, H/2048); each block owns a
+// 2048-column tile (256 threads x 8) over DW_CHUNK_ROWS rows.
+#define DW_CHUNK_ROWS 64
+
+__global__ void __launch_bounds__(256)
+rmsnorm_bwd_dw_kernel(
+    const unsigned short* __restrict__ dy,
+    const unsigned short* __restrict__ x,
+    const float* __restrict__ invrms,
+    float* __restrict__ dw,  // [H] f32, pre-zeroed
+    long rows, int H) {
+  const int col = (blockIdx.y * 2048) + threadIdx.x * 8;
+  const long row0 = (long)blockIdx.x * DW_CHUNK_ROWS;
+  const long row1 = min(row0 + DW_CHUNK_ROWS, rows);
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (long row = row0; row < row1; ++row) {
+    const float r = invrms[row];
+    ushort8 xv = *(const ushort8*)(x + row * (long)H + col);
+    ushort8 dv = *(const ushort8*)(dy + row * (long)H + col);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      acc[j] = fmaf(bf16_to_f32(dv[j]) * bf16_to_f32(xv[j]), r, acc[j]);
+    }
+  }
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(dw + col + j, acc[j]);
 }
 
 #define DISPATCH_ITERS(H, FN)                                                  \
@@ -159,13 +163,16 @@ extern "C" void rmsnorm_bwd_launch(const void* dy, const void* x,
                                    const void* w, const void* invrms, void* dx,
                                    void* dw, long rows, int H,
                                    hipStream_t stream) {
-  const long nblk = (rows + RMS_ROWS_PER_BLOCK - 1) / RMS_ROWS_PER_BLOCK;
 #define LAUNCH_B(I)                                                            \
-  hipLaunchKernelGGL((rmsnorm_bwd_kernel<I>), dim3((int)nblk), dim3(256), 0,   \
-                     stream, (const unsigned short*)dy,                        \
+  hipLaunchKernelGGL((rmsnorm_bwd_dx_kernel<I>), dim3((int)rows), dim3(256),   \
+                     0, stream, (const unsigned short*)dy,                     \
                      (const unsigned short*)x, (const unsigned short*)w,       \
-                     (const float*)invrms, (unsigned short*)dx, (float*)dw,    \
-                     rows, H)
+                     (const float*)invrms, (unsigned short*)dx, H)
   DISPATCH_ITERS(H, LAUNCH_B);
 #undef LAUNCH_B
+  const int row_chunks = (int)((rows + DW_CHUNK_ROWS - 1) / DW_CHUNK_ROWS);
+  hipLaunchKernelGGL(rmsnorm_bwd_dw_kernel,
+                     dim3(row_chunks, H / 2048), dim3(256), 0, stream,
+                     (const unsigned short*)dy, (const unsigned short*)x,
+                     (const float*)invrms, (float*)dw, rows, H);
 }
