@@ -35,12 +35,64 @@ def bench_mm(fn, iters=20, warmup=5):
     return (time.perf_counter() - t0) / iters
 
 
+_FLUSH = None
+
+
+def bench_mm_cold(fn, iters=8, warmup=2):
+    """Times only the GEMM (CUDA events) with an LLC-sized flush write
+    between iterations — the in-loop situation, where the weights are
+    never cache-resident."""
+    global _FLUSH
+    if _FLUSH is None:
+        _FLUSH = torch.empty(512 * 1024 * 1024 // 4, device="cuda",
+                             dtype=torch.float32)  # 512 MB > 256 MB LLC
+    start = torch.cuda.Event(enable_timing=True)
+    end = torch.cuda.Event(enable_timing=True)
+    total = 0.0
+    for i in range(warmup + iters):
+        _FLUSH.fill_(float(i))
+        start.record()
+        fn()
+        end.record()
+        torch.cuda.synchronize()
+        if i >= warmup:
+            total += start.elapsed_time(end) / 1e3
+    return total / iters
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--m", type=int, default=16384)
     p.add_argument("--iters", type=int, default=20)
+    p.add_argument("--cold", action="store_true",
+                   help="flush the LLC between iterations (in-loop truth)")
+    p.add_argument("--sustain", type=float, default=0.0,
+                   help="loop the gu fwd GEMM for this many seconds and "
+                        "report the rate per 0.5s window (detects clock "
+                        "throttling under sustained load)")
     args = p.parse_args()
     M = args.m
+
+    if args.sustain > 0:
+        dev = torch.device("cuda:0")
+        N, K = 28672, 4096
+        x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+        flops = 2.0 * M * N * K
+        t_end = time.perf_counter() + args.sustain
+        window_t0 = time.perf_counter()
+        n = 0
+        while time.perf_counter() < t_end:
+            x @ w.t()
+            n += 1
+            if n % 50 == 0:
+                torch.cuda.synchronize()
+                now = time.perf_counter()
+                rate = flops * 50 / (now - window_t0) / 1e12
+                print(json.dumps({"t": round(now - (t_end - args.sustain), 2),
+                                  "gu_fwd_TFs": round(rate, 1)}), flush=True)
+                window_t0 = now
+        return 0
 
     shapes = [
         ("qkv", 6144, 4096),
@@ -67,9 +119,13 @@ def main():
             "wgrad-TN": lambda: dy.t() @ x,
             "wgrad-TNp": lambda: x.t() @ dy,
         }
-        row = {"shape": name, "M": M, "N": N, "K": K}
+        row = {"shape": name, "M": M, "N": N, "K": K,
+               "mode": "cold" if args.cold else "hot"}
         for cname, fn in cases.items():
-            dt = bench_mm(fn, iters=args.iters)
+            if args.cold:
+                dt = bench_mm_cold(fn)
+            else:
+                dt = bench_mm(fn, iters=args.iters)
             row[cname] = round(flops / dt / 1e12, 1)  # TF/s
         results.append(row)
         print(json.dumps(row), flush=True)

@@ -3,10 +3,13 @@
 (rows=16384, H=4096). Reports us/call and achieved TB/s vs the HBM
 roofline (fwd: 3 tensor passes; bwd: dx 3 passes + dw re-read 2 passes)."""
 
+import os
 import sys
 import time
 
 import torch
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
 def bench(fn, iters=50, warmup=10):

@@ -204,9 +204,14 @@ class _FusedLinearCE(torch.autograd.Function):
         hip = hip_ops()
         T, H = x2.shape
         dx = torch.empty_like(x2)
-        dw32 = torch.zeros_like(w, dtype=torch.float32)
+        single = T <= chunk
+        # multi-chunk dW accumulates in fp32 (bf16 += across chunks drifts);
+        # a single chunk writes the bf16 wgrad GEMM output directly
+        dw_acc = None if single else torch.zeros_like(w, dtype=torch.float32)
         loss_sum = torch.zeros((), device=x2.device, dtype=torch.float32)
         inv_t = 1.0 / T
+        gscale = torch.full((min(chunk, T),), inv_t, device=x2.device,
+                            dtype=torch.float32)
         for s in range(0, T, chunk):
             e = min(T, s + chunk)
             x_c = x2[s:e]
@@ -214,27 +219,30 @@ class _FusedLinearCE(torch.autograd.Function):
             logits_c = x_c @ w.t()
             loss_c, lse_c = hip.ce_fwd(logits_c, t_c)
             loss_sum += loss_c.sum()
-            gscale = torch.full((e - s,), inv_t, device=x2.device,
-                                dtype=torch.float32)
-            dlog_c = hip.ce_bwd(logits_c, t_c, lse_c, gscale)
+            dlog_c = hip.ce_bwd(logits_c, t_c, lse_c, gscale[:e - s])
             torch.matmul(dlog_c, w, out=dx[s:e])
-            dw32 += dlog_c.t() @ x_c
-        ctx.save_for_backward(dx, dw32)
+            if single:
+                dw_acc = dlog_c.t() @ x_c
+            else:
+                dw_acc += dlog_c.t() @ x_c
+        ctx.save_for_backward(dx, dw_acc)
         return loss_sum * inv_t
 
     @staticmethod
     def backward(ctx, gout: torch.Tensor):
-        dx, dw32 = ctx.saved_tensors
-        g = gout.to(torch.float32)
-        # single fused scale+cast pass for dW; dx scales in place
-        dw = (dw32 * g).to(dx.dtype)
+        dx, dw_acc = ctx.saved_tensors
+        # single scale(+cast) pass for dW; dx scales in place
+        if dw_acc.dtype == dx.dtype:
+            dw = dw_acc * gout.to(dx.dtype)
+        else:
+            dw = (dw_acc * gout.to(torch.float32)).to(dx.dtype)
         dx = dx * gout.to(dx.dtype)
         return dx, dw, None, None
 
 
 def fused_linear_cross_entropy(
     x: torch.Tensor, w: torch.Tensor, targets: torch.Tensor,
-    chunk: int = 4096,
+    chunk: int = 8192,
 ) -> torch.Tensor:
     """mean CE of ``x @ w.T`` against ``targets`` without materializing the
     full logits. x [..., H] bf16, w [V, H] bf16, targets [...] int64."""

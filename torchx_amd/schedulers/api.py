@@ -9,11 +9,15 @@ log_iter`` monitor the data plane.  Run-config schemas are ``runopts``
 from __future__ import annotations
 
 import abc
+import inspect
 import re
-from dataclasses import dataclass, field
+import typing
+from dataclasses import MISSING, dataclass, field, fields, is_dataclass
 from datetime import datetime
 from enum import Enum
-from typing import Any, Dict, Generic, Iterable, List, Mapping, Optional, TypeVar
+from typing import (
+    Any, Dict, Generic, Iterable, Iterator, List, Mapping, Optional, TypeVar,
+)
 
 from torchx_amd.specs import (
     AppDef,
@@ -27,6 +31,167 @@ from torchx_amd.specs import (
 )
 
 T = TypeVar("T")
+
+
+# ---------------------------------------------------------------------------
+# StructuredOpts: typed scheduler config (parity: torchx/schedulers/api.py:79)
+# ---------------------------------------------------------------------------
+
+
+def _snake_to_camel(name: str) -> str:
+    head, *rest = name.split("_")
+    return head + "".join(p.title() for p in rest)
+
+
+def _camel_to_snake(name: str) -> str:
+    return re.sub(r"(?<=[a-z0-9])([A-Z])", r"_\1", name).lower()
+
+
+def _unwrap_optional(tp: Any) -> Any:
+    """Optional[X] / X | None -> X."""
+    if typing.get_origin(tp) is typing.Union:
+        args = [a for a in typing.get_args(tp) if a is not type(None)]
+        if len(args) == 1:
+            return args[0]
+    return tp
+
+
+def _is_opts_type(tp: Any) -> bool:
+    return (isinstance(tp, type) and issubclass(tp, StructuredOpts)
+            and is_dataclass(tp))
+
+
+@dataclass
+class StructuredOpts(Mapping[str, Any]):
+    """Declare scheduler run options as a ``@dataclass``: field types and
+    defaults become the ``runopts`` schema, field docstrings become the
+    help text, nested ``StructuredOpts`` fields flatten to dot-prefixed
+    keys (``k8s.context``), snake_case fields accept camelCase aliases,
+    and a ``cfg_key`` metadata entry overrides the external key for names
+    that cannot be identifiers (``mail-user``)::
+
+        @dataclass
+        class MyOpts(StructuredOpts):
+            cluster: str
+            '''cluster to submit to'''
+            retries: int = 3
+            '''number of retry attempts'''
+
+    ``MyOpts.as_runopts()`` feeds ``torchx runopts``/``torchx configure``;
+    ``MyOpts.from_cfg(cfg)`` gives typed access in ``_submit_dryrun``.
+    The Mapping protocol keeps instances usable where a cfg dict is
+    expected.
+    """
+
+    # -- construction -------------------------------------------------------
+    @classmethod
+    def from_cfg(cls, cfg: Mapping[str, Any]):
+        hints = typing.get_type_hints(cls)
+        kwargs: Dict[str, Any] = {}
+        for f in fields(cls):
+            ftype = _unwrap_optional(hints.get(f.name, str))
+            if _is_opts_type(ftype):
+                prefix = f.name + "."
+                nested = {k[len(prefix):]: v for k, v in cfg.items()
+                          if k.startswith(prefix)}
+                if nested or (f.default is MISSING
+                              and f.default_factory is MISSING):
+                    kwargs[f.name] = ftype.from_cfg(nested)
+                continue
+            key = f.metadata.get("cfg_key", f.name)
+            for cand in (key, f.name, _snake_to_camel(f.name)):
+                if cand in cfg and cfg[cand] is not None:
+                    kwargs[f.name] = cfg[cand]
+                    break
+        return cls(**kwargs)
+
+    # -- runopts schema ------------------------------------------------------
+    @classmethod
+    def field_docstrings(cls) -> Dict[str, str]:
+        """Attribute docstrings (the string literal following each field)."""
+        docs: Dict[str, str] = {}
+        for klass in reversed(cls.__mro__):
+            if not is_dataclass(klass) or klass is StructuredOpts:
+                continue
+            try:
+                src = inspect.getsource(klass)
+            except (OSError, TypeError):
+                continue
+            pat = re.compile(
+                r"^\s+(\w+)\s*:[^\n]+\n\s+(?:\"\"\"(.+?)\"\"\"|'''(.+?)''')",
+                re.MULTILINE | re.DOTALL,
+            )
+            for m in pat.finditer(src):
+                docs[m.group(1)] = (m.group(2) or m.group(3)).strip()
+        hints = typing.get_type_hints(cls)
+        for f in fields(cls):
+            ftype = _unwrap_optional(hints.get(f.name, str))
+            if _is_opts_type(ftype):
+                for k, d in ftype.field_docstrings().items():
+                    docs[f"{f.name}.{k}"] = d
+        return docs
+
+    @classmethod
+    def as_runopts(cls) -> runopts:
+        opts = runopts()
+        hints = typing.get_type_hints(cls)
+        docs = cls.field_docstrings()
+        for f in fields(cls):
+            ftype = _unwrap_optional(hints.get(f.name, str))
+            if _is_opts_type(ftype):
+                for key, sub in ftype.as_runopts():
+                    opts.add(f"{f.name}.{key}", type_=sub.opt_type,
+                             default=sub.default, required=sub.required,
+                             help=sub.help)
+                continue
+            has_default = (f.default is not MISSING
+                           or f.default_factory is not MISSING)
+            default = f.default if f.default is not MISSING else None
+            opts.add(
+                f.metadata.get("cfg_key", f.name),
+                type_=ftype,
+                default=default,
+                required=not has_default,
+                help=docs.get(f.name, f.name),
+            )
+        return opts
+
+    # -- Mapping protocol (dict-style compatibility) -------------------------
+    def __getitem__(self, key: str) -> Any:
+        if "." in key:
+            head, rest = key.split(".", 1)
+            nested = getattr(self, _camel_to_snake(head), None)
+            if isinstance(nested, StructuredOpts):
+                return nested[rest]
+            raise KeyError(key)
+        snake = _camel_to_snake(key)
+        names = {f.name for f in fields(self)}
+        if snake in names:
+            return getattr(self, snake)
+        for f in fields(self):
+            if f.metadata.get("cfg_key") == key:
+                return getattr(self, f.name)
+        raise KeyError(key)
+
+    def __iter__(self) -> Iterator[str]:
+        hints = typing.get_type_hints(type(self))
+        for f in fields(self):
+            ftype = _unwrap_optional(hints.get(f.name, str))
+            if _is_opts_type(ftype):
+                nested = getattr(self, f.name)
+                if isinstance(nested, StructuredOpts):
+                    for k in nested:
+                        yield f"{f.name}.{k}"
+            else:
+                yield f.metadata.get("cfg_key", f.name)
+
+    def __len__(self) -> int:
+        return sum(1 for _ in self)
+
+    def __or__(self, other: "StructuredOpts") -> Dict[str, Any]:
+        merged: Dict[str, Any] = {k: self[k] for k in self}
+        merged.update({k: other[k] for k in other})
+        return merged
 
 
 class Stream(str, Enum):

@@ -32,7 +32,9 @@ from torchx_amd.specs import (
 
 from torchx_amd.workspace.docker_workspace import DockerWorkspaceMixin
 
-from .api import DescribeAppResponse, ListAppResponse, Scheduler, Stream
+from .api import (
+    DescribeAppResponse, ListAppResponse, Scheduler, Stream, StructuredOpts,
+)
 from .devices import get_device_mounts
 from .ids import make_unique
 
@@ -86,6 +88,21 @@ def _replica_name(app_id: str, role: str, idx: int) -> str:
     return f"{app_id}-{role}-{idx}"
 
 
+@dataclass
+class DockerOpts(StructuredOpts):
+    """Typed run options for ``local_docker`` (StructuredOpts parity:
+    reference schedulers/api.py:79-324 + docker_scheduler.py:129)."""
+
+    copy_env: Optional[List[str]] = None
+    """glob patterns of host env vars to copy into containers"""
+
+    env: Optional[Dict[str, str]] = None
+    """extra env vars for all containers"""
+
+    privileged: bool = False
+    """run containers privileged"""
+
+
 class DockerScheduler(DockerWorkspaceMixin, Scheduler[DockerJob]):
     def __init__(self, session_name: str, client: Optional[Any] = None) -> None:
         super().__init__("local_docker", session_name, docker_client=client)
@@ -99,14 +116,7 @@ class DockerScheduler(DockerWorkspaceMixin, Scheduler[DockerJob]):
         return self.__client
 
     def run_opts(self) -> runopts:
-        opts = runopts()
-        opts.add("copy_env", type_=List[str], default=None,
-                 help="glob patterns of host env vars to copy into containers")
-        opts.add("env", type_=Dict[str, str], default=None,
-                 help="extra env vars for all containers")
-        opts.add("privileged", type_=bool, default=False,
-                 help="run containers privileged")
-        return opts
+        return DockerOpts.as_runopts()
 
     def _ensure_network(self) -> None:
         import docker.errors

@@ -40,7 +40,9 @@ from torchx_amd.specs import (
     runopts,
 )
 
-from .api import DescribeAppResponse, ListAppResponse, Scheduler, Stream
+from .api import (
+    DescribeAppResponse, ListAppResponse, Scheduler, Stream, StructuredOpts,
+)
 from .devices import device_env, hip_device_count, partition_devices
 from .ids import make_unique
 from .streams import Tee
@@ -214,6 +216,21 @@ def _cleanup_all() -> None:
 atexit.register(_cleanup_all)
 
 
+@dataclass
+class LocalOpts(StructuredOpts):
+    """Typed run options for ``local_cwd`` (StructuredOpts parity:
+    reference schedulers/api.py:79-324 + local_scheduler.py:177)."""
+
+    log_dir: Optional[str] = None
+    """base dir for replica logs (default: tmp dir)"""
+
+    auto_set_hip_visible_devices: bool = True
+    """partition host GPUs across replicas via HIP_VISIBLE_DEVICES/ROCR_VISIBLE_DEVICES"""
+
+    prepend_cwd: bool = False
+    """put binaries in cwd ahead of PATH (default: PATH wins, cwd appended)"""
+
+
 class LocalScheduler(Scheduler[PopenRequest]):
     """``local_cwd``: runs replica commands from the current working dir."""
 
@@ -225,15 +242,7 @@ class LocalScheduler(Scheduler[PopenRequest]):
         _INSTANCES.append(self)
 
     def run_opts(self) -> runopts:
-        opts = runopts()
-        opts.add("log_dir", type_=str, default=None,
-                 help="base dir for replica logs (default: tmp dir)")
-        opts.add("auto_set_hip_visible_devices", type_=bool, default=True,
-                 help="partition host GPUs across replicas via "
-                      "HIP_VISIBLE_DEVICES/ROCR_VISIBLE_DEVICES")
-        opts.add("prepend_cwd", type_=bool, default=False,
-                 help="resolve entrypoints against cwd before PATH")
-        return opts
+        return LocalOpts.as_runopts()
 
     # -- dryrun -------------------------------------------------------------
     def _submit_dryrun(self, app: AppDef,
