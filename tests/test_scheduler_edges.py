@@ -175,6 +175,79 @@ class TestRunnerSemantics:
                 == "local_cwd://s/parent1")
 
 
+class TestNumaAffinity:
+    def _fake_sysfs(self, tmp_path, cards):
+        # cards: list of (render_minor, vendor, numa_node)
+        for minor, vendor, numa in cards:
+            d = tmp_path / "class" / "drm" / f"renderD{minor}" / "device"
+            d.mkdir(parents=True)
+            (d / "vendor").write_text(vendor + "\n")
+            (d / "numa_node").write_text(str(numa) + "\n")
+        return str(tmp_path)
+
+    def test_numa_map_from_sysfs(self, tmp_path):
+        from torchx_amd.schedulers.devices import _read_device_numa_map
+
+        root = self._fake_sysfs(tmp_path, [
+            (128, "0x1002", 0), (129, "0x1002", 0),
+            (130, "0x1002", 1), (131, "0x1002", 1),
+            (132, "0x10de", 7),  # non-AMD card is skipped entirely
+        ])
+        assert _read_device_numa_map(root) == {0: 0, 1: 0, 2: 1, 3: 1}
+
+    def test_numa_bind_args(self):
+        from torchx_amd.schedulers.devices import numa_bind_args
+
+        nmap = {0: 0, 1: 0, 2: 1, 3: 1}
+        assert numa_bind_args("0,1", nmap, numactl="/usr/bin/numactl") == [
+            "/usr/bin/numactl", "--cpunodebind=0", "--membind=0",
+        ]
+        assert numa_bind_args("2", nmap, numactl="/usr/bin/numactl") == [
+            "/usr/bin/numactl", "--cpunodebind=1", "--membind=1",
+        ]
+        # spans two nodes -> no binding
+        assert numa_bind_args("1,2", nmap, numactl="/usr/bin/numactl") == []
+        # unknown node -> no binding
+        assert numa_bind_args("9", nmap, numactl="/usr/bin/numactl") == []
+        # numactl missing -> no binding
+        assert numa_bind_args("0", nmap, numactl="") == []
+        assert numa_bind_args(None, nmap, numactl="/usr/bin/numactl") == []
+
+    def test_replica_args_carry_numactl_prefix(self, monkeypatch):
+        from torchx_amd.schedulers import devices as dev_mod
+        from torchx_amd.schedulers.local_scheduler import LocalScheduler
+        from torchx_amd.specs import AppDef, Resource
+
+        monkeypatch.setattr(dev_mod, "hip_device_count", lambda: 8)
+        monkeypatch.setattr(
+            "torchx_amd.schedulers.local_scheduler.numa_bind_args",
+            lambda devs: ["/usr/bin/numactl", "--cpunodebind=0",
+                          "--membind=0"] if devs == "0,1,2,3" else
+                         ["/usr/bin/numactl", "--cpunodebind=1",
+                          "--membind=1"],
+        )
+        monkeypatch.setattr(
+            "torchx_amd.schedulers.local_scheduler.partition_devices",
+            lambda rr, gg: {"trainer": ["0,1,2,3", "4,5,6,7"]},
+        )
+        s = LocalScheduler("t")
+        role = _role(num_replicas=2,
+                     resource=Resource(cpu=8, gpu=4, memMB=1024))
+        app = AppDef(name="x", roles=[role])
+        info = s._submit_dryrun(app, {"auto_set_hip_visible_devices": True,
+                                      "numa_affinity": True})
+        p0, p1 = info.request.role_params["trainer"]
+        assert p0.args[:3] == ["/usr/bin/numactl", "--cpunodebind=0",
+                               "--membind=0"]
+        assert p1.args[:3] == ["/usr/bin/numactl", "--cpunodebind=1",
+                               "--membind=1"]
+        assert p0.env["HIP_VISIBLE_DEVICES"] == "0,1,2,3"
+        # opt off -> clean argv
+        info2 = s._submit_dryrun(app, {"auto_set_hip_visible_devices": True,
+                                       "numa_affinity": False})
+        assert "numactl" not in info2.request.role_params["trainer"][0].args[0]
+
+
 class TestLocalSchedulerEdges:
     def test_bind_mount_rejected(self, tmp_path):
         # local_cwd has no mount support (compat matrix: mounts x)

@@ -70,8 +70,33 @@ def main():
                    help="loop the gu fwd GEMM for this many seconds and "
                         "report the rate per 0.5s window (detects clock "
                         "throttling under sustained load)")
+    p.add_argument("--spread", action="store_true",
+                   help="cycle 32 distinct weight + 8 activation buffers "
+                        "(~9 GB working set) like a real 32-layer step — "
+                        "detects TLB/working-set effects the single-buffer "
+                        "probe misses")
     args = p.parse_args()
     M = args.m
+
+    if args.spread:
+        dev = torch.device("cuda:0")
+        N, K = 28672, 4096
+        ws = [torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+              for _ in range(32)]
+        xs = [torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+              for _ in range(8)]
+        flops = 2.0 * M * N * K
+        idx = [0]
+
+        def fn():
+            i = idx[0]
+            idx[0] += 1
+            return xs[i % 8] @ ws[i % 32].t()
+
+        dt = bench_mm(fn, iters=64, warmup=8)
+        print(json.dumps({"mode": "spread32w8x",
+                          "gu_fwd_TFs": round(flops / dt / 1e12, 1)}))
+        return 0
 
     if args.sustain > 0:
         dev = torch.device("cuda:0")

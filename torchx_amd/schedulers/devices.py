@@ -125,3 +125,75 @@ def device_env(devices: Optional[str]) -> Dict[str, str]:
     if devices is None:
         return {}
     return {HIP_VISIBLE_DEVICES: devices, ROCR_VISIBLE_DEVICES: devices}
+
+
+# ---------------------------------------------------------------------------
+# NUMA affinity: pin each replica's CPU threads and page
+# allocations to the NUMA node its GPUs hang off — the MI355X analog of the
+# reference's CPU-affinity handling around local_scheduler.py:858-950.
+# ---------------------------------------------------------------------------
+
+
+def _read_device_numa_map(sysfs: str = "/sys") -> Dict[int, int]:
+    """HIP device index -> NUMA node, from the DRM render nodes' sysfs
+    (renderD* enumeration order matches the HIP device order)."""
+    import glob
+    import os
+    import re
+
+    result: Dict[int, int] = {}
+    paths = sorted(
+        glob.glob(os.path.join(sysfs, "class/drm/renderD*/device")),
+        key=lambda p: int(re.search(r"renderD(\d+)", p).group(1)),
+    )
+    idx = 0
+    for p in paths:
+        try:
+            with open(os.path.join(p, "vendor")) as f:
+                vendor = f.read().strip()
+        except OSError:
+            continue
+        if vendor != "0x1002":  # AMD
+            continue
+        try:
+            with open(os.path.join(p, "numa_node")) as f:
+                numa = int(f.read().strip())
+        except (OSError, ValueError):
+            numa = -1
+        result[idx] = numa
+        idx += 1
+    return result
+
+
+@functools.lru_cache(maxsize=1)
+def device_numa_map() -> Dict[int, int]:
+    return _read_device_numa_map()
+
+
+def numa_bind_args(
+    devices: Optional[str],
+    numa_map: Optional[Dict[int, int]] = None,
+    numactl: Optional[str] = None,
+) -> List[str]:
+    """``numactl`` argv prefix binding a replica to its GPUs' NUMA node.
+
+    Empty when: no device assignment, the devices span several nodes, the
+    node is unknown (-1), or numactl is not installed.
+    """
+    if not devices:
+        return []
+    if numa_map is None:
+        numa_map = device_numa_map()
+    try:
+        nodes = {numa_map.get(int(d), -1) for d in devices.split(",")}
+    except ValueError:
+        return []
+    if len(nodes) != 1:
+        return []
+    node = nodes.pop()
+    if node < 0:
+        return []
+    numactl = numactl if numactl is not None else shutil.which("numactl")
+    if not numactl:
+        return []
+    return [numactl, f"--cpunodebind={node}", f"--membind={node}"]

@@ -43,7 +43,9 @@ from torchx_amd.specs import (
 from .api import (
     DescribeAppResponse, ListAppResponse, Scheduler, Stream, StructuredOpts,
 )
-from .devices import device_env, hip_device_count, partition_devices
+from .devices import (
+    device_env, hip_device_count, numa_bind_args, partition_devices,
+)
 from .ids import make_unique
 from .streams import Tee
 
@@ -230,6 +232,9 @@ class LocalOpts(StructuredOpts):
     prepend_cwd: bool = False
     """put binaries in cwd ahead of PATH (default: PATH wins, cwd appended)"""
 
+    numa_affinity: bool = True
+    """wrap each replica in numactl --cpunodebind/--membind for the NUMA node its GPUs hang off (no-op without numactl or a single-node assignment)"""
+
 
 class LocalScheduler(Scheduler[PopenRequest]):
     """``local_cwd``: runs replica commands from the current working dir."""
@@ -298,11 +303,14 @@ class LocalScheduler(Scheduler[PopenRequest]):
                 )
                 env.setdefault("PET_LOG_DIR", replica_log)
                 devs = device_assignment.get(role.name)
+                numa_prefix: List[str] = []
                 if devs is not None:
                     env.update(device_env(devs[replica_id]))
+                    if cfg.get("numa_affinity"):
+                        numa_prefix = numa_bind_args(devs[replica_id])
                 params.append(
                     ReplicaParam(
-                        args=[replica.entrypoint, *replica.args],
+                        args=[*numa_prefix, replica.entrypoint, *replica.args],
                         env=env,
                         stdout=os.path.join(replica_log, "stdout.log"),
                         stderr=os.path.join(replica_log, "stderr.log"),
