@@ -70,6 +70,9 @@ def main():
                    help="loop the gu fwd GEMM for this many seconds and "
                         "report the rate per 0.5s window (detects clock "
                         "throttling under sustained load)")
+    p.add_argument("--flat", action="store_true",
+                   help="weights as views into one flat buffer "
+                        "(FlatParams layout)")
     p.add_argument("--spread", action="store_true",
                    help="cycle 32 distinct weight + 8 activation buffers "
                         "(~9 GB working set) like a real 32-layer step — "
@@ -128,10 +131,21 @@ def main():
     ]
 
     dev = torch.device("cuda:0")
+    flat = None
+    if args.flat:
+        # weights as views into one big flat allocation (the FlatParams
+        # situation in the model) — catches pointer-offset effects
+        total = sum(N * K for _, N, K in shapes)
+        flat = torch.randn(total, device=dev, dtype=torch.bfloat16)
     results = []
+    off = 0
     for name, N, K in shapes:
         x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
-        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+        if flat is not None:
+            w = flat[off:off + N * K].view(N, K)
+            off += N * K
+        else:
+            w = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
         wt = w.t().contiguous()  # [K, N]
         dy = torch.randn(M, N, device=dev, dtype=torch.bfloat16)
         flops = 2.0 * M * N * K
@@ -145,7 +159,8 @@ def main():
             "wgrad-TNp": lambda: x.t() @ dy,
         }
         row = {"shape": name, "M": M, "N": N, "K": K,
-               "mode": "cold" if args.cold else "hot"}
+               "mode": ("cold" if args.cold else "hot")
+                       + ("+flat" if args.flat else "")}
         for cname, fn in cases.items():
             if args.cold:
                 dt = bench_mm_cold(fn)
@@ -155,7 +170,8 @@ def main():
         results.append(row)
         print(json.dumps(row), flush=True)
         del x, w, wt, dy
-        torch.cuda.empty_cache()
+        if flat is None:
+            torch.cuda.empty_cache()
 
     # totals at current vs best-layout assignment
     tot_cur = tot_best = 0.0

@@ -111,10 +111,11 @@ rmsnorm_bwd_dx_kernel(
 }
 
 // dw column reduction: grid (row_chunks, H/2048); each block owns a
-// 2048-column tile (256 threads x 8) over DW_CHUNK_ROWS rows, walked 4 at
-// a time so four row-loads are in flight per accumulate (the single-row
-// walk was latency-bound at half roofline).
-#define DW_CHUNK_ROWS 32
+// 2048-column tile (256 threads x 8) over DW_CHUNK_ROWS rows.
+// (Measured: 64-row chunks with a plain row walk beat a 4-wide unrolled
+// 32-row variant 157 vs 282 us total bwd — the unroll's register pressure
+// cost more than the load ILP bought.)
+#define DW_CHUNK_ROWS 64
 
 __global__ void __launch_bounds__(256)
 rmsnorm_bwd_dw_kernel(
@@ -127,26 +128,7 @@ rmsnorm_bwd_dw_kernel(
   const long row0 = (long)blockIdx.x * DW_CHUNK_ROWS;
   const long row1 = min(row0 + DW_CHUNK_ROWS, rows);
   float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  long row = row0;
-  for (; row + 4 <= row1; row += 4) {
-    ushort8 xv[4], dv[4];
-    float r[4];
-    #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      r[u] = invrms[row + u];
-      xv[u] = *(const ushort8*)(x + (row + u) * (long)H + col);
-      dv[u] = *(const ushort8*)(dy + (row + u) * (long)H + col);
-    }
-    #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        acc[j] = fmaf(bf16_to_f32(dv[u][j]) * bf16_to_f32(xv[u][j]), r[u],
-                      acc[j]);
-      }
-    }
-  }
-  for (; row < row1; ++row) {
+  for (long row = row0; row < row1; ++row) {
     const float r = invrms[row];
     ushort8 xv = *(const ushort8*)(x + row * (long)H + col);
     ushort8 dv = *(const ushort8*)(dy + row * (long)H + col);

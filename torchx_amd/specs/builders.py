@@ -97,10 +97,12 @@ def _fn_summary(fn: Callable[..., Any]) -> str:
 def create_args_parser(fn: Callable[..., Any]) -> argparse.ArgumentParser:
     sig = inspect.signature(fn)
     helps = _docstring_param_help(fn)
+    from .file_linter import ComponentHelpFormatter
+
     parser = argparse.ArgumentParser(
         prog=fn.__name__,
         description=_fn_summary(fn),
-        formatter_class=argparse.ArgumentDefaultsHelpFormatter,
+        formatter_class=ComponentHelpFormatter,
     )
     for name, p in sig.parameters.items():
         help_txt = helps.get(name, "")

@@ -9,12 +9,28 @@ describing each argument (missing docs are warnings, not errors).
 
 from __future__ import annotations
 
+import argparse
 import ast
 from dataclasses import dataclass
 from typing import List, Optional
 
 _PRIMITIVES = {"int", "float", "str", "bool"}
-_CONTAINERS = {"List", "Dict", "Optional", "list", "dict"}
+_CONTAINERS = {"List", "Dict", "Optional", "Tuple", "list", "dict", "tuple"}
+
+
+class ComponentHelpFormatter(argparse.RawDescriptionHelpFormatter):
+    """Help formatter for component arg parsers: appends ``(required)`` to
+    required arguments and ``(default: X)`` otherwise (parity:
+    torchx/specs/file_linter.py:33-55 TorchXArgumentHelpFormatter —
+    component functions never have an argument that is both)."""
+
+    def _get_help_string(self, action: argparse.Action) -> str:
+        help_txt = action.help or ""
+        if action.default is argparse.SUPPRESS:  # only --help
+            return help_txt
+        if action.required:
+            return f"{help_txt} (required)".strip()
+        return f"{help_txt} (default: {action.default})".strip()
 
 
 @dataclass
@@ -43,7 +59,19 @@ def _optional_inner(ann: ast.expr) -> Optional[ast.expr]:
     return None
 
 
+def _unwrap_annotated(ann: ast.expr) -> ast.expr:
+    """``Annotated[T, meta...]`` -> ``T`` (metadata carries e.g. short
+    flags; only the leading type is validated)."""
+    if (isinstance(ann, ast.Subscript) and isinstance(ann.value, ast.Name)
+            and ann.value.id == "Annotated"):
+        sl = ann.slice
+        if isinstance(sl, ast.Tuple) and sl.elts:
+            return sl.elts[0]
+    return ann
+
+
 def _type_ok(ann: ast.expr, depth: int = 0) -> bool:
+    ann = _unwrap_annotated(ann)
     inner = _optional_inner(ann)
     if inner is not None:
         return _type_ok(inner, depth)
@@ -55,6 +83,10 @@ def _type_ok(ann: ast.expr, depth: int = 0) -> bool:
             return False
         sl = ann.slice
         elts = sl.elts if isinstance(sl, ast.Tuple) else [sl]
+        if base in ("Dict", "dict") and len(elts) != 2:
+            return False  # dict needs exactly K and V type params
+        if base in ("Tuple", "tuple") and len(elts) < 2:
+            return False  # single-element tuples unsupported for components
         return all(_type_ok(e, depth + 1) for e in elts)
     return False
 
