@@ -70,6 +70,9 @@ def main():
                    help="loop the gu fwd GEMM for this many seconds and "
                         "report the rate per 0.5s window (detects clock "
                         "throttling under sustained load)")
+    p.add_argument("--interleave", action="store_true",
+                   help="alternate the gu GEMM with non-GEMM kernels and "
+                        "time only the GEMM (kernel-transition effects)")
     p.add_argument("--flat", action="store_true",
                    help="weights as views into one flat buffer "
                         "(FlatParams layout)")
@@ -99,6 +102,53 @@ def main():
         dt = bench_mm(fn, iters=64, warmup=8)
         print(json.dumps({"mode": "spread32w8x",
                           "gu_fwd_TFs": round(flops / dt / 1e12, 1)}))
+        return 0
+
+    if args.interleave:
+        # alternate the gu fwd GEMM with non-GEMM kernels (rmsnorm /
+        # attention / a memory sweep) and time ONLY the GEMM via events —
+        # reproduces the in-step situation where fwd GEMMs run at ~1.05
+        # PF/s vs 1.5 isolated. If rates drop here, the gap is a
+        # power/clock state transition between kernel types.
+        import os as _os
+        import sys as _sys
+        _sys.path.insert(0, _os.path.dirname(_os.path.dirname(
+            _os.path.abspath(__file__))))
+        from torchx_amd import ops
+
+        dev = torch.device("cuda:0")
+        N, K = 28672, 4096
+        x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+        flops = 2.0 * M * N * K
+        rx = torch.randn(16384, 4096, device=dev, dtype=torch.bfloat16)
+        rw = torch.randn(4096, device=dev, dtype=torch.bfloat16)
+        qkv = torch.randn(4, 512, 64 * 128, device=dev, dtype=torch.bfloat16)
+        cos, sin = ops.rope_tables(512, 128, device=dev)
+        sweep = torch.empty(2 << 30, device=dev, dtype=torch.uint8)
+
+        fillers = {
+            "none": lambda: None,
+            "rmsnorm": lambda: ops.rmsnorm(rx, rw),
+            "attn": lambda: ops.fused_attention_qkv(qkv, cos, sin, 32, 16),
+            "memsweep": lambda: sweep.fill_(1),
+        }
+        start = torch.cuda.Event(enable_timing=True)
+        end = torch.cuda.Event(enable_timing=True)
+        for fname, filler in fillers.items():
+            total = 0.0
+            iters, warmup = 16, 4
+            for i in range(warmup + iters):
+                filler()
+                start.record()
+                x @ w.t()
+                end.record()
+                torch.cuda.synchronize()
+                if i >= warmup:
+                    total += start.elapsed_time(end) / 1e3
+            rate = flops / (total / iters) / 1e12
+            print(json.dumps({"filler": fname,
+                              "gu_fwd_TFs": round(rate, 1)}), flush=True)
         return 0
 
     if args.sustain > 0:
