@@ -136,12 +136,20 @@ def main() -> int:
     opt = FlatAdamW(flat, lr=3e-4)
 
     B, S = args.micro_batch, args.seq_len
-    tokens = torch.randint(0, cfg.vocab_size, (B, S), device=device)
-    targets = torch.roll(tokens, shifts=-1, dims=1)
+    # rotating pool of synthetic batches (pre-generated: no per-step host
+    # work in the timed region; >1 batch so the loss is not pure
+    # single-batch memorization)
+    n_batches = 4
+    batch_pool = [torch.randint(0, cfg.vocab_size, (B, S), device=device)
+                  for _ in range(n_batches)]
+    target_pool = [torch.roll(t, shifts=-1, dims=1) for t in batch_pool]
+    step_i = [0]
 
     def one_step() -> float:
+        i = step_i[0] % n_batches
+        step_i[0] += 1
         opt.zero_grad()
-        loss = model(tokens, targets)
+        loss = model(batch_pool[i], target_pool[i])
         loss.backward()
         ddp.finish()
         opt.step()
@@ -193,7 +201,7 @@ def main() -> int:
         try:
             # free the bench model's HBM first so the spawned trainer and
             # this process never contend for memory
-            del model, flat, ddp, opt, tokens, targets
+            del model, flat, ddp, opt, batch_pool, target_pool
             if use_gpu:
                 torch.cuda.empty_cache()
             from torchx_amd.utils.launch_latency import measure_launch_latency
