@@ -1,0 +1,171 @@
+#!/usr/bin/env python3
+"""Classifier example trainer — the full end-to-end example app family
+(counterpart of the reference's Lightning example,
+torchx/examples/apps/lightning/train.py, rebuilt without Lightning:
+plain torch DDP loop over our launcher's env contract).
+
+Launch as a distributed app:
+
+    torchx run -s local_cwd dist.ddp -j 1x2 \
+        -m torchx_amd.examples.apps.classifier.train -- \
+        --epochs 2 --output_path /tmp/out
+
+Covers the same surface as the reference example: data module (synthetic
+— no network on the boxes), checkpoint save/resume via fsspec, stage
+profiler, experiment-tracker logging (AppRun), and TorchScript export of
+the trained model.
+"""
+
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import sys
+import tempfile
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+from torchx_amd.distributed import init_pg, rank
+from torchx_amd.examples.apps.classifier.data import (
+    ImageDataModule,
+    SyntheticImages,
+    create_random_data,
+    load_data,
+)
+from torchx_amd.examples.apps.classifier.model import (
+    TinyImageModel,
+    export_inference,
+)
+from torchx_amd.examples.apps.classifier.profiler import StageProfiler
+
+log = logging.getLogger(__name__)
+
+
+def parse_args(argv: List[str]) -> argparse.Namespace:
+    p = argparse.ArgumentParser(description="torchx_amd classifier example")
+    p.add_argument("--epochs", type=int, default=2)
+    p.add_argument("--lr", type=float, default=1e-2)
+    p.add_argument("--batch_size", type=int, default=32)
+    p.add_argument("--dataset_path", type=str, default=None,
+                   help="fsspec URI of a dataset made by create_random_data "
+                        "(default: generate in a tmp dir)")
+    p.add_argument("--output_path", type=str, default=None,
+                   help="fsspec dir for checkpoints + exported model")
+    p.add_argument("--load_path", type=str, default="",
+                   help="checkpoint to resume from")
+    p.add_argument("--num_samples", type=int, default=256)
+    p.add_argument("--skip_export", action="store_true")
+    return p.parse_args(argv)
+
+
+def save_checkpoint(model: TinyImageModel, opt, epoch: int,
+                    out_dir: str) -> str:
+    import fsspec
+
+    fs, path = fsspec.core.url_to_fs(out_dir)
+    fs.makedirs(path, exist_ok=True)
+    target = f"{path}/ckpt_epoch{epoch}.pt"
+    with fs.open(target, "wb") as f:
+        torch.save({"model": model.state_dict(),
+                    "opt": opt.state_dict(), "epoch": epoch}, f)
+    return target
+
+
+def load_checkpoint(model: TinyImageModel, opt, path: str) -> int:
+    import fsspec
+
+    fs, p = fsspec.core.url_to_fs(path)
+    with fs.open(p, "rb") as f:
+        blob = torch.load(f, map_location="cpu", weights_only=True)
+    model.load_state_dict(blob["model"])
+    opt.load_state_dict(blob["opt"])
+    return int(blob["epoch"])
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    logging.basicConfig(level=os.environ.get("LOGLEVEL", "INFO"))
+    args = parse_args(argv if argv is not None else sys.argv[1:])
+    device = init_pg()
+    prof = StageProfiler()
+
+    out_dir = args.output_path or tempfile.mkdtemp(prefix="classifier-out-")
+
+    with prof.stage("data"):
+        if args.dataset_path:
+            ds, num_classes = load_data(args.dataset_path)
+        else:
+            num_classes = 8
+            path = create_random_data(
+                os.path.join(out_dir, "data"), n=args.num_samples,
+                num_classes=num_classes,
+            )
+            ds, _ = load_data(path)
+        dm = ImageDataModule(ds, batch_size=args.batch_size)
+
+    with prof.stage("setup"):
+        model = TinyImageModel(num_classes=num_classes).to(device)
+        if dist.is_initialized() and dist.get_world_size() > 1:
+            model_ddp = torch.nn.parallel.DistributedDataParallel(model)
+        else:
+            model_ddp = model
+        opt = torch.optim.SGD(model.parameters(), lr=args.lr, momentum=0.9)
+        start_epoch = 0
+        if args.load_path:
+            start_epoch = load_checkpoint(model, opt, args.load_path) + 1
+
+    for epoch in range(start_epoch, args.epochs):
+        model_ddp.train()
+        with prof.stage("train"):
+            loader = dm.train_loader()
+            if hasattr(loader.sampler, "set_epoch"):
+                loader.sampler.set_epoch(epoch)
+            for x, y in loader:
+                x, y = x.to(device), y.to(device)
+                opt.zero_grad()
+                loss = torch.nn.functional.cross_entropy(model_ddp(x), y)
+                loss.backward()
+                opt.step()
+        model_ddp.eval()
+        with prof.stage("eval"), torch.no_grad():
+            correct = total = 0
+            for x, y in dm.val_loader():
+                x, y = x.to(device), y.to(device)
+                correct += int((model(x).argmax(-1) == y).sum())
+                total += len(y)
+        acc = correct / max(total, 1)
+        log.info("epoch %d: loss %.4f val_acc %.3f", epoch,
+                 float(loss), acc)
+        if rank() == 0:
+            with prof.stage("checkpoint"):
+                ckpt = save_checkpoint(model, opt, epoch, out_dir)
+                log.info("checkpoint: %s", ckpt)
+
+    if rank() == 0:
+        # experiment tracking: metrics + artifacts through the launcher's
+        # injected TORCHX_JOB_ID / TORCHX_TRACKERS env (tracker/api.py)
+        try:
+            from torchx_amd.tracker.api import AppRun
+
+            run = AppRun.run_from_env()
+            run.add_metadata(val_acc=acc, epochs=args.epochs,
+                             **prof.summary())
+            run.add_artifact("checkpoints", out_dir)
+        except Exception as e:  # noqa: BLE001 — tracking is best-effort
+            log.debug("tracker unavailable: %s", e)
+
+        if not args.skip_export:
+            with prof.stage("export"):
+                path = export_inference(
+                    model, os.path.join(out_dir, "model_scripted.pt"))
+                log.info("exported inference model: %s", path)
+    prof.report()
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
