@@ -110,6 +110,35 @@ def test_cross_entropy(dev):
     assert rel_err(logits.grad, lr_.grad) < 3e-2
 
 
+def test_transpose_bf16(dev):
+    ops = _hip()
+    hip = ops.hip_ops()
+    for R, C in [(256, 128), (4096, 6144), (128, 128256 // 2)]:
+        x = torch.randn(R, C, device=dev, dtype=torch.bfloat16)
+        t = hip.transpose_bf16(x)
+        assert t.shape == (C, R)
+        assert torch.equal(t, x.t().contiguous())
+
+
+def test_fast_linear(dev):
+    ops = _hip()
+    torch.manual_seed(5)
+    x = torch.randn(4, 128, 1024, device=dev, dtype=torch.bfloat16,
+                    requires_grad=True)
+    w = torch.randn(2048, 1024, device=dev, dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = ops.fast_linear(x, w)
+    xr = x.detach().float().requires_grad_(True)
+    wr = w.detach().float().requires_grad_(True)
+    yr = xr @ wr.t()
+    assert rel_err(y, yr) < 2e-2
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    yr.backward(dy.float())
+    assert rel_err(x.grad, xr.grad) < 2e-2
+    assert rel_err(w.grad, wr.grad) < 2e-2
+
+
 def test_fused_linear_cross_entropy(dev):
     ops = _hip()
     T, H, V = 384, 1024, 4096

@@ -71,6 +71,15 @@ def llama_gpu_tiny(vocab: int = 512) -> LlamaConfig:
     )
 
 
+def _lin(mod: nn.Module, x: torch.Tensor) -> torch.Tensor:
+    """Route plain bias-free nn.Linear through ops.fast_linear (dgrad via
+    the transposed-weight operand, ~15% faster); anything else (Fp8Linear,
+    CPU) goes through the module."""
+    if type(mod) is nn.Linear and mod.bias is None and x.is_cuda:
+        return ops.fast_linear(x, mod.weight)
+    return mod(x)
+
+
 class LlamaBlock(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
@@ -93,14 +102,14 @@ class LlamaBlock(nn.Module):
         # attention: rope + flash straight off the packed qkv GEMM output
         # (no split/cat/contiguous traffic — ops.fused_attention_qkv)
         xn = ops.rmsnorm(x, self.attn_norm, cfg.rms_eps)
-        qkv = self.wqkv(xn)
+        qkv = _lin(self.wqkv, xn)
         attn = ops.fused_attention_qkv(
             qkv, cos, sin, cfg.num_heads, cfg.num_kv_heads, causal=True
         )
-        x = x + self.wo(attn.reshape(B, S, cfg.q_dim))
+        x = x + _lin(self.wo, attn.reshape(B, S, cfg.q_dim))
         # mlp: swiglu over the packed gate|up buffer
         xn = ops.rmsnorm(x, self.mlp_norm, cfg.rms_eps)
-        x = x + self.wdown(ops.swiglu_packed(self.wgu(xn)))
+        x = x + _lin(self.wdown, ops.swiglu_packed(_lin(self.wgu, xn)))
         return x
 
 

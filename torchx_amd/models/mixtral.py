@@ -24,7 +24,7 @@ import torch.nn.functional as F
 from torchx_amd import ops
 from torchx_amd.parallel.ep import exchange_counts, expert_all_to_all
 
-from .llama import LlamaConfig
+from .llama import LlamaConfig, _lin
 
 
 @dataclass
@@ -73,7 +73,7 @@ class Expert(nn.Module):
                                bias=False, dtype=torch.bfloat16)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.wdown(ops.swiglu_packed(self.wgu(x)))
+        return _lin(self.wdown, ops.swiglu_packed(_lin(self.wgu, x)))
 
 
 class MoELayer(nn.Module):
@@ -206,11 +206,11 @@ class MixtralBlock(nn.Module):
         cfg = self.cfg
         B, S, H = x.shape
         xn = ops.rmsnorm(x, self.attn_norm, cfg.rms_eps)
-        qkv = self.wqkv(xn)
+        qkv = _lin(self.wqkv, xn)
         attn = ops.fused_attention_qkv(
             qkv, cos, sin, cfg.num_heads, cfg.num_kv_heads, causal=True
         )
-        x = x + self.wo(attn.reshape(B, S, cfg.q_dim))
+        x = x + _lin(self.wo, attn.reshape(B, S, cfg.q_dim))
         xn = ops.rmsnorm(x, self.mlp_norm, cfg.rms_eps)
         return x + self.moe(xn)
 

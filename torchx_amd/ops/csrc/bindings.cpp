@@ -35,6 +35,7 @@ void swiglu_gu_bwd_launch(const void*, const void*, void*, long, long,
 void mfma_probe_launch(const void*, const void*, void*, hipStream_t);
 void fp8_cast_transpose_launch(const void*, void*, void*, void*,
                                const void*, long, long, int, hipStream_t);
+void transpose_bf16_launch(const void*, void*, long, long, hipStream_t);
 }
 
 namespace {
@@ -307,6 +308,18 @@ std::vector<at::Tensor> fp8_cast_transpose(at::Tensor x, at::Tensor scale,
   return {out, out_t, amax};
 }
 
+// ---- bf16 transpose -------------------------------------------------------
+at::Tensor transpose_bf16(at::Tensor x) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.dim() == 2, "transpose_bf16: 2D only");
+  const long R = x.size(0), C = x.size(1);
+  TORCH_CHECK(R % 64 == 0 && C % 64 == 0,
+              "transpose_bf16: dims must be multiples of 64");
+  auto out = at::empty({C, R}, x.options());
+  transpose_bf16_launch(x.data_ptr(), out.data_ptr(), R, C, cur_stream());
+  return out;
+}
+
 // ---- probe ----------------------------------------------------------------
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
   check_bf16(a, "a");
@@ -336,4 +349,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_gu_bwd", &swiglu_gu_bwd);
   m.def("fp8_cast_transpose", &fp8_cast_transpose);
   m.def("mfma_probe", &mfma_probe);
+  m.def("transpose_bf16", &transpose_bf16);
 }

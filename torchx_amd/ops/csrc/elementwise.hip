@@ -289,3 +289,55 @@ extern "C" void swiglu_gu_bwd_launch(const void* dout, const void* gu,
                      (const unsigned short*)gu, (unsigned short*)dgu, n8,
                      I / 8);
 }
+
+// ---------------------------------------------------------------------------
+// bf16 2D transpose: in [R, C] -> out [C, R].  64x64 LDS tiles, 8-wide
+// vector loads AND stores (both global streams coalesced; the strided
+// direction goes through LDS).  Feeds the dgrad-NT path: dx = dy @ Wt^T
+// runs ~15% faster than dy @ W on these shapes, and torch's strided copy
+// for .t().contiguous() is several times slower than this.
+// R, C must be multiples of 64 (all transformer weight dims are).
+// ---------------------------------------------------------------------------
+#define TP 64
+
+__global__ void __launch_bounds__(256)
+transpose_bf16_kernel(const unsigned short* __restrict__ in,
+                      unsigned short* __restrict__ out, int R, int C) {
+  // +1 short of row padding: store-pass lanes hit rows 8 apart, and
+  // 8*(64+1) shorts staggers them 4 banks apart -> conflict-free
+  // (any even padding aliases: 8*(64+p)*2B % 128B == 0 for p in {0,2,4,8})
+  __shared__ unsigned short tile[TP][TP + 1];
+  const int tile_r = blockIdx.y * TP;  // input row of this tile
+  const int tile_c = blockIdx.x * TP;  // input col of this tile
+
+  // load: 256 threads x 8 = 2048 elems/pass, 2 passes for 64x64
+  #pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    const int idx = pass * 2048 + threadIdx.x * 8;
+    const int r = idx / TP;        // 0..63
+    const int c = idx % TP;        // multiple of 8
+    ushort8 v = *(const ushort8*)(in + (long)(tile_r + r) * C + tile_c + c);
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) tile[r][c + j] = v[j];
+  }
+  __syncthreads();
+  // store: thread writes out rows (= input cols) 8-wide
+  #pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    const int idx = pass * 2048 + threadIdx.x * 8;
+    const int oc = idx / TP;       // output row = input col, 0..63
+    const int orr = idx % TP;      // output col base = input row
+    ushort8 v;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = tile[orr + j][oc];
+    *(ushort8*)(out + (long)(tile_c + oc) * R + tile_r + orr) = v;
+  }
+}
+
+extern "C" void transpose_bf16_launch(const void* in, void* out, long R,
+                                      long C, hipStream_t stream) {
+  hipLaunchKernelGGL(transpose_bf16_kernel,
+                     dim3((int)(C / TP), (int)(R / TP)), dim3(256), 0,
+                     stream, (const unsigned short*)in, (unsigned short*)out,
+                     (int)R, (int)C);
+}
