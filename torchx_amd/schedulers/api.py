@@ -109,20 +109,25 @@ class StructuredOpts(Mapping[str, Any]):
     @classmethod
     def field_docstrings(cls) -> Dict[str, str]:
         """Attribute docstrings (the string literal following each field)."""
+        import ast
+        import textwrap
+
         docs: Dict[str, str] = {}
         for klass in reversed(cls.__mro__):
             if not is_dataclass(klass) or klass is StructuredOpts:
                 continue
             try:
-                src = inspect.getsource(klass)
-            except (OSError, TypeError):
+                src = textwrap.dedent(inspect.getsource(klass))
+                body = ast.parse(src).body[0].body
+            except (OSError, TypeError, SyntaxError, IndexError):
                 continue
-            pat = re.compile(
-                r"^\s+(\w+)\s*:[^\n]+\n\s+(?:\"\"\"(.+?)\"\"\"|'''(.+?)''')",
-                re.MULTILINE | re.DOTALL,
-            )
-            for m in pat.finditer(src):
-                docs[m.group(1)] = (m.group(2) or m.group(3)).strip()
+            for stmt, nxt in zip(body, body[1:]):
+                if (isinstance(stmt, ast.AnnAssign)
+                        and isinstance(stmt.target, ast.Name)
+                        and isinstance(nxt, ast.Expr)
+                        and isinstance(nxt.value, ast.Constant)
+                        and isinstance(nxt.value.value, str)):
+                    docs[stmt.target.id] = nxt.value.value.strip()
         hints = typing.get_type_hints(cls)
         for f in fields(cls):
             ftype = _unwrap_optional(hints.get(f.name, str))
