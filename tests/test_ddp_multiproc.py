@@ -9,6 +9,8 @@ from pathlib import Path
 
 import pytest
 
+pytestmark = pytest.mark.subprocess_heavy
+
 REPO = Path(__file__).resolve().parent.parent
 
 WORKER = r"""

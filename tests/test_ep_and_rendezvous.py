@@ -3,8 +3,12 @@ round-trip) and the c10d rendezvous module."""
 
 import os
 import subprocess
+
+import pytest
 import sys
 from pathlib import Path
+
+pytestmark = pytest.mark.subprocess_heavy
 
 REPO = Path(__file__).resolve().parent.parent
 

@@ -5,10 +5,14 @@ gloo DDP run through torch.distributed.run."""
 
 import os
 import subprocess
+
+import pytest
 import sys
 from pathlib import Path
 
 import torch
+
+pytestmark = pytest.mark.subprocess_heavy
 
 REPO = Path(__file__).resolve().parent.parent
 

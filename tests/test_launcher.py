@@ -14,6 +14,8 @@ from torchx_amd.runner import get_runner
 from torchx_amd.specs import AppState
 
 
+pytestmark = pytest.mark.subprocess_heavy
+
 def _wait(runner, handle, timeout=120.0):
     deadline = time.time() + timeout
     while time.time() < deadline:
