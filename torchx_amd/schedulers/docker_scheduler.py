@@ -122,13 +122,23 @@ class DockerScheduler(DockerWorkspaceMixin, Scheduler[DockerJob]):
         import docker.errors
 
         client = self._client()
-        try:
-            client.networks.create(
-                name=NETWORK_NAME, driver="bridge", check_duplicate=True
-            )
-        except docker.errors.APIError as e:
-            if "already exists" not in str(e):
-                raise
+        # cross-process lock: concurrent creates can BOTH succeed in docker
+        # and leave duplicate bridge networks (parity:
+        # docker_scheduler.py:105-126 filelock)
+        import tempfile
+
+        from filelock import FileLock
+
+        lock_path = os.path.join(tempfile.gettempdir(),
+                                 "torchx_amd_docker_network.lock")
+        with FileLock(lock_path, timeout=60):
+            try:
+                client.networks.create(
+                    name=NETWORK_NAME, driver="bridge", check_duplicate=True
+                )
+            except docker.errors.APIError as e:
+                if "already exists" not in str(e):
+                    raise
 
     # -- dryrun -------------------------------------------------------------
     def _submit_dryrun(self, app: AppDef,
