@@ -862,7 +862,12 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
         }
 
         const int qs0 = q0 + 32 * sb;
-        const bool mask_tile = causal || (qs0 + 32 > S) || !krow_valid;
+        // causal masking only bites when SOME (q, k) pair in this
+        // 32q x 32k(own) tile has k > q, i.e. qs0 < kw0 + 32; interior
+        // tiles (q strictly above every owned k) skip the 16-row mask
+        // VALU entirely (it used to run on EVERY causal tile)
+        const bool mask_tile = (causal && qs0 < kw0 + 32)
+                               || (qs0 + 32 > S) || !krow_valid;
         float cv[16];
         #pragma unroll
         for (int r = 0; r < 16; ++r) {
