@@ -585,13 +585,11 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
       #pragma clang loop unroll(disable)
       for (int sb = 0; sb < 2; ++sb) {
         f32x16 acc_s = {}, acc_dp = {};
-        __builtin_amdgcn_s_setprio(1);  // T5: keep the MFMA pipe fed
         #pragma unroll
         for (int c = 0; c < 8; ++c) {
           acc_s = mfma32(kimg_frag(kcur, sb, c), qfrag[c], acc_s);
           acc_dp = mfma32(kimg_frag(vcur, sb, c), dofrag[c], acc_dp);
         }
-        __builtin_amdgcn_s_setprio(0);
         const int k0 = kv0 + 32 * sb;
         const bool mask_tile =
             (causal && k0 + 31 > qw0) || (k0 + 32 > S) || !row_valid;
@@ -611,14 +609,12 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
         mbf16x8 df0, df1;
         cvals_to_frags(ds, hi, &df0, &df1);
         // dQ^T[d, q] += K^T . dS^T
-        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int dt = 0; dt < 4; ++dt) {
           acc_dq[dt] = mfma32(vimg_frag(ktr, dt, 2 * sb), df0, acc_dq[dt]);
           acc_dq[dt] =
               mfma32(vimg_frag(ktr, dt, 2 * sb + 1), df1, acc_dq[dt]);
         }
-        __builtin_amdgcn_s_setprio(0);
       }
     }
 
@@ -857,7 +853,6 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
         // contraction dim of the accumulate MFMAs is the C-row dim);
         // dP[q, k own] for the dK waves
         f32x16 acc_s = {}, acc_dp = {};
-        __builtin_amdgcn_s_setprio(1);  // T5: keep the MFMA pipe fed
         #pragma unroll
         for (int c = 0; c < 8; ++c) {
           acc_s = mfma32(kimg_frag(qn_c, sb, c), kfrag[c], acc_s);
@@ -865,7 +860,6 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
             acc_dp = mfma32(kimg_frag(don_c, sb, c), vfrag[c], acc_dp);
           }
         }
-        __builtin_amdgcn_s_setprio(0);
 
         const int qs0 = q0 + 32 * sb;
         // causal masking only bites when SOME (q, k) pair in this
@@ -896,14 +890,12 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
         // dV[k,d] += P^T . dO  (B rows from dO^T image)
         // dK[k,d] += dS^T . Q  (B rows from Q^T image)
         const char* acc_img = is_dk ? qt_c : dot_c;
-        __builtin_amdgcn_s_setprio(1);
         #pragma unroll
         for (int dt = 0; dt < 4; ++dt) {
           acc[dt] = mfma32(f0, vimg_frag(acc_img, dt, 2 * sb), acc[dt]);
           acc[dt] =
               mfma32(f1, vimg_frag(acc_img, dt, 2 * sb + 1), acc[dt]);
         }
-        __builtin_amdgcn_s_setprio(0);
       }
     }
 
