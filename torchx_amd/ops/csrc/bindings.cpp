@@ -33,6 +33,8 @@ void swiglu_gu_fwd_launch(const void*, void*, long, long, hipStream_t);
 void swiglu_gu_bwd_launch(const void*, const void*, void*, long, long,
                           hipStream_t);
 void mfma_probe_launch(const void*, const void*, void*, hipStream_t);
+void mfma_probe_tr_launch(const void*, const void*, void*, void*,
+                          hipStream_t);
 void fp8_cast_transpose_launch(const void*, void*, void*, void*,
                                const void*, long, long, int, hipStream_t);
 void transpose_bf16_launch(const void*, void*, long, long, hipStream_t);
@@ -321,6 +323,18 @@ at::Tensor transpose_bf16(at::Tensor x) {
 }
 
 // ---- probe ----------------------------------------------------------------
+std::vector<at::Tensor> mfma_probe_tr(at::Tensor a, at::Tensor b) {
+  check_bf16(a, "a");
+  check_bf16(b, "b");
+  TORCH_CHECK(a.size(0) == 32 && a.size(1) == 16, "a must be [32,16]");
+  TORCH_CHECK(b.size(0) == 16 && b.size(1) == 32, "b must be [16,32]");
+  auto c = at::empty({32, 32}, a.options().dtype(at::kFloat));
+  auto raw = at::empty({64, 8}, a.options());
+  mfma_probe_tr_launch(a.data_ptr(), b.data_ptr(), c.data_ptr(),
+                       raw.data_ptr(), cur_stream());
+  return {c, raw};
+}
+
 at::Tensor mfma_probe(at::Tensor a, at::Tensor b) {
   check_bf16(a, "a");
   check_bf16(b, "b");
@@ -349,5 +363,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("swiglu_gu_bwd", &swiglu_gu_bwd);
   m.def("fp8_cast_transpose", &fp8_cast_transpose);
   m.def("mfma_probe", &mfma_probe);
+  m.def("mfma_probe_tr", &mfma_probe_tr);
   m.def("transpose_bf16", &transpose_bf16);
 }
