@@ -36,10 +36,11 @@ def main() -> int:
                    choices=["bf16", "fp8"],
                    help="fp8: OCP e4m3/e5m2 GEMMs via hipBLASLt scaled-mm "
                         "(opt-in; the headline metric is bf16)")
-    p.add_argument("--hipgraph", type=str, default="auto",
+    p.add_argument("--hipgraph", type=str, default="off",
                    choices=["auto", "off"],
                    help="capture fwd+bwd in a hipGraph for 1-GPU runs "
-                        "(optimizer stays eager); off = fully eager")
+                        "(measured neutral: the step is dense big-kernel "
+                        "work; kept as an option)")
     p.add_argument("--launch-latency", type=str, default="auto",
                    choices=["auto", "on", "off"],
                    help="also measure launch->first-step latency through the "
