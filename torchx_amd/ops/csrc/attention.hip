@@ -132,6 +132,41 @@ __device__ __forceinline__ mbf16x8 vimg_frag(const char* vimg, int dt, int ks) {
   return *(const mbf16x8*)(vimg + d * 128 + g * 16);
 }
 
+// ---------------------------------------------------------------------------
+// T10 hardware transpose read: assemble the same fragment a vimg_frag of a
+// transposed image gives — lane holds 8 elements (rows ks*16+(lane>>5)*8+j
+// of the NATURAL 64-row image, at its own column dt*32+(lane&31)) — with
+// two ds_read_tr16_b64 per fragment, straight off the XOR-swizzled natural
+// image.  Kills the explicit LDS transposes (and their images) in the
+// backward kernels.  Per-16-lane-cluster semantics (probe-verified,
+// tools/probe_tr_read.py): lane j addresses the 4-bf16 chunk
+// (row r0 + (j>>2), cols cbase + 4*(j&3)); lane i receives the 4 rows at
+// column cbase + i, packed low-to-high.
+// ---------------------------------------------------------------------------
+typedef short short4_v __attribute__((ext_vector_type(4)));
+typedef __attribute__((address_space(3))) short4_v* lds_s4p;
+
+__device__ __forceinline__ mbf16x8 nat_tr_frag(const char* nat, int ks,
+                                               int dt) {
+  const int lane = threadIdx.x & 63;
+  const int j = lane & 15;
+  const int r_lo = ks * 16 + ((lane >> 5) * 8) + (j >> 2);
+  const int r_hi = r_lo + 4;
+  const int d0 = dt * 32 + ((lane >> 4) & 1) * 16 + 4 * (j & 3);
+  const int g16 = (d0 >> 3) << 4;
+  const int off = (d0 & 7) * 2;
+  const int b_lo = r_lo * 256 + ((g16 ^ ((r_lo & 15) << 4)) | off);
+  const int b_hi = r_hi * 256 + ((g16 ^ ((r_hi & 15) << 4)) | off);
+  short4_v lo = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (lds_s4p)(nat + b_lo));
+  short4_v hi = __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (lds_s4p)(nat + b_hi));
+  uint2_v ul = __builtin_bit_cast(uint2_v, lo);
+  uint2_v uh = __builtin_bit_cast(uint2_v, hi);
+  uint4_v u = {ul[0], ul[1], uh[0], uh[1]};
+  return __builtin_bit_cast(mbf16x8, u);
+}
+
 // Async-stage one 64-row K tile: 16 global_load_lds_dwordx4 per block
 // (16/NW per wave), source-permuted so the lane-linear LDS image lands
 // swizzled (T2 note: swizzle moves to the SOURCE with glds staging).
@@ -492,16 +527,16 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
                    unsigned short* __restrict__ dq,
                    int B, int S, int Hq, int Hkv, float scale, int causal,
                    long q_rs, long kv_rs, long dq_rs) {
-  // v2: KV tiles of 64; K and V natural images arrive by async
-  // global_load_lds (2-deep rings, swizzled); K^T single rot-placed image
-  // reg-staged with the T14 split; exp2-domain probabilities.  80 KiB LDS
-  // -> 2 blocks/CU.
-  __shared__ __align__(16) char smem[4 * KIMG_BYTES + VIMG_BYTES];
+  // v3 (T10): KV tiles of 64; K and V natural images arrive by async
+  // global_load_lds (2-deep rings, swizzled); the dQ-accumulate A-operand
+  // (K^T) is read straight off the natural K image with hardware
+  // transpose reads — no K^T build, no rot image.  64 KiB LDS -> true
+  // 2 blocks/CU.
+  __shared__ __align__(16) char smem[4 * KIMG_BYTES];
   char* kcur = smem;
   char* knxt = smem + KIMG_BYTES;
   char* vcur = smem + 2 * KIMG_BYTES;
   char* vnxt = smem + 3 * KIMG_BYTES;
-  char* ktr = smem + 4 * KIMG_BYTES;
 
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
   const int G = Hq / Hkv;
@@ -556,8 +591,6 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   stage_k_glds(vb, 0, kv_seq_stride, S, vcur);
   asm volatile("s_waitcnt vmcnt(0)");
   __syncthreads();
-  lds_nat_to_tr<256>(kcur, ktr);
-  __syncthreads();
 
   for (int t = 0; t < ntiles; ++t) {
     const int kv0 = t * FKV;
@@ -608,23 +641,21 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
         }
         mbf16x8 df0, df1;
         cvals_to_frags(ds, hi, &df0, &df1);
-        // dQ^T[d, q] += K^T . dS^T
+        // dQ^T[d, q] += K^T . dS^T; K^T fragments by hardware transpose
+        // read from the natural K image (T10)
         #pragma unroll
         for (int dt = 0; dt < 4; ++dt) {
-          acc_dq[dt] = mfma32(vimg_frag(ktr, dt, 2 * sb), df0, acc_dq[dt]);
           acc_dq[dt] =
-              mfma32(vimg_frag(ktr, dt, 2 * sb + 1), df1, acc_dq[dt]);
+              mfma32(nat_tr_frag(kcur, 2 * sb, dt), df0, acc_dq[dt]);
+          acc_dq[dt] =
+              mfma32(nat_tr_frag(kcur, 2 * sb + 1, dt), df1, acc_dq[dt]);
         }
       }
     }
 
     if (has_next) {
       asm volatile("s_waitcnt vmcnt(0)");
-      __syncthreads();   // all waves done reading ktr; nxt glds visible
-      // K^T rebuilt from the just-landed K natural image: an LDS round
-      // trip instead of a second global K stream
-      lds_nat_to_tr<256>(knxt, ktr);
-      __syncthreads();   // ktr visible
+      __syncthreads();   // all waves done reading cur; nxt glds visible
       char* tk = kcur; kcur = knxt; knxt = tk;
       char* tv = vcur; vcur = vnxt; vnxt = tv;
     }
@@ -722,16 +753,15 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     unsigned short* __restrict__ dv,
                     int B, int S, int Hq, int Hkv, float scale, int causal,
                     long q_rs, long kv_rs, long dout_rs, long out_rs) {
-  __shared__ __align__(16) char smem[8 * KIMG_BYTES];
+  // T10: the accumulate B-operands (Q^T / dO^T) are hardware transpose
+  // reads off the natural images — no Q^T/dO^T builds, half the LDS
+  // (64 KiB + lse) -> 2 blocks/CU.
+  __shared__ __align__(16) char smem[4 * KIMG_BYTES];
   __shared__ float lse_buf[2][FKV], del_buf[2][FKV];
   char* qn_c = smem;                       // Q natural cur
   char* don_c = smem + KIMG_BYTES;         // dO natural cur
-  char* qt_c = smem + 2 * KIMG_BYTES;      // Q^T cur
-  char* dot_c = smem + 3 * KIMG_BYTES;     // dO^T cur
-  char* qn_n = smem + 4 * KIMG_BYTES;
-  char* don_n = smem + 5 * KIMG_BYTES;
-  char* qt_n = smem + 6 * KIMG_BYTES;
-  char* dot_n = smem + 7 * KIMG_BYTES;
+  char* qn_n = smem + 2 * KIMG_BYTES;
+  char* don_n = smem + 3 * KIMG_BYTES;
 
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
   const int G = Hq / Hkv;
@@ -813,13 +843,10 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
     }
   };
 
-  // prologue: tile (gh=0, t=t0) — glds the natural images, then build
-  // the transposed images from LDS
+  // prologue: tile (gh=0, t=t0) — glds the natural images
   stage_k_glds8(head_q(0), (long)t0 * FKV, q_seq_stride, S, qn_c);
   stage_k_glds8(head_do(0), (long)t0 * FKV, dout_rs, S, don_c);
   asm volatile("s_waitcnt vmcnt(0)");
-  lds_nat_to_tr_own8(qn_c, qt_c);
-  lds_nat_to_tr_own8(don_c, dot_c);
   stage_lse(0, t0, 0);
   __syncthreads();
 
@@ -887,24 +914,21 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
         mbf16x8 f0, f1;
         cvals_to_frags(cv, hi, &f0, &f1);
 
-        // dV[k,d] += P^T . dO  (B rows from dO^T image)
-        // dK[k,d] += dS^T . Q  (B rows from Q^T image)
-        const char* acc_img = is_dk ? qt_c : dot_c;
+        // dV[k,d] += P^T . dO ; dK[k,d] += dS^T . Q — B fragments by
+        // hardware transpose read from the natural images (T10)
+        const char* acc_img = is_dk ? qn_c : don_c;
         #pragma unroll
         for (int dt = 0; dt < 4; ++dt) {
-          acc[dt] = mfma32(f0, vimg_frag(acc_img, dt, 2 * sb), acc[dt]);
           acc[dt] =
-              mfma32(f1, vimg_frag(acc_img, dt, 2 * sb + 1), acc[dt]);
+              mfma32(f0, nat_tr_frag(acc_img, 2 * sb, dt), acc[dt]);
+          acc[dt] =
+              mfma32(f1, nat_tr_frag(acc_img, 2 * sb + 1, dt), acc[dt]);
         }
       }
     }
 
     if (has_next) {
       asm volatile("s_waitcnt vmcnt(0)");  // own glds of the nxt tiles done
-      // wave-local transpose of the rows THIS wave staged (no pre-barrier)
-      // into the double-buffered tr images; lse double-buffered likewise
-      lds_nat_to_tr_own8(qn_n, qt_n);
-      lds_nat_to_tr_own8(don_n, dot_n);
       stage_lse(ngh, t0 + (idx + 1) % nt_eff, lse_nxt);
       __syncthreads();   // ONE barrier per tile publishes everything
       lse_cur = lse_buf[lse_nxt];
@@ -913,8 +937,6 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
       char* tp;
       tp = qn_c; qn_c = qn_n; qn_n = tp;
       tp = don_c; don_c = don_n; don_n = tp;
-      tp = qt_c; qt_c = qt_n; qt_n = tp;
-      tp = dot_c; dot_c = dot_n; dot_n = tp;
     }
   }
 
