@@ -304,9 +304,10 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
                 float* __restrict__ lse,  // [B,Hq,S]
                 int B, int S, int Hq, int Hkv, float scale, int causal,
                 long q_rs, long kv_rs) {  // per-seq-row element strides
-  // K 2-ring + single V natural scratch (glds) + 2-deep V^T images,
-  // each transposed wave-locally from the scratch (as in the backward)
-  __shared__ __align__(16) char smem[3 * KIMG_BYTES + 2 * VIMG_BYTES];
+  // K and V natural 2-rings (glds); the PV A-operand (V^T) is read off
+  // the natural V image with hardware transpose reads (T10) — no V^T
+  // build, 64 KiB LDS -> true 2 blocks/CU.
+  __shared__ __align__(16) char smem[4 * KIMG_BYTES];
 
   const int BQ = NW * QBLK;
   const int nqt = (S + BQ - 1) / BQ;
@@ -352,15 +353,13 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
 
   char* k0 = smem;                       // K tile t   (cur)
   char* k1 = smem + KIMG_BYTES;          // K tile t+1 (in flight)
-  char* vnat = smem + 2 * KIMG_BYTES;    // V natural glds scratch
-  char* vcur = smem + 3 * KIMG_BYTES;    // V^T cur
-  char* vnxt = vcur + VIMG_BYTES;        // V^T nxt
+  char* vcur = smem + 2 * KIMG_BYTES;    // V natural cur
+  char* vnxt = smem + 3 * KIMG_BYTES;    // V natural nxt
 
-  // prologue: glds K0 + V0 natural, wave-local transpose V0
+  // prologue: glds K0 + V0 natural
   stage_k_glds<NW>(kb, 0, kv_seq_stride, S, k0);
-  stage_k_glds<NW>(vb, 0, kv_seq_stride, S, vnat);
+  stage_k_glds<NW>(vb, 0, kv_seq_stride, S, vcur);
   asm volatile("s_waitcnt vmcnt(0)");
-  lds_nat_to_tr_own<NW>(vnat, vcur);
   __syncthreads();
 
   for (int t = 0; t < ntiles; ++t) {
@@ -370,7 +369,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
       // issue next K/V glds BEFORE compute (T14: their latency hides
       // under this tile's MFMAs; the stream is L2-resident anyway)
       stage_k_glds<NW>(kb, kv0 + FKV, kv_seq_stride, S, k1);
-      stage_k_glds<NW>(vb, kv0 + FKV, kv_seq_stride, S, vnat);
+      stage_k_glds<NW>(vb, kv0 + FKV, kv_seq_stride, S, vnxt);
     }
 
     const bool needed = wave_active && (!causal || kv0 <= qw_max);
@@ -439,13 +438,14 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
       cvals_to_frags(sv, hi, &pf[0], &pf[1]);
       cvals_to_frags(sv + 16, hi, &pf[2], &pf[3]);
 
-      // O^T[d, q] += V^T . P  (A = V^T from the swizzled LDS image)
+      // O^T[d, q] += V^T . P  (A = V^T by hardware transpose read off
+      // the natural V image, T10)
       __builtin_amdgcn_s_setprio(1);
       #pragma unroll
       for (int dt = 0; dt < 4; ++dt) {
         #pragma unroll
         for (int ks = 0; ks < 4; ++ks) {
-          acc_o[dt] = mfma32(vimg_frag(vcur, dt, ks), pf[ks], acc_o[dt]);
+          acc_o[dt] = mfma32(nat_tr_frag(vcur, ks, dt), pf[ks], acc_o[dt]);
         }
       }
       __builtin_amdgcn_s_setprio(0);
@@ -453,11 +453,7 @@ attn_fwd_kernel(const unsigned short* __restrict__ q,
 
     if (has_next) {
       asm volatile("s_waitcnt vmcnt(0)");  // own K/V glds landed
-      // wave-local transpose of the V rows THIS wave staged; one barrier
-      // per tile publishes K(t+1) + V^T(t+1) (the overwritten buffers
-      // were last read before the PREVIOUS barrier)
-      lds_nat_to_tr_own<NW>(vnat, vnxt);
-      __syncthreads();
+      __syncthreads();   // everyone done reading cur; nxt visible
       char* tk = k0; k0 = k1; k1 = tk;
       char* tv = vcur; vcur = vnxt; vnxt = tv;
     }
