@@ -200,6 +200,21 @@ def main():
         dy = torch.randn(M, N, device=dev, dtype=torch.bfloat16)
         flops = 2.0 * M * N * K
 
+        try:
+            sys.path.insert(0, __import__("os").path.dirname(
+                __import__("os").path.dirname(
+                    __import__("os").path.abspath(__file__))))
+            from torchx_amd import ops as _txops
+            _hip = _txops.hip_ops(required=True)
+        except Exception:
+            _hip = None
+
+        def wgrad_via_t():
+            # transpose dy with the LDS-tiled kernel, then run wgrad as a
+            # plain NN GEMM (includes the transpose cost)
+            dyt = _hip.transpose_bf16(dy)
+            return dyt @ x
+
         cases = {
             "fwd-NT": lambda: x @ w.t(),
             "fwd-NN": lambda: x @ wt,
@@ -208,6 +223,8 @@ def main():
             "wgrad-TN": lambda: dy.t() @ x,
             "wgrad-TNp": lambda: x.t() @ dy,
         }
+        if _hip is not None:
+            cases["wgrad-viaT"] = wgrad_via_t
         row = {"shape": name, "M": M, "N": N, "K": K,
                "mode": ("cold" if args.cold else "hot")
                        + ("+flat" if args.flat else "")}
