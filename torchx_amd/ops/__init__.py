@@ -347,7 +347,15 @@ class _FastLinear(torch.autograd.Function):
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         x2 = x.reshape(-1, x.shape[-1])
         dx = (dy2 @ wt.t()).view_as(x)
-        dw = dy2.t() @ x2
+        # wgrad: for narrow outputs (qkv/wo/down shapes) transposing dy
+        # once and running the wgrad as a plain NN GEMM beats the TN form
+        # by ~15% incl. the transpose (tools/gemm_probe.py wgrad-viaT);
+        # fat outputs (gu, lm_head) are neutral and skip the transpose
+        M, N = dy2.shape
+        if N <= 8192 and M % 64 == 0 and N % 64 == 0:
+            dw = hip.transpose_bf16(dy2) @ x2
+        else:
+            dw = dy2.t() @ x2
         return dx, dw
 
 
