@@ -169,11 +169,15 @@ __global__ void rope_qkv_kernel(
   typedef unsigned short us4 __attribute__((ext_vector_type(4)));
   typedef float f4 __attribute__((ext_vector_type(4)));
   const int half_d = 64;
-  for (long p4 = grid_stride_begin(); p4 < np4; p4 += grid_stride()) {
-    const long hp = p4 >> 4;           // (row, head)
-    const int i = (int)(p4 & 15) * 4;  // pair index within head
-    const long row = hp / nh;
-    const int h = (int)(hp % nh);
+  // 2D grid: x covers (head, pair-group) with shifts/masks only, y strides
+  // rows — the old flat grid-stride form paid a 64-bit div/mod (by nh=40,
+  // not a power of two) on every thread
+  const long hp4 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (hp4 >= (long)nh * 16) return;
+  const int h = (int)(hp4 >> 4);
+  const int i = (int)(hp4 & 15) * 4;
+  const long rows = np4 / ((long)nh * 16);
+  for (long row = blockIdx.y; row < rows; row += gridDim.y) {
     const int pos = (int)(row % seq_len);
     const long base = row * row_stride + (long)h * 128;
     us4 u1 = *(const us4*)(x + base + i);
@@ -199,10 +203,9 @@ extern "C" void rope_qkv_launch(
     hipStream_t stream) {
   const long np4 = rows * nh * 16;
   const int block = 256;
-  long grid = (np4 + block - 1) / block;
-  if (grid > 65535L * 8) grid = 65535L * 8;
-  if (grid == 0) grid = 1;
-  hipLaunchKernelGGL(rope_qkv_kernel, dim3((int)grid), dim3(block), 0,
+  const int gx = (int)(((long)nh * 16 + block - 1) / block);
+  const int gy = (int)((rows < 65535) ? (rows > 0 ? rows : 1) : 65535);
+  hipLaunchKernelGGL(rope_qkv_kernel, dim3(gx, gy), dim3(block), 0,
                      stream, (const unsigned short*)x, (unsigned short*)y,
                      (const float*)cos_t, (const float*)sin_t, np4, nh,
                      row_stride, seq_len, sign);
@@ -213,13 +216,17 @@ extern "C" void rope_qkv_launch(
 // with g = gu[:, :I], u = gu[:, I:].  Avoids the two .contiguous() copies
 // of the chunked views and the torch.cat in backward.
 // ---------------------------------------------------------------------------
+// 2D grid (cols x rows): the old 1D grid-stride form spent a 64-bit
+// integer division (p / i8, i8 = I/8 = 1792 for llama) on EVERY thread —
+// ~25% over the HBM roofline on a pure streaming kernel.
 __global__ void swiglu_gu_fwd_kernel(
     const unsigned short* __restrict__ gu,
     unsigned short* __restrict__ out,
-    long n8,     // rows * I / 8
+    long rows,
     long i8) {   // I / 8
-  for (long p = grid_stride_begin(); p < n8; p += grid_stride()) {
-    const long row = p / i8, c8 = p % i8;
+  const long c8 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c8 >= i8) return;
+  for (long row = blockIdx.y; row < rows; row += gridDim.y) {
     const long base = row * 2 * i8 * 8 + c8 * 8;
     ushort8 gv = *(const ushort8*)(gu + base);
     ushort8 uv = *(const ushort8*)(gu + base + i8 * 8);
@@ -232,7 +239,7 @@ __global__ void swiglu_gu_fwd_kernel(
           1.0f / (1.0f + __builtin_amdgcn_exp2f(-1.44269504f * gf));
       ov[j] = f32_to_bf16(gf * sig * uf);
     }
-    *(ushort8*)(out + p * 8) = ov;
+    *(ushort8*)(out + (row * i8 + c8) * 8) = ov;
   }
 }
 
@@ -240,11 +247,12 @@ __global__ void swiglu_gu_bwd_kernel(
     const unsigned short* __restrict__ dout,
     const unsigned short* __restrict__ gu,
     unsigned short* __restrict__ dgu,
-    long n8, long i8) {
-  for (long p = grid_stride_begin(); p < n8; p += grid_stride()) {
-    const long row = p / i8, c8 = p % i8;
+    long rows, long i8) {
+  const long c8 = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c8 >= i8) return;
+  for (long row = blockIdx.y; row < rows; row += gridDim.y) {
     const long base = row * 2 * i8 * 8 + c8 * 8;
-    ushort8 dov = *(const ushort8*)(dout + p * 8);
+    ushort8 dov = *(const ushort8*)(dout + (row * i8 + c8) * 8);
     ushort8 gv = *(const ushort8*)(gu + base);
     ushort8 uv = *(const ushort8*)(gu + base + i8 * 8);
     ushort8 dgv, duv;
@@ -266,28 +274,26 @@ __global__ void swiglu_gu_bwd_kernel(
 
 extern "C" void swiglu_gu_fwd_launch(const void* gu, void* out, long rows,
                                      long I, hipStream_t stream) {
-  const long n8 = rows * I / 8;
+  const long i8 = I / 8;
   const int block = 256;
-  long grid = (n8 + block - 1) / block;
-  if (grid > 65535L * 8) grid = 65535L * 8;
-  if (grid == 0) grid = 1;
-  hipLaunchKernelGGL(swiglu_gu_fwd_kernel, dim3((int)grid), dim3(block), 0,
+  const int gx = (int)((i8 + block - 1) / block);
+  const int gy = (int)((rows < 65535) ? (rows > 0 ? rows : 1) : 65535);
+  hipLaunchKernelGGL(swiglu_gu_fwd_kernel, dim3(gx, gy), dim3(block), 0,
                      stream, (const unsigned short*)gu, (unsigned short*)out,
-                     n8, I / 8);
+                     rows, i8);
 }
 
 extern "C" void swiglu_gu_bwd_launch(const void* dout, const void* gu,
                                      void* dgu, long rows, long I,
                                      hipStream_t stream) {
-  const long n8 = rows * I / 8;
+  const long i8 = I / 8;
   const int block = 256;
-  long grid = (n8 + block - 1) / block;
-  if (grid > 65535L * 8) grid = 65535L * 8;
-  if (grid == 0) grid = 1;
-  hipLaunchKernelGGL(swiglu_gu_bwd_kernel, dim3((int)grid), dim3(block), 0,
+  const int gx = (int)((i8 + block - 1) / block);
+  const int gy = (int)((rows < 65535) ? (rows > 0 ? rows : 1) : 65535);
+  hipLaunchKernelGGL(swiglu_gu_bwd_kernel, dim3(gx, gy), dim3(block), 0,
                      stream, (const unsigned short*)dout,
-                     (const unsigned short*)gu, (unsigned short*)dgu, n8,
-                     I / 8);
+                     (const unsigned short*)gu, (unsigned short*)dgu, rows,
+                     i8);
 }
 
 // ---------------------------------------------------------------------------
