@@ -492,16 +492,26 @@ attn_bwd_pre_kernel(const unsigned short* __restrict__ dout,
                     float* __restrict__ delta,  // [B,Hq,S]
                     long rows,  // B*S*Hq
                     int S, int Hq) {
-  // one wave per row; lane covers 2 elements of HD=128
-  const long row = (long)blockIdx.x * 4 + threadIdx.x / 64;
+  // 16 lanes per row x 8 elems (16-B loads; the 1-row-per-wave version's
+  // 4-B loads ran 3x off roofline); 4 rows per wave, 16 per block
+  const long row = (long)blockIdx.x * 16 + threadIdx.x / 16;
   if (row >= rows) return;
-  const int lane = threadIdx.x & 63;
-  const unsigned short* dp = dout + row * HD + lane * 2;
-  const unsigned short* op = o + row * HD + lane * 2;
-  float acc = bf16_to_f32(dp[0]) * bf16_to_f32(op[0]) +
-              bf16_to_f32(dp[1]) * bf16_to_f32(op[1]);
-  acc = wave_reduce_sum(acc);
-  if (lane == 0) {
+  const int sl = threadIdx.x & 15;
+  const unsigned short* dp = dout + row * HD + sl * 8;
+  const unsigned short* op = o + row * HD + sl * 8;
+  ushort8 dv = *(const ushort8*)dp;
+  ushort8 ov = *(const ushort8*)op;
+  float acc = 0.f;
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    acc = fmaf(bf16_to_f32(dv[j]), bf16_to_f32(ov[j]), acc);
+  }
+  // reduce across the 16-lane group
+  #pragma unroll
+  for (int off = 8; off >= 1; off >>= 1) {
+    acc += __shfl_down(acc, off, 64);
+  }
+  if (sl == 0) {
     const long b = row / ((long)S * Hq);
     const long s = (row / Hq) % S;
     const long h = row % Hq;
@@ -1005,7 +1015,7 @@ extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
                                 long q_rs, long kv_rs, long dqkv_q_rs,
                                 long dqkv_kv_rs, hipStream_t stream) {
   const long rows = (long)B * S * Hq;
-  hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3((int)((rows + 3) / 4)),
+  hipLaunchKernelGGL(attn_bwd_pre_kernel, dim3((int)((rows + 15) / 16)),
                      dim3(256), 0, stream, (const unsigned short*)dout,
                      (const unsigned short*)o, (float*)delta, rows, S, Hq);
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
