@@ -509,7 +509,7 @@ attn_bwd_pre_kernel(const unsigned short* __restrict__ dout,
   // reduce across the 16-lane group
   #pragma unroll
   for (int off = 8; off >= 1; off >>= 1) {
-    acc += __shfl_down(acc, off, 64);
+    acc += __shfl_down(acc, off, 16);  // width 16: stay within the group
   }
   if (sl == 0) {
     const long b = row / ((long)S * Hq);
