@@ -135,6 +135,34 @@ def _skip_if_dup_gpu(out):
                     + text[-500:])
 
 
+def test_rccl_world1_ep_all_to_all():
+    """The EP all_to_all_single RCCL branch executes on hardware with a
+    world-1 communicator (two-ranks-one-device is refused by this RCCL —
+    see DUP_GPU_MARKERS — so the multi-rank exchange itself is covered by
+    gloo on CPU and by the driver's 8-GPU run)."""
+    import torch
+    import torch.distributed as dist
+
+    from torchx_amd.parallel.ep import exchange_counts, expert_all_to_all
+
+    store = dist.TCPStore("127.0.0.1", 0, is_master=True,
+                          wait_for_workers=False)
+    dist.init_process_group("nccl", store=store, rank=0, world_size=1)
+    try:
+        dev = torch.device("cuda", 0)
+        send = torch.tensor([5], device=dev)
+        recv = exchange_counts(send)
+        assert recv.tolist() == [5]
+        x = torch.randn(5, 64, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True)
+        out = expert_all_to_all(x, [5], [5], None)  # nccl branch
+        assert torch.equal(out, x.detach())
+        out.sum().backward()
+        assert x.grad is not None
+    finally:
+        dist.destroy_process_group()
+
+
 def test_rccl_two_ranks_one_gpu_ddp_and_ep(tmp_path):
     script = tmp_path / "w.py"
     script.write_text(WORKER % {"repo": str(REPO)})
