@@ -87,6 +87,65 @@ def test_flatddp_gloo_world2(tmp_path):
     assert "DDP_MULTIPROC_OK" in out, out[-4000:]
 
 
+DIRECT_GRAD_WORKER = r"""
+import os, sys
+sys.path.insert(0, %(repo)r)
+import torch
+import torch.distributed as dist
+from torchx_amd.models.llama import llama_tiny, LlamaModel
+from torchx_amd.parallel import FlatParams, FlatDDP, FlatAdamW
+from torchx_amd.parallel.ddp import grad_ready
+
+dist.init_process_group("gloo")
+rank = dist.get_rank()
+torch.manual_seed(7)
+
+cfg = llama_tiny()
+model = LlamaModel(cfg)
+flat = FlatParams(model, torch.device("cpu"))
+ddp = FlatDDP(flat, bucket_bytes=1 << 18)
+opt = FlatAdamW(flat, lr=1e-3)
+
+# simulate fast_linear's DIRECT grad production: write each param's grad
+# slot in place (rank-dependent) and notify the bucket hook manually —
+# torch's post-accumulate hook never fires on this path
+opt.zero_grad()
+for slot in flat.slots:
+    p = slot.param
+    with torch.no_grad():
+        p.grad.add_(torch.full_like(p, float(rank + 1)))
+    grad_ready(p)
+ddp.finish()
+
+# after finish, every grad must be the cross-rank mean: (1+2)/2 = 1.5
+for slot in flat.slots:
+    g = slot.param.grad.float()
+    assert torch.allclose(g, torch.full_like(g, 1.5), atol=1e-2), (
+        slot.name if hasattr(slot, "name") else "param", g.mean())
+if rank == 0:
+    print("DIRECT_GRAD_OK", flush=True)
+dist.destroy_process_group()
+"""
+
+
+def test_direct_grad_ready_bucket_path(tmp_path):
+    """grad_ready() (the manual bucket notification fast_linear's direct
+    wgrad accumulation uses) drives the same bucketed all-reduce as the
+    torch post-accumulate hook — exactly what the 8-GPU run exercises."""
+    script = tmp_path / "worker.py"
+    script.write_text(DIRECT_GRAD_WORKER % {"repo": str(REPO)})
+    env = dict(os.environ)
+    env["PYTHONPATH"] = str(REPO) + os.pathsep + env.get("PYTHONPATH", "")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes", "1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--standalone", str(script)],
+        env=env, capture_output=True, text=True, timeout=300, cwd=str(REPO),
+    )
+    assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-3000:]
+    assert "DIRECT_GRAD_OK" in out.stdout + out.stderr
+
+
 def test_agent_two_nodes(tmp_path):
     """Two agent processes (nnodes=2) rendezvous and form one world of 4
     (the multi-node path of the launcher, on one host via 127.0.0.1)."""

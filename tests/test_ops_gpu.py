@@ -139,6 +139,30 @@ def test_fast_linear(dev):
     assert rel_err(w.grad, wr.grad) < 2e-2
 
 
+def test_fast_linear_direct_grad_accumulate(dev):
+    """With a pre-existing (flat-buffer-style) grad slot, fast_linear's
+    wgrad accumulates in place via addmm_ and never materializes dw."""
+    ops = _hip()
+    torch.manual_seed(9)
+    for N in (2048, 12288):  # via-transpose and TN paths
+        x = torch.randn(256, 1024, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True)
+        w = torch.randn(N, 1024, device=dev, dtype=torch.bfloat16,
+                        requires_grad=True) * 0.05
+        w = w.detach().requires_grad_(True)
+        seed_grad = torch.randn_like(w) * 0.01
+        w.grad = seed_grad.clone()
+        y = ops.fast_linear(x, w)
+        dy = torch.randn_like(y)
+        y.backward(dy)
+        # reference: seed + dy^T @ x in fp32
+        ref = seed_grad.float() + dy.float().t() @ x.detach().float()
+        err = (w.grad.float() - ref).abs().max().item()
+        scale = ref.abs().max().item()
+        assert err < 3e-2 * scale, (N, err, scale)
+        assert x.grad is not None
+
+
 def test_fused_linear_cross_entropy(dev):
     ops = _hip()
     T, H, V = 384, 1024, 4096

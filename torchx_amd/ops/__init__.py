@@ -352,10 +352,23 @@ class _FastLinear(torch.autograd.Function):
         # by ~15% incl. the transpose (tools/gemm_probe.py wgrad-viaT);
         # fat outputs (gu, lm_head) are neutral and skip the transpose
         M, N = dy2.shape
-        if N <= 8192 and M % 64 == 0 and N % 64 == 0:
-            dw = hip.transpose_bf16(dy2) @ x2
-        else:
-            dw = dy2.t() @ x2
+        via_t = N <= 8192 and M % 64 == 0 and N % 64 == 0
+        # direct accumulation: when the param already has a (flat-buffer)
+        # grad slot, the wgrad GEMM accumulates straight into it with
+        # beta=1 — no dw buffer write/read and no AccumulateGrad RMW pass.
+        # Safe because each weight is consumed once per step; the DDP
+        # bucket hook is notified manually (grad_ready).
+        wg = w.grad if w.is_leaf else None
+        if wg is not None and wg.is_contiguous():
+            if via_t:
+                wg.addmm_(hip.transpose_bf16(dy2), x2)
+            else:
+                wg.addmm_(dy2.t(), x2)
+            from torchx_amd.parallel.ddp import grad_ready
+
+            grad_ready(w)
+            return dx, None
+        dw = (hip.transpose_bf16(dy2) @ x2) if via_t else (dy2.t() @ x2)
         return dx, dw
 
 

@@ -25,6 +25,19 @@ import torch.distributed as dist
 from .flat import FlatParams
 
 
+# params whose grads are produced by DIRECT in-place accumulation (e.g.
+# fast_linear's wgrad addmm_ into the flat slot) never trigger torch's
+# post-accumulate hook; the producer calls grad_ready(param) instead so
+# bucket overlap still works.  id(param) -> FlatDDP.
+_PARAM_DDP: Dict[int, "FlatDDP"] = {}
+
+
+def grad_ready(p: torch.Tensor) -> None:
+    ddp = _PARAM_DDP.get(id(p))
+    if ddp is not None:
+        ddp._hook(p)
+
+
 @dataclass
 class _Bucket:
     group: str
@@ -99,6 +112,7 @@ class FlatDDP:
     def _register_hooks(self) -> None:
         for slot in self.flat.slots:
             slot.param.register_post_accumulate_grad_hook(self._hook)
+            _PARAM_DDP[id(slot.param)] = self
 
     def _broadcast_params(self) -> None:
         for g, p16, _ in self.flat.groups():
