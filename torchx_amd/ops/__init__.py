@@ -360,10 +360,14 @@ class _FastLinear(torch.autograd.Function):
         # bucket hook is notified manually (grad_ready).
         wg = w.grad if w.is_leaf else None
         if wg is not None and wg.is_contiguous():
-            if via_t:
-                wg.addmm_(hip.transpose_bf16(dy2), x2)
+            a = hip.transpose_bf16(dy2) if via_t else dy2.t()
+            if getattr(w, "_wgrad_fresh", False):
+                # first touch since zero_grad: overwrite (beta=0) — skips
+                # both the zeroed-C read in the GEMM and staleness
+                torch.matmul(a, x2, out=wg)
+                w._wgrad_fresh = False
             else:
-                wg.addmm_(dy2.t(), x2)
+                wg.addmm_(a, x2)
             from torchx_amd.parallel.ddp import grad_ready
 
             grad_ready(w)

@@ -90,6 +90,10 @@ class FlatParams:
     def zero_grad(self) -> None:
         for g in self.flat_grad.values():
             g.zero_()
+        # first-touch flag: fast_linear's direct wgrad writes its slot
+        # with beta=0 (no C read) on the first touch after a zero_grad
+        for slot in self.slots:
+            slot.param._wgrad_fresh = True
 
     def groups(self) -> Iterator[Tuple[str, torch.Tensor, torch.Tensor]]:
         for g in self.flat_p16:
