@@ -137,7 +137,7 @@ def main(argv: Optional[List[str]] = None) -> int:
                 total += len(y)
         acc = correct / max(total, 1)
         log.info("epoch %d: loss %.4f val_acc %.3f", epoch,
-                 float(loss), acc)
+                 float(loss.detach()), acc)
         if rank() == 0:
             with prof.stage("checkpoint"):
                 ckpt = save_checkpoint(model, opt, epoch, out_dir)
