@@ -428,6 +428,11 @@ class LocalScheduler(Scheduler[PopenRequest]):
         should_tail: bool = False,
         streams: Optional[Stream] = None,
     ) -> Iterable[str]:
+        if since or until:
+            log.warning(
+                "since/until are ignored by local_cwd log_iter; all log "
+                "lines will be returned (parity: local_scheduler.py:1082)"
+            )
         app = self._apps.get(app_id)
         if app is None:
             raise ValueError(f"unknown app {app_id}")
