@@ -55,6 +55,8 @@ def _component_defaults() -> Dict[str, Dict[str, str]]:
 
 
 def cmd_run(args: argparse.Namespace) -> int:
+    # context name for downstream tracking (parity: cli/cmd_run.py:446)
+    os.environ.setdefault("TORCHX_CONTEXT_NAME", "cli_run")
     stdin_scheduler_args: Optional[Dict[str, object]] = None
     if getattr(args, "stdin", False):
         # JSON run spec from stdin: {"component": ..., "component_args":
