@@ -36,6 +36,10 @@ def main() -> int:
                    choices=["bf16", "fp8"],
                    help="fp8: OCP e4m3/e5m2 GEMMs via hipBLASLt scaled-mm "
                         "(opt-in; the headline metric is bf16)")
+    p.add_argument("--zero1", action="store_true",
+                   help="ZeRO-1: reduce-scatter grads + sharded optimizer "
+                        "state + all-gather params (default: FlatDDP "
+                        "bucketed all-reduce)")
     p.add_argument("--hipgraph", type=str, default="off",
                    choices=["auto", "off"],
                    help="capture fwd+bwd in a hipGraph for 1-GPU runs "
@@ -137,8 +141,14 @@ def main() -> int:
 
             convert_to_fp8(model)
         flat = FlatParams(model, device)
-        ddp = FlatDDP(flat)
-    opt = FlatAdamW(flat, lr=3e-4)
+        ddp = FlatDDP(flat) if not args.zero1 else None
+    if args.zero1 and not is_moe:
+        from torchx_amd.parallel import FlatZeRO1
+
+        opt = FlatZeRO1(flat, lr=3e-4)
+        ddp = opt  # .finish() is a no-op; reduction happens in step()
+    else:
+        opt = FlatAdamW(flat, lr=3e-4)
 
     B, S = args.micro_batch, args.seq_len
     # rotating pool of synthetic batches (pre-generated: no per-step host
