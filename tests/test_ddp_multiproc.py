@@ -204,6 +204,17 @@ for g, p16, _ in flat.groups():
     dist.broadcast(other, src=0)
     assert torch.equal(other, p16), g
 
+# consolidation: gather both shards' state dicts and merge -> must match
+# the full FlatAdamW state from the reference run
+sd = z.state_dict()
+gathered = [None, None]
+dist.all_gather_object(gathered, sd)
+full = FlatZeRO1.consolidate(gathered)
+for g, st in opt.state.items():
+    for k in ("p32", "m", "v"):
+        d = (full["state"][g][k] - st[k]).abs().max().item()
+        assert d < 1e-2, (g, k, d)
+
 if rank == 0:
     print("ZERO1_OK", flush=True)
 dist.destroy_process_group()

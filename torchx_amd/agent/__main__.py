@@ -59,15 +59,34 @@ class Worker:
         self.pumps = pumps
 
 
-def _pump(src, dst, prefix: str) -> threading.Thread:
+def _pump(src, dst, prefix: str,
+          file_path: Optional[str] = None) -> threading.Thread:
+    """Prefix-tee a worker stream to the agent's stream AND (torchrun
+    --tee parity) a per-rank file under the log dir."""
+
     def run() -> None:
+        f = None
+        if file_path:
+            try:
+                os.makedirs(os.path.dirname(file_path), exist_ok=True)
+                f = open(file_path, "ab")
+            except OSError:
+                f = None
         for line in iter(src.readline, b""):
             try:
                 dst.buffer.write(prefix.encode() + line)
                 dst.flush()
             except ValueError:
                 break
+            if f is not None:
+                try:
+                    f.write(line)
+                    f.flush()
+                except OSError:
+                    f = None
         src.close()
+        if f is not None:
+            f.close()
 
     t = threading.Thread(target=run, daemon=True)
     t.start()
@@ -129,8 +148,13 @@ def spawn_workers(args, rdzv: RendezvousResult, restart_count: int,
         )
         pumps = []
         if args.tee:
-            pumps.append(_pump(proc.stdout, sys.stdout, f"[rank{grank}]: "))
-            pumps.append(_pump(proc.stderr, sys.stderr, f"[rank{grank}]: "))
+            # per-rank files under the log dir (torchrun --tee/PET_LOG_DIR
+            # layout: <log_dir>/<restart>/<local_rank>/std{out,err}.log)
+            rank_dir = os.path.join(log_dir, str(restart_count), str(lr))
+            pumps.append(_pump(proc.stdout, sys.stdout, f"[rank{grank}]: ",
+                               os.path.join(rank_dir, "stdout.log")))
+            pumps.append(_pump(proc.stderr, sys.stderr, f"[rank{grank}]: ",
+                               os.path.join(rank_dir, "stderr.log")))
         workers.append(Worker(proc, lr, grank, error_file, pumps))
     return workers
 

@@ -145,8 +145,36 @@ class FlatZeRO1:
             "lr": self.lr,
             "rank": self.rank,
             "world_size": self.ws,
+            "local_groups": sorted(self.local_groups),
             "state": {g: dict(st) for g, st in self.state.items()},
         }
+
+    @staticmethod
+    def consolidate(shards) -> Dict:
+        """Merge per-rank ZeRO-1 ``state_dict``s into one full
+        FlatAdamW-compatible state dict (for world-size changes or
+        single-process analysis). ``shards`` is the list of all ranks'
+        dicts, any order. Local (EP-sharded) groups keep rank 0's copy —
+        they are rank-local by construction."""
+        shards = sorted(shards, key=lambda s: s["rank"])
+        ws = shards[0]["world_size"]
+        if len(shards) != ws or [s["rank"] for s in shards] != list(range(ws)):
+            raise ValueError(
+                f"need all {ws} rank shards exactly once, got ranks "
+                f"{[s.get('rank') for s in shards]}"
+            )
+        local = set(shards[0].get("local_groups", []))
+        state: Dict[str, Dict[str, torch.Tensor]] = {}
+        for g in shards[0]["state"]:
+            if g in local:
+                state[g] = dict(shards[0]["state"][g])
+            else:
+                state[g] = {
+                    k: torch.cat([s["state"][g][k] for s in shards])
+                    for k in shards[0]["state"][g]
+                }
+        return {"step": shards[0]["step"], "lr": shards[0]["lr"],
+                "state": state}
 
     def load_state_dict(self, sd: Dict) -> None:
         assert sd.get("world_size", 1) == self.ws, (
