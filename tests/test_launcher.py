@@ -105,6 +105,13 @@ def test_ddp_gloo_world2(tmp_path):
             f"{status}\nlogs:\n" + "\n".join(lines[-30:])
         )
         assert any("computed world size = 2" in ln for ln in lines), lines[-30:]
+        # agent wrote per-rank stdout files (torchrun --tee layout:
+        # <PET_LOG_DIR>/<restart>/<local_rank>/stdout.log)
+        import glob
+
+        rank_logs = glob.glob(str(tmp_path) + "/**/0/0/stdout.log",
+                              recursive=True)
+        assert rank_logs, list(tmp_path.rglob("*.log"))
 
 
 def test_ddp_elastic_restart(tmp_path):
