@@ -694,14 +694,15 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
 // ---------------------------------------------------------------------------
 // 512-thread (8-wave) staging variants: same image formats as the 4-wave
 // helpers, work split across twice the threads.
+template <int TFKV = FKV>
 __device__ __forceinline__ void stage_k_glds8(
     const unsigned short* __restrict__ kb, long kv0, long stride_elems,
     int S, char* kimg) {
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x & 63;
   #pragma unroll
-  for (int j = 0; j < 2; ++j) {
-    const int i = wave * 2 + j;
+  for (int j = 0; j < TFKV / 32; ++j) {
+    const int i = wave * (TFKV / 32) + j;
     const int row = i * 4 + (lane >> 4);
     const int colbyte = ((lane & 15) * 16) ^ ((row & 15) << 4);
     long srow = kv0 + row;
@@ -748,6 +749,7 @@ __device__ __forceinline__ void lds_nat_to_tr_own8(const char* nat,
   }
 }
 
+template <int TFKV>
 __global__ void __launch_bounds__(512, 2)
 attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
                     const unsigned short* __restrict__ k,
@@ -762,12 +764,14 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   // T10: the accumulate B-operands (Q^T / dO^T) are hardware transpose
   // reads off the natural images — no Q^T/dO^T builds, half the LDS
   // (64 KiB + lse) -> 2 blocks/CU.
-  __shared__ __align__(16) char smem[4 * KIMG_BYTES];
-  __shared__ float lse_buf[2][FKV], del_buf[2][FKV];
+  // TFKV=64: 64 KiB LDS -> 2 blocks/CU; TFKV=128: 128 KiB, 1 block but
+  // half the barriers/loop iterations (A/B via TORCHX_AMD_DKV_FKV)
+  __shared__ __align__(16) char smem[4 * (TFKV * 256)];
+  __shared__ float lse_buf[2][TFKV], del_buf[2][TFKV];
   char* qn_c = smem;                       // Q natural cur
-  char* don_c = smem + KIMG_BYTES;         // dO natural cur
-  char* qn_n = smem + 2 * KIMG_BYTES;
-  char* don_n = smem + 3 * KIMG_BYTES;
+  char* don_c = smem + TFKV * 256;         // dO natural cur
+  char* qn_n = smem + 2 * (TFKV * 256);
+  char* don_n = smem + 3 * (TFKV * 256);
 
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
   const int G = Hq / Hkv;
@@ -824,8 +828,8 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   f32x16 acc[4] = {};
   const float sc2 = scale * LOG2E;
 
-  const int t0 = causal ? (kv0_blk / FKV) : 0;
-  const int nt = (S + FKV - 1) / FKV;
+  const int t0 = causal ? (kv0_blk / TFKV) : 0;
+  const int nt = (S + TFKV - 1) / TFKV;
   const int nt_eff = nt - t0;
   const int total = G * nt_eff;
 
@@ -837,21 +841,21 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   };
   auto stage_lse = [&](int gh, int t, int buf) {
     const int hq_ = hkv * G + gh;
-    const int q0_ = t * FKV;
-    if (threadIdx.x < FKV) {
+    const int q0_ = t * TFKV;
+    if (threadIdx.x < TFKV) {
       const float* lse_b = lse + ((long)b * Hq + hq_) * S;
       const int qg = q0_ + threadIdx.x;
       lse_buf[buf][threadIdx.x] = (qg < S) ? lse_b[qg] * LOG2E : 0.f;
-    } else if (threadIdx.x < 2 * FKV) {
+    } else if (threadIdx.x < 2 * TFKV) {
       const float* del_b = delta + ((long)b * Hq + hq_) * S;
-      const int qg = q0_ + (threadIdx.x - FKV);
-      del_buf[buf][threadIdx.x - FKV] = (qg < S) ? del_b[qg] : 0.f;
+      const int qg = q0_ + (threadIdx.x - TFKV);
+      del_buf[buf][threadIdx.x - TFKV] = (qg < S) ? del_b[qg] : 0.f;
     }
   };
 
   // prologue: tile (gh=0, t=t0) — glds the natural images
-  stage_k_glds8(head_q(0), (long)t0 * FKV, q_seq_stride, S, qn_c);
-  stage_k_glds8(head_do(0), (long)t0 * FKV, dout_rs, S, don_c);
+  stage_k_glds8<TFKV>(head_q(0), (long)t0 * TFKV, q_seq_stride, S, qn_c);
+  stage_k_glds8<TFKV>(head_do(0), (long)t0 * TFKV, dout_rs, S, don_c);
   asm volatile("s_waitcnt vmcnt(0)");
   stage_lse(0, t0, 0);
   __syncthreads();
@@ -862,26 +866,27 @@ attn_bwd_dkv_kernel(const unsigned short* __restrict__ q,
   for (int idx = 0; idx < total; ++idx) {
     const int gh = idx / nt_eff;
     const int t = t0 + idx % nt_eff;
-    const int q0 = t * FKV;
+    const int q0 = t * TFKV;
     const int hq = hkv * G + gh;
     const bool has_next = (idx + 1) < total;
     const int ngh = (idx + 1) / nt_eff;
-    const long nrow0 = (long)(t0 + (idx + 1) % nt_eff) * FKV;
+    const long nrow0 = (long)(t0 + (idx + 1) % nt_eff) * TFKV;
     if (has_next) {
       // async glds issue BEFORE compute (their latency hides under the
       // MFMAs; the address temporaries are consumed by the instruction
       // immediately so this does not raise register pressure).  The
       // reg-staged tr loads stay in the tail: their 32 data registers
       // across the compute phase are what spilled.
-      stage_k_glds8(head_q(ngh), nrow0, q_seq_stride, S, qn_n);
-      stage_k_glds8(head_do(ngh), nrow0, dout_rs, S, don_n);
+      stage_k_glds8<TFKV>(head_q(ngh), nrow0, q_seq_stride, S, qn_n);
+      stage_k_glds8<TFKV>(head_do(ngh), nrow0, dout_rs, S, don_n);
     }
 
     const bool needed =
-        wave_active && (!causal || q0 + FKV - 1 >= kw0);
+        wave_active && (!causal || q0 + TFKV - 1 >= kw0);
     if (needed) {
       #pragma clang loop unroll(disable)
-      for (int sb = 0; sb < 2; ++sb) {
+      for (int sb = 0; sb < TFKV / 32; ++sb) {
+        if (causal && q0 + 32 * sb + 32 <= kw0) continue;  // fully masked
         // S[q, k own] (lane owns the k column; q rows via c_row so the
         // contraction dim of the accumulate MFMAs is the C-row dim);
         // dP[q, k own] for the dK waves
@@ -1027,11 +1032,26 @@ extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
                      scale, causal, q_rs, kv_rs, dqkv_q_rs);
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
   const long dout_rs = (long)Hq * HD;
-  hipLaunchKernelGGL(attn_bwd_dkv_kernel, dim3(B * Hkv * nkt),
-                     dim3(512), 0, stream, (const unsigned short*)q,
-                     (const unsigned short*)k, (const unsigned short*)v,
-                     (const unsigned short*)dout, (const float*)lse,
-                     (const float*)delta, (unsigned short*)dk,
-                     (unsigned short*)dv, B, S, Hq, Hkv,
-                     scale, causal, q_rs, kv_rs, dout_rs, dqkv_kv_rs);
+  static int dkv_fkv = -1;
+  if (dkv_fkv < 0) {
+    const char* e = getenv("TORCHX_AMD_DKV_FKV");
+    dkv_fkv = (e && atoi(e) == 128) ? 128 : 64;
+  }
+  if (dkv_fkv == 128 && S % 128 == 0) {
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<128>), dim3(B * Hkv * nkt),
+                       dim3(512), 0, stream, (const unsigned short*)q,
+                       (const unsigned short*)k, (const unsigned short*)v,
+                       (const unsigned short*)dout, (const float*)lse,
+                       (const float*)delta, (unsigned short*)dk,
+                       (unsigned short*)dv, B, S, Hq, Hkv,
+                       scale, causal, q_rs, kv_rs, dout_rs, dqkv_kv_rs);
+  } else {
+    hipLaunchKernelGGL((attn_bwd_dkv_kernel<64>), dim3(B * Hkv * nkt),
+                       dim3(512), 0, stream, (const unsigned short*)q,
+                       (const unsigned short*)k, (const unsigned short*)v,
+                       (const unsigned short*)dout, (const float*)lse,
+                       (const float*)delta, (unsigned short*)dk,
+                       (unsigned short*)dv, B, S, Hq, Hkv,
+                       scale, causal, q_rs, kv_rs, dout_rs, dqkv_kv_rs);
+  }
 }
