@@ -1032,10 +1032,12 @@ extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
                      scale, causal, q_rs, kv_rs, dqkv_q_rs);
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
   const long dout_rs = (long)Hq * HD;
+  // 128-row q tiles measured +4% over 64 (fewer barriers/iterations beat
+  // the 2-blocks/CU the 64-row variant's smaller LDS allows)
   static int dkv_fkv = -1;
   if (dkv_fkv < 0) {
     const char* e = getenv("TORCHX_AMD_DKV_FKV");
-    dkv_fkv = (e && atoi(e) == 128) ? 128 : 64;
+    dkv_fkv = (e && atoi(e) == 64) ? 64 : 128;
   }
   if (dkv_fkv == 128 && S % 128 == 0) {
     hipLaunchKernelGGL((attn_bwd_dkv_kernel<128>), dim3(B * Hkv * nkt),
