@@ -171,15 +171,15 @@ __device__ __forceinline__ mbf16x8 nat_tr_frag(const char* nat, int ks,
 // (16/NW per wave), source-permuted so the lane-linear LDS image lands
 // swizzled (T2 note: swizzle moves to the SOURCE with glds staging).
 // Rows beyond S clamp to S-1 (values masked in softmax).
-template <int NW = 4>
+template <int NW = 4, int TFKV = FKV>
 __device__ __forceinline__ void stage_k_glds(
     const unsigned short* __restrict__ kb, long kv0, long stride_elems,
     int S, char* kimg) {
   const int wave = threadIdx.x / 64;
   const int lane = threadIdx.x & 63;
   #pragma unroll
-  for (int j = 0; j < 16 / NW; ++j) {
-    const int i = wave * (16 / NW) + j;
+  for (int j = 0; j < (TFKV / 4) / NW; ++j) {
+    const int i = wave * ((TFKV / 4) / NW) + j;
     const int row = i * 4 + (lane >> 4);
     const int colbyte = ((lane & 15) * 16) ^ ((row & 15) << 4);
     long srow = kv0 + row;
@@ -523,6 +523,7 @@ attn_bwd_pre_kernel(const unsigned short* __restrict__ dout,
 // Backward dQ: grid over q tiles; inner loop over kv tiles.
 // dQ[q,d] = scale * sum_k (P*(dP - delta))[q,k] * K[k,d]
 // ---------------------------------------------------------------------------
+template <int TFKV>
 __global__ void __launch_bounds__(256, 2)
 attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
                    const unsigned short* __restrict__ k,
@@ -538,11 +539,11 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   // (K^T) is read straight off the natural K image with hardware
   // transpose reads — no K^T build, no rot image.  64 KiB LDS -> true
   // 2 blocks/CU.
-  __shared__ __align__(16) char smem[4 * KIMG_BYTES];
+  __shared__ __align__(16) char smem[4 * (TFKV * 256)];
   char* kcur = smem;
-  char* knxt = smem + KIMG_BYTES;
-  char* vcur = smem + 2 * KIMG_BYTES;
-  char* vnxt = smem + 3 * KIMG_BYTES;
+  char* knxt = smem + TFKV * 256;
+  char* vcur = smem + 2 * (TFKV * 256);
+  char* vnxt = smem + 3 * (TFKV * 256);
 
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
   const int G = Hq / Hkv;
@@ -590,22 +591,22 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
   const float sc2 = scale * LOG2E;
 
   const int kv_limit = causal ? min(S, q0_blk + BLOCK_Q) : S;
-  const int ntiles = (kv_limit + FKV - 1) / FKV;
+  const int ntiles = (kv_limit + TFKV - 1) / TFKV;
   const int qw_max = min(qw0 + QBLK - 1, S - 1);
 
-  stage_k_glds(kb, 0, kv_seq_stride, S, kcur);
-  stage_k_glds(vb, 0, kv_seq_stride, S, vcur);
+  stage_k_glds<4, TFKV>(kb, 0, kv_seq_stride, S, kcur);
+  stage_k_glds<4, TFKV>(vb, 0, kv_seq_stride, S, vcur);
   asm volatile("s_waitcnt vmcnt(0)");
   __syncthreads();
 
   for (int t = 0; t < ntiles; ++t) {
-    const int kv0 = t * FKV;
+    const int kv0 = t * TFKV;
     const bool has_next = (t + 1) < ntiles;
     if (has_next) {
       // next tile's glds issue BEFORE compute; with the XCD swizzle the
       // K/V stream is L2-resident, one tile of cover is plenty
-      stage_k_glds(kb, kv0 + FKV, kv_seq_stride, S, knxt);
-      stage_k_glds(vb, kv0 + FKV, kv_seq_stride, S, vnxt);
+      stage_k_glds<4, TFKV>(kb, kv0 + TFKV, kv_seq_stride, S, knxt);
+      stage_k_glds<4, TFKV>(vb, kv0 + TFKV, kv_seq_stride, S, vnxt);
     }
 
     const bool needed = wave_active && (!causal || kv0 <= qw_max);
@@ -622,7 +623,8 @@ attn_bwd_dq_kernel(const unsigned short* __restrict__ q,
       // couple them -> half the live accumulators of a fused pass; the
       // loop must NOT unroll or both subs' accumulators go live at once)
       #pragma clang loop unroll(disable)
-      for (int sb = 0; sb < 2; ++sb) {
+      for (int sb = 0; sb < TFKV / 32; ++sb) {
+        if (causal && kv0 + 32 * sb > qw_max) break;  // rest fully masked
         f32x16 acc_s = {}, acc_dp = {};
         #pragma unroll
         for (int c = 0; c < 8; ++c) {
@@ -1024,12 +1026,26 @@ extern "C" void attn_bwd_launch(const void* q, const void* k, const void* v,
                      dim3(256), 0, stream, (const unsigned short*)dout,
                      (const unsigned short*)o, (float*)delta, rows, S, Hq);
   const int nqt = (S + BLOCK_Q - 1) / BLOCK_Q;
-  hipLaunchKernelGGL(attn_bwd_dq_kernel, dim3(B * Hq * nqt), dim3(256), 0,
-                     stream, (const unsigned short*)q,
-                     (const unsigned short*)k, (const unsigned short*)v,
-                     (const unsigned short*)dout, (const float*)lse,
-                     (const float*)delta, (unsigned short*)dq, B, S, Hq, Hkv,
-                     scale, causal, q_rs, kv_rs, dqkv_q_rs);
+  static int dq_fkv = -1;
+  if (dq_fkv < 0) {
+    const char* e = getenv("TORCHX_AMD_DQ_FKV");
+    dq_fkv = (e && atoi(e) == 128) ? 128 : 64;
+  }
+  if (dq_fkv == 128 && S % 128 == 0) {
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<128>), dim3(B * Hq * nqt),
+                       dim3(256), 0, stream, (const unsigned short*)q,
+                       (const unsigned short*)k, (const unsigned short*)v,
+                       (const unsigned short*)dout, (const float*)lse,
+                       (const float*)delta, (unsigned short*)dq, B, S, Hq,
+                       Hkv, scale, causal, q_rs, kv_rs, dqkv_q_rs);
+  } else {
+    hipLaunchKernelGGL((attn_bwd_dq_kernel<64>), dim3(B * Hq * nqt),
+                       dim3(256), 0, stream, (const unsigned short*)q,
+                       (const unsigned short*)k, (const unsigned short*)v,
+                       (const unsigned short*)dout, (const float*)lse,
+                       (const float*)delta, (unsigned short*)dq, B, S, Hq,
+                       Hkv, scale, causal, q_rs, kv_rs, dqkv_q_rs);
+  }
   const int nkt = (S + BLOCK_K - 1) / BLOCK_K;
   const long dout_rs = (long)Hq * HD;
   // 128-row q tiles measured +4% over 64 (fewer barriers/iterations beat
