@@ -237,6 +237,8 @@ def main() -> int:
     ms_per_step = elapsed / args.steps * 1000.0
     tokens_per_step = B * S * world
     tok_s = tokens_per_step / (elapsed / args.steps)
+    peak_hbm_gb = (round(torch.cuda.max_memory_allocated() / 1e9, 2)
+                   if use_gpu else None)
 
     # launch->first-step latency (the other half of the BASELINE metric):
     # the full dist.ddp launcher path — materialize -> Popen -> agent ->
@@ -283,6 +285,8 @@ def main() -> int:
                 "final_loss": loss,
             },
         }
+        if peak_hbm_gb is not None:
+            result["peak_hbm_gb"] = peak_hbm_gb
         if launch_latency is not None:
             result["launch_to_first_step_s"] = launch_latency["value"]
             result["launch_submit_s"] = launch_latency["submit_seconds"]
