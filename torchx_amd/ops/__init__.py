@@ -257,7 +257,18 @@ def fused_linear_cross_entropy(
     x2 = x.reshape(-1, H)
     t = targets.reshape(-1)
     if _on_gpu(x):
-        return _FusedLinearCE.apply(x2.contiguous(), w, t, chunk)
+        if torch.is_grad_enabled() and (x.requires_grad or w.requires_grad):
+            return _FusedLinearCE.apply(x2.contiguous(), w, t, chunk)
+        # eval path: loss only, no dx/dW work
+        hip = hip_ops()
+        x2 = x2.contiguous()
+        T = x2.shape[0]
+        loss_sum = torch.zeros((), device=x.device, dtype=torch.float32)
+        for s in range(0, T, chunk):
+            e = min(T, s + chunk)
+            loss_c, _ = hip.ce_fwd(x2[s:e] @ w.t(), t[s:e].contiguous())
+            loss_sum += loss_c.sum()
+        return loss_sum / T
     return reference.cross_entropy(x2 @ w.t(), t)
 
 
