@@ -131,3 +131,25 @@ class TestIds:
         # degenerate names still produce a valid id
         assert make_unique("///").startswith("app-")
         assert len(random_id(4)) == 4
+
+
+def test_settings_constants_wire_stable():
+    from torchx_amd import settings as s
+
+    assert s.ENV_TORCHX_JOB_ID == "TORCHX_JOB_ID"
+    assert s.ENV_TORCHX_TRACKERS == "TORCHX_TRACKERS"
+    assert s.ENV_TORCHXCONFIG == "TORCHXCONFIG"
+    assert s.ENV_TORCHX_RANK0_HOST == "TORCHX_RANK0_HOST"
+    assert s.ENV_HIP_VISIBLE_DEVICES == "HIP_VISIBLE_DEVICES"
+
+
+def test_notebook_workspace_roundtrip():
+    from torchx_amd.notebook import get_workspace, write_workspace_file
+
+    path = write_workspace_file("pkg/mod.py", "VALUE = 41\n")
+    assert path.startswith(get_workspace())
+    import fsspec
+
+    fs, p = fsspec.core.url_to_fs(path)
+    with fs.open(p) as f:
+        assert b"VALUE = 41" in f.read()
