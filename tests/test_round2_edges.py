@@ -150,3 +150,27 @@ class TestLogIterSinceUntil:
         assert any("ignored" in r.message for r in caplog.records)
         assert any("one" in ln for ln in lines)
         s.close()
+
+
+class TestWorkerNumaPrefix:
+    def test_maps_local_rank_through_visible_devices(self, monkeypatch):
+        from torchx_amd.agent.__main__ import worker_numa_prefix
+        from torchx_amd.schedulers import devices as dev_mod
+
+        calls = []
+
+        def fake_bind(dev):
+            calls.append(dev)
+            return ["/usr/bin/numactl", "--cpunodebind=1", "--membind=1"]
+
+        monkeypatch.setattr(dev_mod, "numa_bind_args", fake_bind)
+        # replica owns global devices 4..7; worker 2 -> global 6
+        pref = worker_numa_prefix(2, {"HIP_VISIBLE_DEVICES": "4,5,6,7"})
+        assert calls == ["6"]
+        assert pref[0].endswith("numactl")
+        # no visible restriction: worker lr maps to global lr
+        calls.clear()
+        worker_numa_prefix(3, {})
+        assert calls == ["3"]
+        # out-of-range worker -> no binding
+        assert worker_numa_prefix(9, {"HIP_VISIBLE_DEVICES": "0"}) == []

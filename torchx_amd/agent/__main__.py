@@ -100,6 +100,27 @@ def _default_nproc() -> int:
     return n if n > 0 else 1
 
 
+def worker_numa_prefix(local_rank: int,
+                       env: Optional[Dict[str, str]] = None) -> List[str]:
+    """numactl argv prefix binding worker ``local_rank`` to the NUMA node
+    of ITS GPU (SURVEY §7 step 2 at worker granularity — the replica-level
+    binding no-ops when one agent owns GPUs on several nodes).  Resolves
+    the worker's GLOBAL device index through HIP_VISIBLE_DEVICES."""
+    from torchx_amd.schedulers.devices import numa_bind_args
+
+    env = env if env is not None else os.environ
+    visible = env.get("HIP_VISIBLE_DEVICES") or env.get(
+        "ROCR_VISIBLE_DEVICES")
+    if visible:
+        devs = [d.strip() for d in visible.split(",") if d.strip()]
+        if local_rank >= len(devs):
+            return []
+        dev = devs[local_rank]
+    else:
+        dev = str(local_rank)
+    return numa_bind_args(dev)
+
+
 def spawn_workers(args, rdzv: RendezvousResult, restart_count: int,
                   log_dir: str) -> List[Worker]:
     nproc = args.nproc_per_node
@@ -139,8 +160,9 @@ def spawn_workers(args, rdzv: RendezvousResult, restart_count: int,
         }
         stdout = subprocess.PIPE if args.tee else None
         stderr = subprocess.PIPE if args.tee else None
+        numa_prefix = worker_numa_prefix(lr) if args.numa_affinity else []
         proc = subprocess.Popen(
-            prof_prefix + cmd_base + args.script_args,
+            numa_prefix + prof_prefix + cmd_base + args.script_args,
             env=env,
             stdout=stdout,
             stderr=stderr,
@@ -227,6 +249,9 @@ def main(argv: Optional[List[str]] = None) -> int:
     p.add_argument("--tee", action="store_true", default=True)
     p.add_argument("--no-tee", dest="tee", action="store_false")
     p.add_argument("--no-python", action="store_true")
+    p.add_argument("--numa-affinity", action="store_true", default=True)
+    p.add_argument("--no-numa-affinity", dest="numa_affinity",
+                   action="store_false")
     p.add_argument("--rocprof", action="store_true",
                    help="wrap each worker in rocprofv3 --kernel-trace "
                         "--stats (output under the log dir)")
