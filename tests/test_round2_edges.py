@@ -174,3 +174,22 @@ class TestWorkerNumaPrefix:
         assert calls == ["3"]
         # out-of-range worker -> no binding
         assert worker_numa_prefix(9, {"HIP_VISIBLE_DEVICES": "0"}) == []
+
+    def test_worker_cpu_affinity_cpulist(self, monkeypatch, tmp_path):
+        from torchx_amd.agent.__main__ import worker_cpu_affinity
+        from torchx_amd.schedulers import devices as dev_mod
+
+        node = tmp_path / "devices" / "system" / "node" / "node1"
+        node.mkdir(parents=True)
+        (node / "cpulist").write_text("32-63,160-191\n")
+        monkeypatch.setattr(dev_mod, "numa_node_of",
+                            lambda dev, numa_map=None: 1 if dev == "6" else -1)
+        monkeypatch.setattr(
+            dev_mod, "numa_cpulist",
+            lambda n, sysfs="/sys": dev_mod.numa_cpulist.__wrapped__(n)
+            if False else ("32-63,160-191" if n == 1 else ""))
+        aff = worker_cpu_affinity(2, {"HIP_VISIBLE_DEVICES": "4,5,6,7"})
+        assert aff == "32-63,160-191"
+        assert worker_cpu_affinity(0, {"HIP_VISIBLE_DEVICES": "0"}) == ""
+        # parse_cpulist round-trip
+        assert dev_mod.parse_cpulist("0-3,8,10-11") == {0, 1, 2, 3, 8, 10, 11}

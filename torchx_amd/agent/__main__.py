@@ -121,6 +121,27 @@ def worker_numa_prefix(local_rank: int,
     return numa_bind_args(dev)
 
 
+def worker_cpu_affinity(local_rank: int,
+                        env: Optional[Dict[str, str]] = None) -> str:
+    """cpulist of the worker GPU's NUMA node ("" if unknown): the
+    numactl-free affinity path, exported as TORCHX_AMD_CPU_AFFINITY and
+    applied by torchx_amd.distributed.init_pg via sched_setaffinity
+    (numactl is not installed on the target fleet)."""
+    from torchx_amd.schedulers.devices import numa_cpulist, numa_node_of
+
+    env = env if env is not None else os.environ
+    visible = env.get("HIP_VISIBLE_DEVICES") or env.get(
+        "ROCR_VISIBLE_DEVICES")
+    if visible:
+        devs = [d.strip() for d in visible.split(",") if d.strip()]
+        if local_rank >= len(devs):
+            return ""
+        dev = devs[local_rank]
+    else:
+        dev = str(local_rank)
+    return numa_cpulist(numa_node_of(dev))
+
+
 def spawn_workers(args, rdzv: RendezvousResult, restart_count: int,
                   log_dir: str) -> List[Worker]:
     nproc = args.nproc_per_node
@@ -158,6 +179,10 @@ def spawn_workers(args, rdzv: RendezvousResult, restart_count: int,
             "TORCHELASTIC_ERROR_FILE": error_file,
             "OMP_NUM_THREADS": os.environ.get("OMP_NUM_THREADS", "1"),
         }
+        if args.numa_affinity:
+            aff = worker_cpu_affinity(lr)
+            if aff:
+                env["TORCHX_AMD_CPU_AFFINITY"] = aff
         stdout = subprocess.PIPE if args.tee else None
         stderr = subprocess.PIPE if args.tee else None
         numa_prefix = worker_numa_prefix(lr) if args.numa_affinity else []

@@ -56,8 +56,25 @@ def _auto_backend() -> str:
     return "gloo"
 
 
+def _apply_cpu_affinity() -> None:
+    """Pin this worker to its GPU's NUMA-node CPUs if the agent exported
+    TORCHX_AMD_CPU_AFFINITY (the numactl-free affinity path)."""
+    spec = os.environ.get("TORCHX_AMD_CPU_AFFINITY")
+    if not spec:
+        return
+    try:
+        from torchx_amd.schedulers.devices import parse_cpulist
+
+        cpus = parse_cpulist(spec)
+        if cpus:
+            os.sched_setaffinity(0, cpus)
+    except (OSError, ValueError):  # affinity is best-effort
+        pass
+
+
 def init_pg(backend: str = "auto") -> torch.device:
     """Initialize the default process group and return this rank's device."""
+    _apply_cpu_affinity()
     if backend == "auto":
         backend = _auto_backend()
     if not dist.is_initialized():

@@ -170,6 +170,49 @@ def device_numa_map() -> Dict[int, int]:
     return _read_device_numa_map()
 
 
+def numa_node_of(devices: Optional[str],
+                 numa_map: Optional[Dict[int, int]] = None) -> int:
+    """The single NUMA node a device string ("0" / "0,1") sits on, or -1
+    (unknown / spans nodes / no assignment)."""
+    if not devices:
+        return -1
+    if numa_map is None:
+        numa_map = device_numa_map()
+    try:
+        nodes = {numa_map.get(int(d), -1) for d in devices.split(",")}
+    except ValueError:
+        return -1
+    if len(nodes) != 1:
+        return -1
+    return nodes.pop()
+
+
+def numa_cpulist(node: int, sysfs: str = "/sys") -> str:
+    """The kernel's cpulist string for a NUMA node ("0-31,64-95"), or ""."""
+    if node < 0:
+        return ""
+    try:
+        with open(f"{sysfs}/devices/system/node/node{node}/cpulist") as f:
+            return f.read().strip()
+    except OSError:
+        return ""
+
+
+def parse_cpulist(spec: str) -> set:
+    """"0-3,8,10-11" -> {0,1,2,3,8,10,11}."""
+    cpus: set = set()
+    for part in spec.split(","):
+        part = part.strip()
+        if not part:
+            continue
+        lo, _, hi = part.partition("-")
+        if hi:
+            cpus.update(range(int(lo), int(hi) + 1))
+        else:
+            cpus.add(int(lo))
+    return cpus
+
+
 def numa_bind_args(
     devices: Optional[str],
     numa_map: Optional[Dict[int, int]] = None,
