@@ -62,6 +62,18 @@ def main() -> int:
         dev_idx = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(dev_idx)
         device = torch.device("cuda", dev_idx)
+        # bind this rank's host threads to its GPU's NUMA node (the same
+        # affinity the agent gives launched workers; best-effort)
+        try:
+            from torchx_amd.schedulers.devices import (
+                numa_cpulist, numa_node_of, parse_cpulist,
+            )
+
+            cpus = parse_cpulist(numa_cpulist(numa_node_of(str(dev_idx))))
+            if cpus:
+                os.sched_setaffinity(0, cpus)
+        except Exception:  # noqa: BLE001
+            pass
         backend = "nccl"  # RCCL on ROCm
         # tuned hipBLASLt algorithm table (tools/gemm_tune.py), read-only
         tuned = os.path.join(os.path.dirname(os.path.abspath(__file__)),
