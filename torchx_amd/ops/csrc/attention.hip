@@ -994,17 +994,17 @@ extern "C" void attn_fwd_launch(const void* q, const void* k, const void* v,
                                 void* o, void* lse, int B, int S, int Hq,
                                 int Hkv, float scale, int causal,
                                 long q_rs, long kv_rs, hipStream_t stream) {
-  // S-dependent block size: 8-wave blocks amortize staging better once
-  // the causal wave-skew is small relative to the tile count (measured:
-  // +14% at S=8192, -3% at S=4096 pre-T10; TORCHX_AMD_FWD_8WAVE=1/0
-  // overrides for A/B)
+  // Post-T10 the 4-wave block wins at EVERY length (S=4096: 618 vs 558
+  // TF/s; S=8192: 706 vs 648 — the pre-T10 dispatch sent S>=8192 to the
+  // 8-wave variant, which lost its edge once the V^T build disappeared
+  // and the 4-wave kernel fit 2 blocks/CU).  TORCHX_AMD_FWD_8WAVE=1
+  // forces the 8-wave path for re-measurement.
   static int fwd8 = -1;
   if (fwd8 < 0) {
     const char* e = getenv("TORCHX_AMD_FWD_8WAVE");
-    fwd8 = e ? (atoi(e) ? 1 : 0) : -2;  // -2 = auto
+    fwd8 = (e && atoi(e)) ? 1 : 0;
   }
-  const bool use8 = (fwd8 == 1) || (fwd8 == -2 && S >= 8192);
-  if (use8) {
+  if (fwd8 == 1) {
     const int nqt = (S + 8 * QBLK - 1) / (8 * QBLK);
     hipLaunchKernelGGL((attn_fwd_kernel<8>), dim3(B * Hq * nqt), dim3(512),
                        0, stream, (const unsigned short*)q,
