@@ -28,7 +28,7 @@ from typing import List, Optional
 import torch
 import torch.distributed as dist
 
-from torchx_amd.distributed import init_pg, rank
+from torchx_amd.distributed import init_pg, on_rank0_first, rank
 from torchx_amd.examples.apps.classifier.data import (
     ImageDataModule,
     SyntheticImages,
@@ -98,10 +98,15 @@ def main(argv: Optional[List[str]] = None) -> int:
             ds, num_classes = load_data(args.dataset_path)
         else:
             num_classes = 8
-            path = create_random_data(
-                os.path.join(out_dir, "data"), n=args.num_samples,
-                num_classes=num_classes,
-            )
+            path = os.path.join(out_dir, "data", "data.pt")
+            # rank 0 materializes the dataset; everyone else waits then
+            # loads (concurrent writers raced on the same file)
+            with on_rank0_first():
+                if rank() == 0:
+                    path = create_random_data(
+                        os.path.join(out_dir, "data"), n=args.num_samples,
+                        num_classes=num_classes,
+                    )
             ds, _ = load_data(path)
         dm = ImageDataModule(ds, batch_size=args.batch_size)
 
