@@ -14,13 +14,11 @@ BSHD directly).
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import Optional
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from torchx_amd import ops
 

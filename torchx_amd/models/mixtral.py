@@ -12,14 +12,12 @@ all-reduce must therefore skip expert parameters when EP is active
 
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import List, Optional
 
 import torch
 import torch.distributed as dist
 import torch.nn as nn
-import torch.nn.functional as F
 
 from torchx_amd import ops
 from torchx_amd.parallel.ep import exchange_counts, expert_all_to_all

@@ -16,7 +16,7 @@ in CPU CI (gloo lacks all_to_all); the RCCL path is the production one.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist

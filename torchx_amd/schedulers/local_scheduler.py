@@ -22,7 +22,6 @@ import shutil
 import signal
 import subprocess
 import tempfile
-import threading
 import time
 from dataclasses import dataclass, field
 from datetime import datetime
