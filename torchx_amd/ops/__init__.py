@@ -466,3 +466,23 @@ def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
         return _SwiGLUPacked.apply(gu)
     g, u = gu.chunk(2, dim=-1)
     return reference.swiglu(g, u)
+
+
+# ---------------------------------------------------------------------------
+# Decode attention (serving): single-position attention over a KV cache —
+# ops/csrc/decode_attn.hip, flash-decode style (wave-local online softmax,
+# one merge barrier). No training/backward path: inference only.
+# ---------------------------------------------------------------------------
+
+
+def decode_attention(q: torch.Tensor, kcache: torch.Tensor,
+                     vcache: torch.Tensor, length: int,
+                     scale: Optional[float] = None) -> torch.Tensor:
+    """q [B, Hq, 128] bf16; k/v caches [B, T, Hkv, 128] bf16 with the
+    first ``length`` positions valid -> o [B, Hq, 128]."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        return hip_ops().decode_attn(q.contiguous(), kcache, vcache,
+                                     length, scale)
+    return reference.decode_attention(q, kcache, vcache, length, scale)

@@ -38,6 +38,8 @@ void mfma_probe_tr_launch(const void*, const void*, void*, void*,
 void fp8_cast_transpose_launch(const void*, void*, void*, void*,
                                const void*, long, long, int, hipStream_t);
 void transpose_bf16_launch(const void*, void*, long, long, hipStream_t);
+void decode_attn_launch(const void*, const void*, const void*, void*, int,
+                        int, int, int, int, float, hipStream_t);
 }
 
 namespace {
@@ -322,6 +324,26 @@ at::Tensor transpose_bf16(at::Tensor x) {
   return out;
 }
 
+// ---- decode attention -----------------------------------------------------
+// q [B, Hq, 128] bf16; k/v caches [B, T, Hkv, 128] bf16; first L rows
+// valid. Returns o [B, Hq, 128].
+at::Tensor decode_attn(at::Tensor q, at::Tensor kc, at::Tensor vc, long L,
+                       double scale) {
+  check_bf16(q, "q");
+  check_bf16(kc, "kc");
+  check_bf16(vc, "vc");
+  const int B = q.size(0), Hq = q.size(1);
+  const int T = kc.size(1), Hkv = kc.size(2);
+  TORCH_CHECK(q.size(2) == 128 && kc.size(3) == 128, "head_dim must be 128");
+  TORCH_CHECK(L >= 1 && L <= T, "invalid cache length");
+  TORCH_CHECK(Hq % Hkv == 0, "GQA group mismatch");
+  auto o = at::empty_like(q);
+  decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                     o.data_ptr(), B, Hq, Hkv, T, (int)L, (float)scale,
+                     cur_stream());
+  return o;
+}
+
 // ---- probe ----------------------------------------------------------------
 std::vector<at::Tensor> mfma_probe_tr(at::Tensor a, at::Tensor b) {
   check_bf16(a, "a");
@@ -365,4 +387,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe", &mfma_probe);
   m.def("mfma_probe_tr", &mfma_probe_tr);
   m.def("transpose_bf16", &transpose_bf16);
+  m.def("decode_attn", &decode_attn);
 }

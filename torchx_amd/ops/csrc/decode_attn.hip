@@ -1,0 +1,120 @@
+// Single-position (decode) attention over a KV cache, bf16, GQA,
+// HD = 128 — the serving-side counterpart of the training flash kernels.
+//
+//   o[b,hq,:] = softmax(scale * q[b,hq,:] . K[b,:L,hkv,:]^T) . V[b,:L,hkv,:]
+//
+// One 256-thread block per (b, hq). Work split: 16 lanes per cached row
+// (16 x 8 d = 128, ushort8 loads), so a wave covers 4 rows per iteration
+// and the block 16; each wave keeps its OWN online-softmax state (m, l,
+// 8-float o partial per lane) over its rows — flash-decode style — and
+// the four waves merge through LDS once at the end (one barrier).
+// HBM-bound by the K/V cache stream (L x 512 B per (b, hkv), shared by
+// the GQA group via L2).
+#include "common.h"
+
+__global__ void __launch_bounds__(256)
+decode_attn_kernel(const unsigned short* __restrict__ q,   // [B, Hq, 128]
+                   const unsigned short* __restrict__ kc,  // [B, T, Hkv, 128]
+                   const unsigned short* __restrict__ vc,  // [B, T, Hkv, 128]
+                   unsigned short* __restrict__ o,         // [B, Hq, 128]
+                   int B, int Hq, int Hkv, int T, int L, float scale) {
+  const int b = blockIdx.x / Hq;
+  const int hq = blockIdx.x % Hq;
+  const int hkv = hq / (Hq / Hkv);
+
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const int tsub = lane >> 4;        // wave's row slot 0..3
+  const int dchunk = (lane & 15) * 8;  // this lane's 8 d's
+
+  // q fragment for this lane's d-chunk (f32)
+  const unsigned short* qp = q + ((long)b * Hq + hq) * HD + dchunk;
+  float qv[8];
+  {
+    ushort8 v = *(const ushort8*)qp;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) qv[j] = bf16_to_f32(v[j]) * scale;
+  }
+
+  const long row_stride = (long)Hkv * HD;
+  const unsigned short* kb = kc + (long)b * T * row_stride + (long)hkv * HD;
+  const unsigned short* vb = vc + (long)b * T * row_stride + (long)hkv * HD;
+
+  float m_run = -3.0e38f, l_run = 0.f;
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+
+  // rows: wave w owns t = 16*i + 4*w + tsub
+  for (int t = 4 * wave + tsub; t < L; t += 16) {
+    const unsigned short* kr = kb + (long)t * row_stride + dchunk;
+    ushort8 kv8 = *(const ushort8*)kr;
+    float s = 0.f;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) s = fmaf(qv[j], bf16_to_f32(kv8[j]), s);
+    // reduce the 16-lane group -> every lane of the group has the score
+    #pragma unroll
+    for (int off = 8; off >= 1; off >>= 1) {
+      s += __shfl_xor(s, off, 16);
+    }
+    // online softmax (wave-local state; all 16 lanes of the group agree)
+    const float m_new = fmaxf(m_run, s);
+    const float alpha = __builtin_amdgcn_exp2f(
+        1.44269504f * (m_run - m_new));
+    const float p = __builtin_amdgcn_exp2f(1.44269504f * (s - m_new));
+    l_run = l_run * alpha + p;
+    const unsigned short* vr = vb + (long)t * row_stride + dchunk;
+    ushort8 vv8 = *(const ushort8*)vr;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      acc[j] = fmaf(acc[j], alpha, p * bf16_to_f32(vv8[j]));
+    }
+    m_run = m_new;
+  }
+
+  // combine the 4 row-slots of each wave, then the 4 waves: 16 partial
+  // (m, l, o[128]) states -> LDS, block-combined by wave 0
+  __shared__ float sm[16], sl[16], so[16][128];
+  const int slot = wave * 4 + tsub;
+  if ((lane & 15) == 0) {
+    sm[slot] = m_run;
+    sl[slot] = l_run;
+  }
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) so[slot][dchunk + j] = acc[j];
+  __syncthreads();
+
+  if (wave == 0) {
+    // lane group 0..15 handles d-chunk as before; combine 16 slots
+    float m_t = -3.0e38f;
+    #pragma unroll
+    for (int i = 0; i < 16; ++i) m_t = fmaxf(m_t, sm[i]);
+    float l_t = 0.f;
+    float out[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
+    for (int i = 0; i < 16; ++i) {
+      const float w = __builtin_amdgcn_exp2f(1.44269504f * (sm[i] - m_t));
+      l_t += sl[i] * w;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        out[j] = fmaf(so[i][dchunk + j], w, out[j]);
+      }
+    }
+    const float inv = (l_t > 0.f) ? 1.0f / l_t : 0.f;
+    if (tsub == 0) {  // 16 lanes cover all 128 d
+      unsigned short* op = o + ((long)b * Hq + hq) * HD + dchunk;
+      ushort8 ov;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) ov[j] = f32_to_bf16(out[j] * inv);
+      *(ushort8*)op = ov;
+    }
+  }
+}
+
+extern "C" void decode_attn_launch(const void* q, const void* kc,
+                                   const void* vc, void* o, int B, int Hq,
+                                   int Hkv, int T, int L, float scale,
+                                   hipStream_t stream) {
+  hipLaunchKernelGGL(decode_attn_kernel, dim3(B * Hq), dim3(256), 0, stream,
+                     (const unsigned short*)q, (const unsigned short*)kc,
+                     (const unsigned short*)vc, (unsigned short*)o, B, Hq,
+                     Hkv, T, L, scale);
+}

@@ -100,3 +100,20 @@ def adamw_step(
     denom = (v / bc2).sqrt().add_(eps)
     p32.add_(mhat / denom + weight_decay * p32, alpha=-lr)
     p16.copy_(p32.to(p16.dtype))
+
+
+def decode_attention(q: torch.Tensor, kcache: torch.Tensor,
+                     vcache: torch.Tensor, length: int,
+                     scale: float) -> torch.Tensor:
+    """fp32 reference for single-position attention over a KV cache.
+    q [B, Hq, D]; k/v caches [B, T, Hkv, D]; first ``length`` rows valid."""
+    B, Hq, D = q.shape
+    Hkv = kcache.shape[2]
+    G = Hq // Hkv
+    k = kcache[:, :length].float()                      # [B, L, Hkv, D]
+    v = vcache[:, :length].float()
+    k = k.repeat_interleave(G, dim=2)                   # [B, L, Hq, D]
+    v = v.repeat_interleave(G, dim=2)
+    s = torch.einsum("bhd,blhd->bhl", q.float(), k) * scale
+    p = torch.softmax(s, dim=-1)
+    return torch.einsum("bhl,blhd->bhd", p, v).to(q.dtype)
