@@ -24,6 +24,7 @@ SOURCES = [
     "adamw.hip",
     "cross_entropy.hip",
     "attention.hip",
+    "decode_attn.hip",
     "fp8_cast.hip",
     "probe.hip",
     "bindings.cpp",

@@ -12,6 +12,8 @@
 // the GQA group via L2).
 #include "common.h"
 
+#define HD 128
+
 __global__ void __launch_bounds__(256)
 decode_attn_kernel(const unsigned short* __restrict__ q,   // [B, Hq, 128]
                    const unsigned short* __restrict__ kc,  // [B, T, Hkv, 128]
