@@ -97,3 +97,19 @@ def test_generate_gpu_end_to_end():
     err = (dec.float() - full.float()).abs().max().item()
     scale = full.float().abs().max().item() + 1e-6
     assert err < 5e-2 * scale, (err, scale)
+
+
+@pytest.mark.gpu
+def test_generate_graphed_matches_eager():
+    from torchx_amd.models.generate import generate_graphed
+    from torchx_amd.models.llama import llama_gpu_tiny
+
+    dev = torch.device("cuda:0")
+    torch.manual_seed(4)
+    cfg = llama_gpu_tiny()
+    model = LlamaModel(cfg, device=dev)
+    tokens = torch.randint(0, cfg.vocab_size, (2, 24), device=dev)
+    eager = generate(model, tokens, max_new_tokens=10)
+    graphed = generate_graphed(model, tokens, max_new_tokens=10)
+    # greedy token streams must agree (identical kernels, identical math)
+    assert torch.equal(eager, graphed), (eager[:, 24:], graphed[:, 24:])

@@ -143,3 +143,102 @@ def generate(
         nxt = pick(decode_step(model, nxt, caches))
         out.append(nxt)
     return torch.cat(out, dim=1)
+
+
+class GraphedDecoder:
+    """hipGraph-captured decode loop: the whole per-token step (embed,
+    32 layers, lm_head, argmax, cache append, position bump) replays as
+    ONE graph with zero host work — the eager decode step is launch-bound
+    (~450 small kernels). The cache length and rope position are driven
+    by a device int32 scalar that the captured step increments itself,
+    so one capture serves every subsequent token.
+
+    Greedy-only (the argmax feeds back inside the graph).
+    """
+
+    def __init__(self, model: LlamaModel, caches: list, batch: int,
+                 first_token: torch.Tensor, start_pos: int):
+        self.model = model
+        self.caches = caches
+        cfg = model.cfg
+        dev = next(model.parameters()).device
+        self.tok = first_token.clone()                      # [B, 1] int64
+        self.pos32 = torch.tensor([start_pos], dtype=torch.int32,
+                                  device=dev)
+        self.pos64 = torch.tensor([start_pos], dtype=torch.int64,
+                                  device=dev)
+        torch.cuda.synchronize()
+        self.graph = torch.cuda.CUDAGraph()
+        # the warmup AND the capture each execute one REAL step — their
+        # produced tokens are collected in init_tokens
+        self.init_tokens = []
+        self._step_body()                      # eager warm (same code path)
+        torch.cuda.synchronize()
+        self.init_tokens.append(self.tok.clone())
+        with torch.cuda.graph(self.graph):
+            self._step_body()
+        torch.cuda.synchronize()
+        self.init_tokens.append(self.tok.clone())
+
+    @torch.no_grad()
+    def _step_body(self) -> None:
+        model, cfg = self.model, self.model.cfg
+        B = self.tok.shape[0]
+        cos = model.rope_cos.index_select(0, self.pos64)
+        sin = model.rope_sin.index_select(0, self.pos64)
+        x = model.embed(self.tok)
+        for blk, cache in zip(model.blocks, self.caches):
+            xn = ops.rmsnorm(x, blk.attn_norm, cfg.rms_eps)
+            q, k, v = _split_qkv(blk.wqkv(xn), cfg)
+            q = ops.rope(q, cos, sin)
+            k = ops.rope(k, cos, sin)
+            cache.k.index_copy_(1, self.pos64, k)
+            cache.v.index_copy_(1, self.pos64, v)
+            from torchx_amd.ops import hip_ops
+
+            o = hip_ops().decode_attn_dev(
+                q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous(),
+                cache.k, cache.v, self.pos32,
+                1.0 / cfg.head_dim ** 0.5)
+            x = x + blk.wo(o.reshape(B, 1, cfg.q_dim))
+            xn = ops.rmsnorm(x, blk.mlp_norm, cfg.rms_eps)
+            x = x + blk.wdown(ops.swiglu_packed(blk.wgu(xn)))
+        x = ops.rmsnorm(x, model.final_norm, cfg.rms_eps)
+        logits = model.lm_head(x[:, -1])
+        self.tok.copy_(logits.argmax(-1, keepdim=True))
+        self.pos32.add_(1)
+        self.pos64.add_(1)
+
+    def step(self) -> torch.Tensor:
+        """Replay one decode step; returns the new token [B, 1]."""
+        self.graph.replay()
+        return self.tok
+
+
+@torch.no_grad()
+def generate_graphed(model: LlamaModel, tokens: torch.Tensor,
+                     max_new_tokens: int,
+                     max_len: Optional[int] = None) -> torch.Tensor:
+    """Greedy generation with the hipGraph decode loop (GPU only)."""
+    cfg = model.cfg
+    B, S0 = tokens.shape
+    total = S0 + max_new_tokens
+    max_len = max_len or total
+    assert total <= cfg.max_seq_len and max_len >= total
+    dev = tokens.device
+    caches = [KVCache.empty(cfg, B, max_len, dev)
+              for _ in range(cfg.num_layers)]
+    out = [tokens]
+    nxt = prefill(model, tokens, caches).argmax(-1, keepdim=True)
+    out.append(nxt)
+    if max_new_tokens <= 2:
+        for _ in range(max_new_tokens - 1):
+            nxt = decode_step(model, nxt, caches).argmax(-1, keepdim=True)
+            out.append(nxt)
+        return torch.cat(out, dim=1)
+    dec = GraphedDecoder(model, caches, B, nxt, start_pos=S0)
+    out.extend(t for t in dec.init_tokens[:max_new_tokens - 1])
+    remaining = max_new_tokens - 1 - len(dec.init_tokens)
+    for _ in range(remaining):
+        out.append(dec.step().clone())
+    return torch.cat(out, dim=1)

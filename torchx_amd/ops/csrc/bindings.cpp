@@ -38,8 +38,9 @@ void mfma_probe_tr_launch(const void*, const void*, void*, void*,
 void fp8_cast_transpose_launch(const void*, void*, void*, void*,
                                const void*, long, long, int, hipStream_t);
 void transpose_bf16_launch(const void*, void*, long, long, hipStream_t);
-void decode_attn_launch(const void*, const void*, const void*, void*, int,
-                        int, int, int, int, float, hipStream_t);
+void decode_attn_launch(const void*, const void*, const void*, void*,
+                        const void*, int, int, int, int, int, float,
+                        hipStream_t);
 }
 
 namespace {
@@ -339,8 +340,25 @@ at::Tensor decode_attn(at::Tensor q, at::Tensor kc, at::Tensor vc, long L,
   TORCH_CHECK(Hq % Hkv == 0, "GQA group mismatch");
   auto o = at::empty_like(q);
   decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
-                     o.data_ptr(), B, Hq, Hkv, T, (int)L, (float)scale,
-                     cur_stream());
+                     o.data_ptr(), nullptr, B, Hq, Hkv, T, (int)L,
+                     (float)scale, cur_stream());
+  return o;
+}
+
+// graph-capturable variant: cache length = pos + 1 read on DEVICE from an
+// int32 scalar tensor, so a captured decode step stays valid as the cache
+// grows between replays.
+at::Tensor decode_attn_dev(at::Tensor q, at::Tensor kc, at::Tensor vc,
+                           at::Tensor pos, double scale) {
+  check_bf16(q, "q");
+  TORCH_CHECK(pos.scalar_type() == at::kInt && pos.numel() == 1,
+              "pos must be an int32 scalar tensor on device");
+  const int B = q.size(0), Hq = q.size(1);
+  const int T = kc.size(1), Hkv = kc.size(2);
+  auto o = at::empty_like(q);
+  decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                     o.data_ptr(), pos.data_ptr(), B, Hq, Hkv, T, 0,
+                     (float)scale, cur_stream());
   return o;
 }
 
@@ -388,4 +406,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe_tr", &mfma_probe_tr);
   m.def("transpose_bf16", &transpose_bf16);
   m.def("decode_attn", &decode_attn);
+  m.def("decode_attn_dev", &decode_attn_dev);
 }

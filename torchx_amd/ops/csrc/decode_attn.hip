@@ -19,7 +19,9 @@ decode_attn_kernel(const unsigned short* __restrict__ q,   // [B, Hq, 128]
                    const unsigned short* __restrict__ kc,  // [B, T, Hkv, 128]
                    const unsigned short* __restrict__ vc,  // [B, T, Hkv, 128]
                    unsigned short* __restrict__ o,         // [B, Hq, 128]
-                   int B, int Hq, int Hkv, int T, int L, float scale) {
+                   const int* __restrict__ len_dev,  // optional: L = *len_dev + 1
+                   int B, int Hq, int Hkv, int T, int L_host, float scale) {
+  const int L = len_dev ? (*len_dev + 1) : L_host;
   const int b = blockIdx.x / Hq;
   const int hq = blockIdx.x % Hq;
   const int hkv = hq / (Hq / Hkv);
@@ -112,11 +114,12 @@ decode_attn_kernel(const unsigned short* __restrict__ q,   // [B, Hq, 128]
 }
 
 extern "C" void decode_attn_launch(const void* q, const void* kc,
-                                   const void* vc, void* o, int B, int Hq,
+                                   const void* vc, void* o,
+                                   const void* len_dev, int B, int Hq,
                                    int Hkv, int T, int L, float scale,
                                    hipStream_t stream) {
   hipLaunchKernelGGL(decode_attn_kernel, dim3(B * Hq), dim3(256), 0, stream,
                      (const unsigned short*)q, (const unsigned short*)kc,
-                     (const unsigned short*)vc, (unsigned short*)o, B, Hq,
-                     Hkv, T, L, scale);
+                     (const unsigned short*)vc, (unsigned short*)o,
+                     (const int*)len_dev, B, Hq, Hkv, T, L, scale);
 }
