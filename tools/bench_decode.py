@@ -20,6 +20,8 @@ def main():
     p.add_argument("--batch", type=int, default=4)
     p.add_argument("--prompt", type=int, default=512)
     p.add_argument("--new", type=int, default=128)
+    p.add_argument("--graph", action="store_true",
+                   help="hipGraph decode loop (whole step as one replay)")
     args = p.parse_args()
 
     from torchx_amd.models.generate import KVCache, decode_step, prefill
@@ -41,16 +43,29 @@ def main():
     t_prefill = time.perf_counter() - t0
 
     nxt = logits.argmax(-1, keepdim=True)
-    # warm a few decode steps, then time
-    for _ in range(4):
-        nxt = decode_step(model, nxt, caches).argmax(-1, keepdim=True)
-    torch.cuda.synchronize()
-    t0 = time.perf_counter()
     steps = N
-    for _ in range(steps):
-        nxt = decode_step(model, nxt, caches).argmax(-1, keepdim=True)
-    torch.cuda.synchronize()
-    t_dec = time.perf_counter() - t0
+    if args.graph:
+        from torchx_amd.models.generate import GraphedDecoder
+
+        dec = GraphedDecoder(model, caches, B, nxt, start_pos=S0)
+        for _ in range(2):
+            dec.step()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            dec.step()
+        torch.cuda.synchronize()
+        t_dec = time.perf_counter() - t0
+    else:
+        # warm a few decode steps, then time
+        for _ in range(4):
+            nxt = decode_step(model, nxt, caches).argmax(-1, keepdim=True)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(steps):
+            nxt = decode_step(model, nxt, caches).argmax(-1, keepdim=True)
+        torch.cuda.synchronize()
+        t_dec = time.perf_counter() - t0
 
     print(json.dumps({
         "metric": "decode_tokens_per_second",
@@ -58,7 +73,7 @@ def main():
         "ms_per_decode_step": t_dec / steps * 1e3,
         "prefill_tokens_per_second": B * S0 / t_prefill,
         "batch": B, "prompt": S0, "new_tokens": steps,
-        "model": args.model, "dtype": "bf16", "data": "synthetic",
+        "model": args.model, "dtype": "bf16", "data": "synthetic", "graphed": bool(args.graph),
     }))
     return 0
 
