@@ -169,8 +169,9 @@ class GraphedDecoder:
                                   device=dev)
         torch.cuda.synchronize()
         self.graph = torch.cuda.CUDAGraph()
-        # the warmup AND the capture each execute one REAL step — their
-        # produced tokens are collected in init_tokens
+        # the eager warmup executes one REAL step (its token is collected
+        # in init_tokens); the capture only RECORDS — nothing runs and no
+        # state advances during it
         self.init_tokens = []
         self._step_body()                      # eager warm (same code path)
         torch.cuda.synchronize()
@@ -178,7 +179,6 @@ class GraphedDecoder:
         with torch.cuda.graph(self.graph):
             self._step_body()
         torch.cuda.synchronize()
-        self.init_tokens.append(self.tok.clone())
 
     @torch.no_grad()
     def _step_body(self) -> None:
