@@ -132,3 +132,30 @@ def test_generate_graphed_matches_eager():
             f"graphed tail {graphed[:, 24:].tolist()}; "
             f"first eager tok {e_log.argmax(-1).tolist()} vs graphed warm "
             f"{dec.init_tokens[0].reshape(-1).tolist()}")
+
+
+def test_decode_linear_cpu_fallback():
+    from torchx_amd import ops
+
+    x = torch.randn(2, 1, 512)
+    w = torch.randn(64, 512)
+    out = ops.decode_linear(x, w)
+    assert out.shape == (2, 1, 64)
+    assert torch.allclose(out, x @ w.t(), atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_gemv_bf16_matches_matmul():
+    from torchx_amd.ops import hip_ops
+
+    dev = torch.device("cuda:0")
+    torch.manual_seed(5)
+    for M, N, K in [(1, 100, 512), (4, 4096, 1024), (8, 333, 1536),
+                    (3, 17, 512), (4, 128256 // 16, 1024)]:
+        x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+        out = hip_ops().gemv_bf16(x, w)
+        ref = x.float() @ w.float().t()
+        err = (out.float() - ref).abs().max().item()
+        tol = 2e-2 * K ** 0.5
+        assert err < tol, (M, N, K, err, tol)

@@ -43,6 +43,14 @@ class KVCache:
         )
 
 
+def _dlin(mod: torch.nn.Linear, x: torch.Tensor) -> torch.Tensor:
+    """Decode-side linear: route the bias-free Llama projections through
+    the skinny-M GEMV kernel (weight-stream bound at decode batch sizes)."""
+    if mod.bias is None:
+        return ops.decode_linear(x, mod.weight)
+    return mod(x)
+
+
 def _split_qkv(qkv: torch.Tensor, cfg: LlamaConfig):
     B, S, _ = qkv.shape
     q, k, v = qkv.split([cfg.q_dim, cfg.kv_dim, cfg.kv_dim], dim=-1)
@@ -90,7 +98,7 @@ def decode_step(model: LlamaModel, token: torch.Tensor,
     x = model.embed(token)  # [B, 1, H]
     for blk, cache in zip(model.blocks, caches):
         xn = ops.rmsnorm(x, blk.attn_norm, cfg.rms_eps)
-        q, k, v = _split_qkv(blk.wqkv(xn), cfg)
+        q, k, v = _split_qkv(_dlin(blk.wqkv, xn), cfg)
         q = ops.rope(q, cos, sin)       # tables sliced at pos -> index 0
         k = ops.rope(k, cos, sin)
         cache.k[:, pos:pos + 1] = k
@@ -98,11 +106,11 @@ def decode_step(model: LlamaModel, token: torch.Tensor,
         cache.length = pos + 1
         o = ops.decode_attention(q.reshape(B, cfg.num_heads, cfg.head_dim),
                                  cache.k, cache.v, cache.length)
-        x = x + blk.wo(o.reshape(B, 1, cfg.q_dim))
+        x = x + _dlin(blk.wo, o.reshape(B, 1, cfg.q_dim))
         xn = ops.rmsnorm(x, blk.mlp_norm, cfg.rms_eps)
-        x = x + blk.wdown(ops.swiglu_packed(blk.wgu(xn)))
+        x = x + _dlin(blk.wdown, ops.swiglu_packed(_dlin(blk.wgu, xn)))
     x = ops.rmsnorm(x, model.final_norm, cfg.rms_eps)
-    return model.lm_head(x[:, -1])
+    return _dlin(model.lm_head, x[:, -1])
 
 
 @torch.no_grad()
@@ -189,7 +197,7 @@ class GraphedDecoder:
         x = model.embed(self.tok)
         for blk, cache in zip(model.blocks, self.caches):
             xn = ops.rmsnorm(x, blk.attn_norm, cfg.rms_eps)
-            q, k, v = _split_qkv(blk.wqkv(xn), cfg)
+            q, k, v = _split_qkv(_dlin(blk.wqkv, xn), cfg)
             q = ops.rope(q, cos, sin)
             k = ops.rope(k, cos, sin)
             cache.k.index_copy_(1, self.pos64, k)
@@ -200,11 +208,11 @@ class GraphedDecoder:
                 q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous(),
                 cache.k, cache.v, self.pos32,
                 1.0 / cfg.head_dim ** 0.5)
-            x = x + blk.wo(o.reshape(B, 1, cfg.q_dim))
+            x = x + _dlin(blk.wo, o.reshape(B, 1, cfg.q_dim))
             xn = ops.rmsnorm(x, blk.mlp_norm, cfg.rms_eps)
-            x = x + blk.wdown(ops.swiglu_packed(blk.wgu(xn)))
+            x = x + _dlin(blk.wdown, ops.swiglu_packed(_dlin(blk.wgu, xn)))
         x = ops.rmsnorm(x, model.final_norm, cfg.rms_eps)
-        logits = model.lm_head(x[:, -1])
+        logits = _dlin(model.lm_head, x[:, -1])
         self.tok.copy_(logits.argmax(-1, keepdim=True))
         self.pos32.add_(1)
         self.pos64.add_(1)

@@ -486,3 +486,21 @@ def decode_attention(q: torch.Tensor, kcache: torch.Tensor,
         return hip_ops().decode_attn(q.contiguous(), kcache, vcache,
                                      length, scale)
     return reference.decode_attention(q, kcache, vcache, length, scale)
+
+
+def decode_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Skinny-M linear for the decode path: x [..., M, K] with M <= 8
+    total rows -> x @ w.T via the W-stream-bound ``gemv_bf16`` kernel
+    (ops/csrc/gemv.hip). hipBLASLt's tile kernels reach only ~1/3 of the
+    HBM roof at decode batch sizes; this kernel reads each weight byte
+    exactly once. Falls back to torch.matmul off-GPU or when the shape
+    doesn't qualify (M > 8 or K % 512 != 0). Inference only (no autograd).
+    """
+    x2 = x.reshape(-1, x.shape[-1])
+    M, K = x2.shape
+    if _on_gpu(x) and M <= 8 and K % 512 == 0 \
+            and w.dtype == torch.bfloat16 and w.is_contiguous():
+        out = hip_ops().gemv_bf16(x2.contiguous(), w)
+    else:
+        out = x2 @ w.t()
+    return out.reshape(*x.shape[:-1], w.shape[0])

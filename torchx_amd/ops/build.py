@@ -25,6 +25,7 @@ SOURCES = [
     "cross_entropy.hip",
     "attention.hip",
     "decode_attn.hip",
+    "gemv.hip",
     "fp8_cast.hip",
     "probe.hip",
     "bindings.cpp",

@@ -41,6 +41,8 @@ void transpose_bf16_launch(const void*, void*, long, long, hipStream_t);
 void decode_attn_launch(const void*, const void*, const void*, void*,
                         const void*, int, int, int, int, int, float,
                         hipStream_t);
+void gemv_bf16_launch(const void*, const void*, void*, int, int, int,
+                      hipStream_t);
 }
 
 namespace {
@@ -362,6 +364,22 @@ at::Tensor decode_attn_dev(at::Tensor q, at::Tensor kc, at::Tensor vc,
   return o;
 }
 
+// ---- skinny-M GEMV (decode linears) ---------------------------------------
+// x [M, K] bf16 (M <= 8), w [N, K] bf16 (nn.Linear layout) -> out [M, N].
+at::Tensor gemv_bf16(at::Tensor x, at::Tensor w) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2, "gemv: 2D only");
+  const long M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(M >= 1 && M <= 8, "gemv: M must be 1..8");
+  TORCH_CHECK(w.size(1) == K, "gemv: K mismatch");
+  TORCH_CHECK(K % 512 == 0, "gemv: K must be a multiple of 512");
+  auto out = at::empty({M, N}, x.options());
+  gemv_bf16_launch(x.data_ptr(), w.data_ptr(), out.data_ptr(), (int)M,
+                   (int)N, (int)K, cur_stream());
+  return out;
+}
+
 // ---- probe ----------------------------------------------------------------
 std::vector<at::Tensor> mfma_probe_tr(at::Tensor a, at::Tensor b) {
   check_bf16(a, "a");
@@ -407,4 +425,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("transpose_bf16", &transpose_bf16);
   m.def("decode_attn", &decode_attn);
   m.def("decode_attn_dev", &decode_attn_dev);
+  m.def("gemv_bf16", &gemv_bf16);
 }
