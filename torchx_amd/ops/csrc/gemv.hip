@@ -16,7 +16,19 @@
 // and shared across the wave's rows. Accumulate f32, reduce each
 // (m, row) partial across the wave with 6 shfl_xor steps, lane 0 writes.
 // K loop unrolled 4x to keep ~8 outstanding 16-B loads per lane.
+// The MAC itself is v_dot2c_f32_bf16 (2 bf16 MACs/VALU-op, no converts):
+// scalar convert+fma costs ~9 VALU ops per W element at M=4, enough to
+// make the kernel VALU-bound instead of W-stream-bound.
 #include "common.h"
+
+typedef __bf16 bf16x2 __attribute__((ext_vector_type(2)));
+
+__device__ __forceinline__ float dot2_bf16(ushort8 a, ushort8 b, float acc,
+                                           int j2) {
+  return __builtin_amdgcn_fdot2_f32_bf16(((const bf16x2*)&a)[j2],
+                                         ((const bf16x2*)&b)[j2], acc,
+                                         false);
+}
 
 template <int M, int RPW>
 __global__ void __launch_bounds__(256)
@@ -55,12 +67,10 @@ gemv_bf16_kernel(const unsigned short* __restrict__ x,  // [M, K]
       #pragma unroll
       for (int u = 0; u < 4; ++u)
         #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const float wv = bf16_to_f32(wf[u][j]);
+        for (int j2 = 0; j2 < 4; ++j2)
           #pragma unroll
           for (int m = 0; m < M; ++m)
-            acc[m][r] = fmaf(bf16_to_f32(xf[m][u][j]), wv, acc[m][r]);
-        }
+            acc[m][r] = dot2_bf16(xf[m][u], wf[u], acc[m][r], j2);
     }
   }
   for (; kc < K; kc += 512) {  // tail chunks
@@ -73,12 +83,10 @@ gemv_bf16_kernel(const unsigned short* __restrict__ x,  // [M, K]
       if (n0 + r >= N) break;
       ushort8 wf = *(const ushort8*)(w + (long)(n0 + r) * K + kc);
       #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const float wv = bf16_to_f32(wf[j]);
+      for (int j2 = 0; j2 < 4; ++j2)
         #pragma unroll
         for (int m = 0; m < M; ++m)
-          acc[m][r] = fmaf(bf16_to_f32(xf[m][j]), wv, acc[m][r]);
-      }
+          acc[m][r] = dot2_bf16(xf[m], wf, acc[m][r], j2);
     }
   }
 
@@ -103,14 +111,21 @@ gemv_bf16_kernel(const unsigned short* __restrict__ x,  // [M, K]
   }
 }
 
-template <int M>
-static void gemv_launch_m(const void* x, const void* w, void* o, int N,
-                          int K, hipStream_t stream) {
-  constexpr int RPW = 2;
+template <int M, int RPW>
+static void gemv_launch_rpw(const void* x, const void* w, void* o, int N,
+                            int K, hipStream_t stream) {
   const int blocks = (N + 4 * RPW - 1) / (4 * RPW);
   hipLaunchKernelGGL((gemv_bf16_kernel<M, RPW>), dim3(blocks), dim3(256), 0,
                      stream, (const unsigned short*)x,
                      (const unsigned short*)w, (unsigned short*)o, N, K);
+}
+
+template <int M>
+static void gemv_launch_m(const void* x, const void* w, void* o, int N,
+                          int K, hipStream_t stream) {
+  // RPW=2 measured best at every decode shape (RPW=4 and an LDS-staged
+  // x variant were both slower; see profiles/README.md)
+  gemv_launch_rpw<M, 2>(x, w, o, N, K, stream);
 }
 
 extern "C" void gemv_bf16_launch(const void* x, const void* w, void* o,
