@@ -43,14 +43,6 @@ class KVCache:
         )
 
 
-def _dlin(mod: torch.nn.Linear, x: torch.Tensor) -> torch.Tensor:
-    """Decode-side linear: route the bias-free Llama projections through
-    the skinny-M GEMV kernel (weight-stream bound at decode batch sizes)."""
-    if mod.bias is None:
-        return ops.decode_linear(x, mod.weight)
-    return mod(x)
-
-
 def _split_qkv(qkv: torch.Tensor, cfg: LlamaConfig):
     B, S, _ = qkv.shape
     q, k, v = qkv.split([cfg.q_dim, cfg.kv_dim, cfg.kv_dim], dim=-1)
@@ -86,31 +78,40 @@ def prefill(model: LlamaModel, tokens: torch.Tensor,
 
 
 @torch.no_grad()
-def decode_step(model: LlamaModel, token: torch.Tensor,
-                caches: list) -> torch.Tensor:
+def decode_step(model: LlamaModel, token: torch.Tensor, caches: list,
+                pos_dev: Optional[torch.Tensor] = None) -> torch.Tensor:
     """One token [B, 1] -> next-position logits [B, vocab], appending to
-    the caches."""
+    the caches. Fully fused decode layer (9 dispatches on GPU): skinny-M
+    GEMV x4, rope+cache-append x1, flash-decode attention x1, fused
+    residual-add+rmsnorm x2, swiglu x1. With ``pos_dev`` (int32 device
+    scalar) the step is hipGraph-capturable: the cache position comes off
+    the device and no host state is read."""
     cfg = model.cfg
     B = token.shape[0]
-    pos = caches[0].length
-    cos = model.rope_cos[pos:pos + 1]
-    sin = model.rope_sin[pos:pos + 1]
-    x = model.embed(token)  # [B, 1, H]
-    for blk, cache in zip(model.blocks, caches):
-        xn = ops.rmsnorm(x, blk.attn_norm, cfg.rms_eps)
-        q, k, v = _split_qkv(_dlin(blk.wqkv, xn), cfg)
-        q = ops.rope(q, cos, sin)       # tables sliced at pos -> index 0
-        k = ops.rope(k, cos, sin)
-        cache.k[:, pos:pos + 1] = k
-        cache.v[:, pos:pos + 1] = v
-        cache.length = pos + 1
-        o = ops.decode_attention(q.reshape(B, cfg.num_heads, cfg.head_dim),
-                                 cache.k, cache.v, cache.length)
-        x = x + _dlin(blk.wo, o.reshape(B, 1, cfg.q_dim))
-        xn = ops.rmsnorm(x, blk.mlp_norm, cfg.rms_eps)
-        x = x + _dlin(blk.wdown, ops.swiglu_packed(_dlin(blk.wgu, xn)))
-    x = ops.rmsnorm(x, model.final_norm, cfg.rms_eps)
-    return _dlin(model.lm_head, x[:, -1])
+    blocks = model.blocks
+    pos = caches[0].length if pos_dev is None else pos_dev
+    x = model.embed(token).reshape(B, -1)  # [B, H] residual stream
+    _, xn = ops.rmsnorm_res(x, None, blocks[0].attn_norm, cfg.rms_eps)
+    for i, (blk, cache) in enumerate(zip(blocks, caches)):
+        qkv = ops.decode_linear(xn, blk.wqkv.weight)
+        q = ops.decode_rope_cache(qkv, cache.k, cache.v, model.rope_cos,
+                                  model.rope_sin, pos, cfg.num_heads)
+        if pos_dev is None:
+            cache.length = pos + 1
+            o = ops.decode_attention(q, cache.k, cache.v, pos + 1)
+        else:
+            from torchx_amd.ops import hip_ops
+
+            o = hip_ops().decode_attn_dev(q, cache.k, cache.v, pos_dev,
+                                          1.0 / cfg.head_dim ** 0.5)
+        a = ops.decode_linear(o.reshape(B, -1), blk.wo.weight)
+        x, xn = ops.rmsnorm_res(x, a, blk.mlp_norm, cfg.rms_eps)
+        gu = ops.decode_linear(xn, blk.wgu.weight)
+        m = ops.decode_linear(ops.swiglu_packed(gu), blk.wdown.weight)
+        w_next = (blocks[i + 1].attn_norm if i + 1 < len(blocks)
+                  else model.final_norm)
+        x, xn = ops.rmsnorm_res(x, m, w_next, cfg.rms_eps)
+    return ops.decode_linear(xn, model.lm_head.weight)
 
 
 @torch.no_grad()
@@ -156,8 +157,8 @@ def generate(
 class GraphedDecoder:
     """hipGraph-captured decode loop: the whole per-token step (embed,
     32 layers, lm_head, argmax, cache append, position bump) replays as
-    ONE graph with zero host work — the eager decode step is launch-bound
-    (~450 small kernels). The cache length and rope position are driven
+    ONE graph with zero host work (the fused eager step is ~290
+    dispatches; un-fused it was ~450 and launch-bound). The cache length and rope position are driven
     by a device int32 scalar that the captured step increments itself,
     so one capture serves every subsequent token.
 
@@ -172,8 +173,6 @@ class GraphedDecoder:
         dev = next(model.parameters()).device
         self.tok = first_token.clone()                      # [B, 1] int64
         self.pos32 = torch.tensor([start_pos], dtype=torch.int32,
-                                  device=dev)
-        self.pos64 = torch.tensor([start_pos], dtype=torch.int64,
                                   device=dev)
         torch.cuda.synchronize()
         self.graph = torch.cuda.CUDAGraph()
@@ -190,32 +189,10 @@ class GraphedDecoder:
 
     @torch.no_grad()
     def _step_body(self) -> None:
-        model, cfg = self.model, self.model.cfg
-        B = self.tok.shape[0]
-        cos = model.rope_cos.index_select(0, self.pos64)
-        sin = model.rope_sin.index_select(0, self.pos64)
-        x = model.embed(self.tok)
-        for blk, cache in zip(model.blocks, self.caches):
-            xn = ops.rmsnorm(x, blk.attn_norm, cfg.rms_eps)
-            q, k, v = _split_qkv(_dlin(blk.wqkv, xn), cfg)
-            q = ops.rope(q, cos, sin)
-            k = ops.rope(k, cos, sin)
-            cache.k.index_copy_(1, self.pos64, k)
-            cache.v.index_copy_(1, self.pos64, v)
-            from torchx_amd.ops import hip_ops
-
-            o = hip_ops().decode_attn_dev(
-                q.reshape(B, cfg.num_heads, cfg.head_dim).contiguous(),
-                cache.k, cache.v, self.pos32,
-                1.0 / cfg.head_dim ** 0.5)
-            x = x + _dlin(blk.wo, o.reshape(B, 1, cfg.q_dim))
-            xn = ops.rmsnorm(x, blk.mlp_norm, cfg.rms_eps)
-            x = x + _dlin(blk.wdown, ops.swiglu_packed(_dlin(blk.wgu, xn)))
-        x = ops.rmsnorm(x, model.final_norm, cfg.rms_eps)
-        logits = _dlin(model.lm_head, x[:, -1])
+        logits = decode_step(self.model, self.tok, self.caches,
+                             pos_dev=self.pos32)
         self.tok.copy_(logits.argmax(-1, keepdim=True))
         self.pos32.add_(1)
-        self.pos64.add_(1)
 
     def step(self) -> torch.Tensor:
         """Replay one decode step; returns the new token [B, 1]."""

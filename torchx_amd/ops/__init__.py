@@ -488,6 +488,48 @@ def decode_attention(q: torch.Tensor, kcache: torch.Tensor,
     return reference.decode_attention(q, kcache, vcache, length, scale)
 
 
+def rmsnorm_res(x: torch.Tensor, res: Optional[torch.Tensor],
+                w: torch.Tensor, eps: float = 1e-5):
+    """Fused residual-add + RMSNorm for the decode path: returns
+    ``(s, y)`` with ``s = x + res`` (the new residual stream, ``x`` when
+    ``res`` is None) and ``y = rmsnorm(s) * w``. Inference only."""
+    if _on_gpu(x):
+        s, y = hip_ops().rmsnorm_res(x.contiguous(),
+                                     res.contiguous() if res is not None
+                                     else None, w.contiguous(), eps)
+        return s, y
+    s = x if res is None else (x + res)
+    return s, reference.rmsnorm(s, w, eps)
+
+
+def decode_rope_cache(qkv: torch.Tensor, kcache: torch.Tensor,
+                      vcache: torch.Tensor, cos: torch.Tensor,
+                      sin: torch.Tensor, pos,
+                      num_heads: int) -> torch.Tensor:
+    """Fused decode rope + KV-cache append: qkv [B, (Hq+2*Hkv)*D] for one
+    position -> roped q [B, Hq, D]; ropes k and appends k/v to the caches
+    at row ``pos`` (host int, or int32 device scalar for hipGraph capture).
+    cos/sin are the FULL f32 [S, D/2] tables."""
+    B = qkv.shape[0]
+    Hkv, D = kcache.shape[2], kcache.shape[3]
+    if _on_gpu(qkv):
+        if isinstance(pos, torch.Tensor):
+            return hip_ops().decode_rope_cache(qkv.contiguous(), kcache,
+                                               vcache, cos, sin, 0, pos)
+        return hip_ops().decode_rope_cache(qkv.contiguous(), kcache, vcache,
+                                           cos, sin, int(pos), None)
+    p = int(pos.item()) if isinstance(pos, torch.Tensor) else int(pos)
+    q, k, v = qkv.split([num_heads * D, Hkv * D, Hkv * D], dim=-1)
+    q = q.reshape(B, 1, num_heads, D)
+    k = k.reshape(B, 1, Hkv, D)
+    cs, sn = cos[p:p + 1], sin[p:p + 1]
+    q = rope(q.contiguous(), cs, sn)
+    k = rope(k.contiguous(), cs, sn)
+    kcache[:, p:p + 1] = k
+    vcache[:, p:p + 1] = v.reshape(B, 1, Hkv, D)
+    return q.reshape(B, num_heads, D)
+
+
 def decode_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Skinny-M linear for the decode path: x [..., M, K] with M <= 8
     total rows -> x @ w.T via the W-stream-bound ``gemv_bf16`` kernel

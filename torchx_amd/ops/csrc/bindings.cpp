@@ -43,6 +43,11 @@ void decode_attn_launch(const void*, const void*, const void*, void*,
                         hipStream_t);
 void gemv_bf16_launch(const void*, const void*, void*, int, int, int,
                       hipStream_t);
+void decode_rope_cache_launch(const void*, void*, void*, void*, const void*,
+                              const void*, const void*, int, int, int, int,
+                              int, hipStream_t);
+void rmsnorm_res_launch(const void*, const void*, const void*, void*, void*,
+                        long, int, float, hipStream_t);
 }
 
 namespace {
@@ -380,6 +385,67 @@ at::Tensor gemv_bf16(at::Tensor x, at::Tensor w) {
   return out;
 }
 
+// ---- fused decode rope + cache append -------------------------------------
+// qkv [B, (Hq+2*Hkv)*128] (one position); ropes q/k with cos/sin [S, 64]
+// f32 tables at row `pos` (host int, or device int32 scalar when `pos_dev`
+// is given), appends k/v to the caches at `pos`, returns q [B, Hq, 128].
+at::Tensor decode_rope_cache(at::Tensor qkv, at::Tensor kc, at::Tensor vc,
+                             at::Tensor cos_t, at::Tensor sin_t, long pos,
+                             c10::optional<at::Tensor> pos_dev) {
+  check_bf16(qkv, "qkv");
+  check_bf16(kc, "kc");
+  check_bf16(vc, "vc");
+  check_f32(cos_t, "cos");
+  check_f32(sin_t, "sin");
+  const int B = qkv.size(0), T = kc.size(1), Hkv = kc.size(2);
+  TORCH_CHECK(kc.size(3) == 128 && cos_t.size(1) == 64,
+              "head_dim must be 128");
+  const long nh = qkv.size(1) / 128;
+  const int Hq = (int)nh - 2 * Hkv;
+  TORCH_CHECK(Hq >= 1 && qkv.size(1) == nh * 128, "qkv width mismatch");
+  const void* pd = nullptr;
+  if (pos_dev.has_value()) {
+    TORCH_CHECK(pos_dev->scalar_type() == at::kInt && pos_dev->numel() == 1,
+                "pos_dev must be an int32 scalar tensor");
+    pd = pos_dev->data_ptr();
+  } else {
+    TORCH_CHECK(pos >= 0 && pos < T, "pos out of cache range");
+    TORCH_CHECK(pos < cos_t.size(0), "pos beyond rope table");
+  }
+  auto q = at::empty({B, (long)Hq, 128}, qkv.options());
+  decode_rope_cache_launch(qkv.data_ptr(), q.data_ptr(), kc.data_ptr(),
+                           vc.data_ptr(), cos_t.data_ptr(), sin_t.data_ptr(),
+                           pd, B, Hq, Hkv, T, (int)pos, cur_stream());
+  return q;
+}
+
+// ---- fused residual add + rmsnorm (inference) ------------------------------
+// s = x + res, y = rmsnorm(s) * w; returns (s, y). res omitted -> plain
+// rmsnorm (s aliases x).
+std::vector<at::Tensor> rmsnorm_res(at::Tensor x,
+                                    c10::optional<at::Tensor> res,
+                                    at::Tensor w, double eps) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  const long H = x.size(-1);
+  TORCH_CHECK(H % 2048 == 0 && H <= 8192, "rmsnorm: H must be k*2048, <=8192");
+  const long R = x.numel() / H;
+  auto y = at::empty_like(x);
+  at::Tensor s = x;
+  const void* rp = nullptr;
+  void* sp = nullptr;
+  if (res.has_value()) {
+    check_bf16(*res, "res");
+    TORCH_CHECK(res->sizes() == x.sizes(), "res shape mismatch");
+    s = at::empty_like(x);
+    rp = res->data_ptr();
+    sp = s.data_ptr();
+  }
+  rmsnorm_res_launch(x.data_ptr(), rp, w.data_ptr(), sp, y.data_ptr(), R,
+                     (int)H, (float)eps, cur_stream());
+  return {s, y};
+}
+
 // ---- probe ----------------------------------------------------------------
 std::vector<at::Tensor> mfma_probe_tr(at::Tensor a, at::Tensor b) {
   check_bf16(a, "a");
@@ -426,4 +492,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_attn", &decode_attn);
   m.def("decode_attn_dev", &decode_attn_dev);
   m.def("gemv_bf16", &gemv_bf16);
+  m.def("decode_rope_cache", &decode_rope_cache,
+        py::arg("qkv"), py::arg("kc"), py::arg("vc"), py::arg("cos"),
+        py::arg("sin"), py::arg("pos") = 0,
+        py::arg("pos_dev") = c10::nullopt);
+  m.def("rmsnorm_res", &rmsnorm_res,
+        py::arg("x"), py::arg("res"), py::arg("w"), py::arg("eps"));
 }

@@ -123,3 +123,59 @@ extern "C" void decode_attn_launch(const void* q, const void* kc,
                      (const unsigned short*)vc, (unsigned short*)o,
                      (const int*)len_dev, B, Hq, Hkv, T, L, scale);
 }
+
+// Fused decode-side rope + cache append: consumes the packed wqkv output
+// [B, (Hq+2*Hkv)*128] for ONE position, ropes q and k (half-split pairing,
+// same convention as rope_fwdbwd_kernel), writes the roped k and the raw v
+// into the caches at row `pos`, and emits q contiguous [B, Hq, 128] for
+// decode_attn. Replaces ~6 tiny kernels (2x rope, 2x index_copy, split
+// copies) per layer per token; one wave per head, latency-floor bound.
+// `pos` comes from the host (eager) or a device int32 scalar (hipGraph).
+__global__ void __launch_bounds__(64)
+decode_rope_cache_kernel(const unsigned short* __restrict__ qkv,
+                         unsigned short* __restrict__ qout,  // [B, Hq, 128]
+                         unsigned short* __restrict__ kc,    // [B,T,Hkv,128]
+                         unsigned short* __restrict__ vc,
+                         const float* __restrict__ cos_t,    // [S, 64]
+                         const float* __restrict__ sin_t,
+                         const int* __restrict__ pos_dev,
+                         int B, int Hq, int Hkv, int T, int pos_host) {
+  const int pos = pos_dev ? *pos_dev : pos_host;
+  const int nh = Hq + 2 * Hkv;
+  const int b = blockIdx.x / nh;
+  const int h = blockIdx.x % nh;
+  const int lane = threadIdx.x;  // 0..63: one rotation pair (lane, lane+64)
+  const unsigned short* src = qkv + ((long)b * nh + h) * HD;
+  if (h < Hq + Hkv) {  // q or k head: rotate
+    const float x1 = bf16_to_f32(src[lane]);
+    const float x2 = bf16_to_f32(src[lane + 64]);
+    const float c = cos_t[(long)pos * 64 + lane];
+    const float s = sin_t[(long)pos * 64 + lane];
+    const unsigned short y1 = f32_to_bf16(fmaf(x1, c, -s * x2));
+    const unsigned short y2 = f32_to_bf16(fmaf(x2, c, s * x1));
+    unsigned short* dst =
+        (h < Hq) ? qout + ((long)b * Hq + h) * HD
+                 : kc + (((long)b * T + pos) * Hkv + (h - Hq)) * HD;
+    dst[lane] = y1;
+    dst[lane + 64] = y2;
+  } else {  // v head: plain copy into the cache
+    unsigned short* dst =
+        vc + (((long)b * T + pos) * Hkv + (h - Hq - Hkv)) * HD;
+    dst[lane] = src[lane];
+    dst[lane + 64] = src[lane + 64];
+  }
+}
+
+extern "C" void decode_rope_cache_launch(const void* qkv, void* qout,
+                                         void* kc, void* vc,
+                                         const void* cos_t, const void* sin_t,
+                                         const void* pos_dev, int B, int Hq,
+                                         int Hkv, int T, int pos_host,
+                                         hipStream_t stream) {
+  hipLaunchKernelGGL(decode_rope_cache_kernel, dim3(B * (Hq + 2 * Hkv)),
+                     dim3(64), 0, stream, (const unsigned short*)qkv,
+                     (unsigned short*)qout, (unsigned short*)kc,
+                     (unsigned short*)vc, (const float*)cos_t,
+                     (const float*)sin_t, (const int*)pos_dev, B, Hq, Hkv, T,
+                     pos_host);
+}

@@ -179,3 +179,82 @@ extern "C" void rmsnorm_bwd_launch(const void* dy, const void* x,
                      (const unsigned short*)dy, (const unsigned short*)x,
                      (const float*)invrms, (float*)dw, rows, H);
 }
+
+// Fused residual-add + RMSNorm (decode/inference): s = x + res (the new
+// residual stream), y = rmsnorm(s) * w. Same one-block-per-row shape as
+// the forward kernel; res == nullptr degrades to plain rmsnorm with no s
+// write. No invrms save — inference only, replaces an elementwise add +
+// rmsnorm pair (2 dispatches -> 1) per use in the decode step.
+template <int ITERS>  // ITERS = H / (256 * 8)
+__global__ void __launch_bounds__(256)
+rmsnorm_res_fwd_kernel(const unsigned short* __restrict__ x,
+                       const unsigned short* __restrict__ res,
+                       const unsigned short* __restrict__ w,
+                       unsigned short* __restrict__ s_out,
+                       unsigned short* __restrict__ y,
+                       int H, float eps) {
+  __shared__ float red[4];
+  const long row = blockIdx.x;
+  const unsigned short* xr = x + row * (long)H;
+  unsigned short* yr = y + row * (long)H;
+
+  float xs[ITERS][8];
+  float acc = 0.f;
+  #pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    int i = (it * 256 + threadIdx.x) * 8;
+    ushort8 v = *(const ushort8*)(xr + i);
+    ushort8 rv;
+    if (res) rv = *(const ushort8*)(res + row * (long)H + i);
+    ushort8 sv;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf16_to_f32(v[j]);
+      if (res) {
+        f += bf16_to_f32(rv[j]);
+        sv[j] = f32_to_bf16(f);
+        f = bf16_to_f32(sv[j]);  // norm sees the bf16-rounded stream
+      }
+      xs[it][j] = f;
+      acc = fmaf(f, f, acc);
+    }
+    if (res) *(ushort8*)(s_out + row * (long)H + i) = sv;
+  }
+  float total = block_reduce<4>(acc, red,
+      [] __device__ (float a, float b) { return a + b; });
+  float r = rsqrtf(total / (float)H + eps);
+
+  #pragma unroll
+  for (int it = 0; it < ITERS; ++it) {
+    int i = (it * 256 + threadIdx.x) * 8;
+    ushort8 wv = *(const ushort8*)(w + i);
+    ushort8 ov;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      ov[j] = f32_to_bf16(xs[it][j] * r * bf16_to_f32(wv[j]));
+    }
+    *(ushort8*)(yr + i) = ov;
+  }
+}
+
+extern "C" void rmsnorm_res_launch(const void* x, const void* res,
+                                   const void* w, void* s_out, void* y,
+                                   long R, int H, float eps,
+                                   hipStream_t stream) {
+  #define RMSRES_CASE(N)                                                   \
+    case N:                                                                \
+      hipLaunchKernelGGL((rmsnorm_res_fwd_kernel<N>), dim3(R), dim3(256),  \
+                         0, stream, (const unsigned short*)x,              \
+                         (const unsigned short*)res,                       \
+                         (const unsigned short*)w, (unsigned short*)s_out, \
+                         (unsigned short*)y, H, eps);                      \
+      break;
+  switch (H / 2048) {
+    RMSRES_CASE(1)
+    RMSRES_CASE(2)
+    RMSRES_CASE(3)
+    RMSRES_CASE(4)
+    default: break;  // binding guards H
+  }
+  #undef RMSRES_CASE
+}
