@@ -43,6 +43,9 @@ void decode_attn_launch(const void*, const void*, const void*, void*,
                         hipStream_t);
 void gemv_bf16_launch(const void*, const void*, void*, int, int, int,
                       hipStream_t);
+void decode_attn_split_launch(const void*, const void*, const void*, void*,
+                              void*, void*, const void*, int, int, int, int,
+                              int, int, float, hipStream_t);
 void decode_rope_cache_launch(const void*, void*, void*, void*, const void*,
                               const void*, const void*, int, int, int, int,
                               int, hipStream_t);
@@ -346,9 +349,21 @@ at::Tensor decode_attn(at::Tensor q, at::Tensor kc, at::Tensor vc, long L,
   TORCH_CHECK(L >= 1 && L <= T, "invalid cache length");
   TORCH_CHECK(Hq % Hkv == 0, "GQA group mismatch");
   auto o = at::empty_like(q);
-  decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
-                     o.data_ptr(), nullptr, B, Hq, Hkv, T, (int)L,
-                     (float)scale, cur_stream());
+  const int split = std::max(1, std::min(16, 512 / (B * Hq)));
+  if (split > 1) {
+    auto ml = at::empty({(long)B * Hq * split * 2},
+                        q.options().dtype(at::kFloat));
+    auto oacc = at::empty({(long)B * Hq * split * 128},
+                          q.options().dtype(at::kFloat));
+    decode_attn_split_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                             ml.data_ptr(), oacc.data_ptr(), o.data_ptr(),
+                             nullptr, B, Hq, Hkv, T, (int)L, split,
+                             (float)scale, cur_stream());
+  } else {
+    decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                       o.data_ptr(), nullptr, B, Hq, Hkv, T, (int)L,
+                       (float)scale, cur_stream());
+  }
   return o;
 }
 
@@ -363,9 +378,21 @@ at::Tensor decode_attn_dev(at::Tensor q, at::Tensor kc, at::Tensor vc,
   const int B = q.size(0), Hq = q.size(1);
   const int T = kc.size(1), Hkv = kc.size(2);
   auto o = at::empty_like(q);
-  decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
-                     o.data_ptr(), pos.data_ptr(), B, Hq, Hkv, T, 0,
-                     (float)scale, cur_stream());
+  const int split = std::max(1, std::min(16, 512 / (B * Hq)));
+  if (split > 1) {
+    auto ml = at::empty({(long)B * Hq * split * 2},
+                        q.options().dtype(at::kFloat));
+    auto oacc = at::empty({(long)B * Hq * split * 128},
+                          q.options().dtype(at::kFloat));
+    decode_attn_split_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                             ml.data_ptr(), oacc.data_ptr(), o.data_ptr(),
+                             pos.data_ptr(), B, Hq, Hkv, T, 0, split,
+                             (float)scale, cur_stream());
+  } else {
+    decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
+                       o.data_ptr(), pos.data_ptr(), B, Hq, Hkv, T, 0,
+                       (float)scale, cur_stream());
+  }
   return o;
 }
 

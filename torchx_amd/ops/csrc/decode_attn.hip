@@ -179,3 +179,143 @@ extern "C" void decode_rope_cache_launch(const void* qkv, void* qout,
                      (const float*)sin_t, (const int*)pos_dev, B, Hq, Hkv, T,
                      pos_host);
 }
+
+// ---- split-K (flash-decode) variant ---------------------------------------
+// At decode batch 4 the single-pass kernel launches only B*Hq = 128 blocks
+// on 256 CUs — half the chip idles and each block serially streams its
+// whole K/V slice (~25 us/layer measured). Pass 1 splits the cache length
+// across gridDim.y blocks, each producing an UNNORMALIZED partial
+// (m, l, o[128]) in f32 scratch; pass 2 (one small block per (b, hq))
+// combines the partials exactly like the in-block wave merge.
+__global__ void __launch_bounds__(256)
+decode_attn_split_kernel(const unsigned short* __restrict__ q,
+                         const unsigned short* __restrict__ kc,
+                         const unsigned short* __restrict__ vc,
+                         float* __restrict__ ml,    // [B*Hq, SPLIT, 2]
+                         float* __restrict__ oacc,  // [B*Hq, SPLIT, 128]
+                         const int* __restrict__ len_dev,
+                         int B, int Hq, int Hkv, int T, int L_host,
+                         float scale) {
+  const int L = len_dev ? (*len_dev + 1) : L_host;
+  const int bq = blockIdx.x;
+  const int b = bq / Hq;
+  const int hq = bq % Hq;
+  const int hkv = hq / (Hq / Hkv);
+  const int split = gridDim.y;
+  const int chunk = (L + split - 1) / split;
+  const int t0 = blockIdx.y * chunk;
+  const int t1 = min(L, t0 + chunk);
+
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const int tsub = lane >> 4;
+  const int dchunk = (lane & 15) * 8;
+
+  const unsigned short* qp = q + ((long)b * Hq + hq) * HD + dchunk;
+  float qv[8];
+  {
+    ushort8 v = *(const ushort8*)qp;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) qv[j] = bf16_to_f32(v[j]) * scale;
+  }
+
+  const long row_stride = (long)Hkv * HD;
+  const unsigned short* kb = kc + (long)b * T * row_stride + (long)hkv * HD;
+  const unsigned short* vb = vc + (long)b * T * row_stride + (long)hkv * HD;
+
+  float m_run = -3.0e38f, l_run = 0.f;
+  float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  for (int t = t0 + 4 * wave + tsub; t < t1; t += 16) {
+    const unsigned short* kr = kb + (long)t * row_stride + dchunk;
+    ushort8 kv8 = *(const ushort8*)kr;
+    float s = 0.f;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j) s = fmaf(qv[j], bf16_to_f32(kv8[j]), s);
+    #pragma unroll
+    for (int off = 8; off >= 1; off >>= 1) s += __shfl_xor(s, off, 16);
+    const float m_new = fmaxf(m_run, s);
+    const float alpha = __builtin_amdgcn_exp2f(1.44269504f * (m_run - m_new));
+    const float p = __builtin_amdgcn_exp2f(1.44269504f * (s - m_new));
+    l_run = l_run * alpha + p;
+    const unsigned short* vr = vb + (long)t * row_stride + dchunk;
+    ushort8 vv8 = *(const ushort8*)vr;
+    #pragma unroll
+    for (int j = 0; j < 8; ++j)
+      acc[j] = fmaf(acc[j], alpha, p * bf16_to_f32(vv8[j]));
+    m_run = m_new;
+  }
+
+  // block-combine the 16 (wave, row-slot) partials, wave 0 writes scratch
+  __shared__ float sm[16], sl[16], so[16][128];
+  const int slot = wave * 4 + tsub;
+  if ((lane & 15) == 0) {
+    sm[slot] = m_run;
+    sl[slot] = l_run;
+  }
+  #pragma unroll
+  for (int j = 0; j < 8; ++j) so[slot][dchunk + j] = acc[j];
+  __syncthreads();
+  if (wave == 0) {
+    float m_t = -3.0e38f;
+    #pragma unroll
+    for (int i = 0; i < 16; ++i) m_t = fmaxf(m_t, sm[i]);
+    float l_t = 0.f;
+    float out[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    #pragma unroll
+    for (int i = 0; i < 16; ++i) {
+      const float w = __builtin_amdgcn_exp2f(1.44269504f * (sm[i] - m_t));
+      l_t += sl[i] * w;
+      #pragma unroll
+      for (int j = 0; j < 8; ++j)
+        out[j] = fmaf(so[i][dchunk + j], w, out[j]);
+    }
+    const long pbase = ((long)bq * split + blockIdx.y);
+    if (lane == 0) {
+      ml[pbase * 2] = m_t;       // m = -inf, l = 0 for an empty slice
+      ml[pbase * 2 + 1] = l_t;
+    }
+    if (tsub == 0) {
+      #pragma unroll
+      for (int j = 0; j < 8; ++j) oacc[pbase * HD + dchunk + j] = out[j];
+    }
+  }
+}
+
+// pass 2: one 128-thread block per (b, hq); thread d combines SPLIT
+// partials and writes the normalized bf16 output element.
+__global__ void __launch_bounds__(128)
+decode_attn_merge_kernel(const float* __restrict__ ml,
+                         const float* __restrict__ oacc,
+                         unsigned short* __restrict__ o,  // [B, Hq, 128]
+                         int split) {
+  const int bq = blockIdx.x;
+  const int d = threadIdx.x;
+  float m_t = -3.0e38f;
+  for (int s = 0; s < split; ++s)
+    m_t = fmaxf(m_t, ml[((long)bq * split + s) * 2]);
+  float l_t = 0.f, out = 0.f;
+  for (int s = 0; s < split; ++s) {
+    const long p = (long)bq * split + s;
+    const float w = __builtin_amdgcn_exp2f(1.44269504f * (ml[p * 2] - m_t));
+    l_t += ml[p * 2 + 1] * w;
+    out = fmaf(oacc[p * HD + d], w, out);
+  }
+  const float inv = (l_t > 0.f) ? 1.0f / l_t : 0.f;
+  o[(long)bq * HD + d] = f32_to_bf16(out * inv);
+}
+
+extern "C" void decode_attn_split_launch(const void* q, const void* kc,
+                                         const void* vc, void* ml,
+                                         void* oacc, void* o,
+                                         const void* len_dev, int B, int Hq,
+                                         int Hkv, int T, int L, int split,
+                                         float scale, hipStream_t stream) {
+  hipLaunchKernelGGL(decode_attn_split_kernel, dim3(B * Hq, split),
+                     dim3(256), 0, stream, (const unsigned short*)q,
+                     (const unsigned short*)kc, (const unsigned short*)vc,
+                     (float*)ml, (float*)oacc, (const int*)len_dev, B, Hq,
+                     Hkv, T, L, scale);
+  hipLaunchKernelGGL(decode_attn_merge_kernel, dim3(B * Hq), dim3(128), 0,
+                     stream, (const float*)ml, (const float*)oacc,
+                     (unsigned short*)o, split);
+}
