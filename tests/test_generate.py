@@ -159,3 +159,79 @@ def test_gemv_bf16_matches_matmul():
         err = (out.float() - ref).abs().max().item()
         tol = 2e-2 * K ** 0.5
         assert err < tol, (M, N, K, err, tol)
+
+
+@pytest.mark.gpu
+def test_rmsnorm_res_gpu_matches_reference():
+    from torchx_amd import ops
+
+    dev = torch.device("cuda:0")
+    torch.manual_seed(6)
+    x = torch.randn(6, 2048, device=dev, dtype=torch.bfloat16)
+    r = torch.randn(6, 2048, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(2048, device=dev, dtype=torch.bfloat16)
+    s, y = ops.rmsnorm_res(x, r, w, 1e-5)
+    s_ref = (x + r)
+    n = s_ref.float()
+    y_ref = (n * torch.rsqrt(n.pow(2).mean(-1, keepdim=True) + 1e-5)
+             ) * w.float()
+    assert torch.allclose(s.float(), s_ref.float(), atol=2e-2)
+    assert torch.allclose(y.float(), y_ref, atol=3e-2), (
+        (y.float() - y_ref).abs().max())
+    # res=None degrades to plain rmsnorm, s aliases x
+    s2, y2 = ops.rmsnorm_res(x, None, w, 1e-5)
+    assert s2.data_ptr() == x.data_ptr()
+    n2 = x.float()
+    y2_ref = (n2 * torch.rsqrt(n2.pow(2).mean(-1, keepdim=True) + 1e-5)
+              ) * w.float()
+    assert torch.allclose(y2.float(), y2_ref, atol=3e-2)
+
+
+@pytest.mark.gpu
+def test_decode_rope_cache_gpu_matches_fallback():
+    from torchx_amd import ops
+
+    dev = torch.device("cuda:0")
+    torch.manual_seed(7)
+    B, Hq, Hkv, D, T, pos = 3, 8, 4, 128, 32, 17
+    qkv = torch.randn(B, (Hq + 2 * Hkv) * D, device=dev,
+                      dtype=torch.bfloat16)
+    cos, sin = ops.rope_tables(T, D, device=dev)
+    kc = torch.zeros(B, T, Hkv, D, device=dev, dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    q = ops.decode_rope_cache(qkv, kc, vc, cos, sin, pos, Hq)
+    # CPU fallback on the same inputs
+    kc2 = torch.zeros(B, T, Hkv, D, dtype=torch.bfloat16)
+    vc2 = torch.zeros_like(kc2)
+    q2 = ops.decode_rope_cache(qkv.cpu(), kc2, vc2, cos.cpu(), sin.cpu(),
+                               pos, Hq)
+    assert torch.allclose(q.cpu().float(), q2.float(), atol=2e-2)
+    assert torch.allclose(kc.cpu().float(), kc2.float(), atol=2e-2)
+    assert torch.equal(vc.cpu(), vc2)
+    # int32 device-scalar position agrees with the host int
+    kc3 = torch.zeros_like(kc)
+    vc3 = torch.zeros_like(vc)
+    p32 = torch.tensor([pos], dtype=torch.int32, device=dev)
+    q3 = ops.decode_rope_cache(qkv, kc3, vc3, cos, sin, p32, Hq)
+    assert torch.equal(q3, q)
+    assert torch.equal(kc3, kc)
+    assert torch.equal(vc3, vc)
+
+
+@pytest.mark.gpu
+def test_decode_attention_split_long_cache():
+    # L > 512 exercises multi-row-per-wave accumulation in every split
+    from torchx_amd import ops
+    from torchx_amd.ops import reference
+
+    dev = torch.device("cuda:0")
+    torch.manual_seed(8)
+    B, Hq, Hkv, T, L, D = 2, 4, 2, 1200, 1101, 128
+    q = torch.randn(B, Hq, D, device=dev, dtype=torch.bfloat16)
+    kc = torch.randn(B, T, Hkv, D, device=dev, dtype=torch.bfloat16)
+    vc = torch.randn(B, T, Hkv, D, device=dev, dtype=torch.bfloat16)
+    o = ops.decode_attention(q, kc, vc, L)
+    ref = reference.decode_attention(q.cpu(), kc.cpu(), vc.cpu(), L,
+                                     1.0 / D ** 0.5)
+    err = (o.cpu().float() - ref.float()).abs().max().item()
+    assert err < 3e-2, err
