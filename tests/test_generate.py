@@ -235,3 +235,27 @@ def test_decode_attention_split_long_cache():
                                      1.0 / D ** 0.5)
     err = (o.cpu().float() - ref.float()).abs().max().item()
     assert err < 3e-2, err
+
+
+def test_generate_app_cpu(capsys):
+    from torchx_amd.apps import generate_main
+
+    rc = generate_main.main(["--model", "tiny", "--batch", "2",
+                             "--prompt-len", "12", "--new-tokens", "4"])
+    assert rc == 0
+    import json as _json
+
+    line = capsys.readouterr().out.strip().splitlines()[-1]
+    rec = _json.loads(line)
+    assert rec["tokens_per_second"] > 0
+    assert rec["new_tokens"] == 4
+
+
+@pytest.mark.gpu
+def test_generate_app_gpu_graphed(capsys):
+    from torchx_amd.apps import generate_main
+
+    rc = generate_main.main(["--model", "gpu_tiny", "--batch", "2",
+                             "--prompt-len", "32", "--new-tokens", "8",
+                             "--graph"])
+    assert rc == 0
