@@ -259,3 +259,21 @@ def test_generate_app_gpu_graphed(capsys):
                              "--prompt-len", "32", "--new-tokens", "8",
                              "--graph"])
     assert rc == 0
+
+
+@pytest.mark.gpu
+def test_gemv_swiglu_matches_reference():
+    from torchx_amd.ops import hip_ops
+
+    dev = torch.device("cuda:0")
+    torch.manual_seed(9)
+    for M, I, K in [(4, 14336, 4096), (1, 100, 512), (8, 333, 1536)]:
+        x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+        w = torch.randn(2 * I, K, device=dev, dtype=torch.bfloat16)
+        out = hip_ops().gemv_swiglu_bf16(x, w)
+        gu = x.float() @ w.float().t()
+        g, u = gu.chunk(2, dim=-1)
+        ref = torch.nn.functional.silu(g) * u
+        err = (out.float() - ref).abs().max().item()
+        scale = ref.abs().max().item() + 1
+        assert err < 2e-2 * K ** 0.5 + 2e-2 * scale, (M, I, K, err)

@@ -81,9 +81,9 @@ def prefill(model: LlamaModel, tokens: torch.Tensor,
 def decode_step(model: LlamaModel, token: torch.Tensor, caches: list,
                 pos_dev: Optional[torch.Tensor] = None) -> torch.Tensor:
     """One token [B, 1] -> next-position logits [B, vocab], appending to
-    the caches. Fully fused decode layer (9 dispatches on GPU): skinny-M
-    GEMV x4, rope+cache-append x1, flash-decode attention x1, fused
-    residual-add+rmsnorm x2, swiglu x1. With ``pos_dev`` (int32 device
+    the caches. Fully fused decode layer (8 dispatches on GPU): skinny-M
+    GEMV x3 + fused GEMV-SwiGLU x1, rope+cache-append x1, split-K
+    flash-decode attention x2, fused residual-add+rmsnorm x2. With ``pos_dev`` (int32 device
     scalar) the step is hipGraph-capturable: the cache position comes off
     the device and no host state is read."""
     cfg = model.cfg
@@ -106,8 +106,8 @@ def decode_step(model: LlamaModel, token: torch.Tensor, caches: list,
                                           1.0 / cfg.head_dim ** 0.5)
         a = ops.decode_linear(o.reshape(B, -1), blk.wo.weight)
         x, xn = ops.rmsnorm_res(x, a, blk.mlp_norm, cfg.rms_eps)
-        gu = ops.decode_linear(xn, blk.wgu.weight)
-        m = ops.decode_linear(ops.swiglu_packed(gu), blk.wdown.weight)
+        m = ops.decode_linear(ops.decode_linear_swiglu(xn, blk.wgu.weight),
+                              blk.wdown.weight)
         w_next = (blocks[i + 1].attn_norm if i + 1 < len(blocks)
                   else model.final_norm)
         x, xn = ops.rmsnorm_res(x, m, w_next, cfg.rms_eps)

@@ -530,6 +530,22 @@ def decode_rope_cache(qkv: torch.Tensor, kcache: torch.Tensor,
     return q.reshape(B, num_heads, D)
 
 
+def decode_linear_swiglu(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Fused decode MLP front half: ``silu(x @ gate.T) * (x @ up.T)`` for
+    the packed wgu weight ``w = [gate; up]`` [2I, K] — one GEMV-shaped
+    dispatch, no [M, 2I] intermediate. Inference only."""
+    x2 = x.reshape(-1, x.shape[-1])
+    M, K = x2.shape
+    if _on_gpu(x) and M <= 8 and K % 512 == 0 and w.shape[0] % 2 == 0 \
+            and w.dtype == torch.bfloat16 and w.is_contiguous():
+        out = hip_ops().gemv_swiglu_bf16(x2.contiguous(), w)
+    else:
+        gu = x2 @ w.t()
+        g, u = gu.chunk(2, dim=-1)
+        out = torch.nn.functional.silu(g.float()).to(u.dtype) * u
+    return out.reshape(*x.shape[:-1], w.shape[0] // 2)
+
+
 def decode_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Skinny-M linear for the decode path: x [..., M, K] with M <= 8
     total rows -> x @ w.T via the W-stream-bound ``gemv_bf16`` kernel

@@ -51,6 +51,8 @@ void decode_rope_cache_launch(const void*, void*, void*, void*, const void*,
                               int, hipStream_t);
 void rmsnorm_res_launch(const void*, const void*, const void*, void*, void*,
                         long, int, float, hipStream_t);
+void gemv_swiglu_launch(const void*, const void*, void*, int, int, int,
+                        hipStream_t);
 }
 
 namespace {
@@ -412,6 +414,23 @@ at::Tensor gemv_bf16(at::Tensor x, at::Tensor w) {
   return out;
 }
 
+// ---- fused skinny-M GEMV + SwiGLU (decode MLP) ----------------------------
+// x [M, K], w [2I, K] packed (gate | up) -> silu(x@gate.T) * (x@up.T),
+// [M, I]: the wgu projection and the swiglu activation in one dispatch.
+at::Tensor gemv_swiglu_bf16(at::Tensor x, at::Tensor w) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2, "gemv_swiglu: 2D only");
+  const long M = x.size(0), K = x.size(1), I = w.size(0) / 2;
+  TORCH_CHECK(M >= 1 && M <= 8, "gemv_swiglu: M must be 1..8");
+  TORCH_CHECK(w.size(0) == 2 * I && w.size(1) == K, "gemv_swiglu: shape");
+  TORCH_CHECK(K % 512 == 0, "gemv_swiglu: K must be a multiple of 512");
+  auto out = at::empty({M, I}, x.options());
+  gemv_swiglu_launch(x.data_ptr(), w.data_ptr(), out.data_ptr(), (int)M,
+                     (int)I, (int)K, cur_stream());
+  return out;
+}
+
 // ---- fused decode rope + cache append -------------------------------------
 // qkv [B, (Hq+2*Hkv)*128] (one position); ropes q/k with cos/sin [S, 64]
 // f32 tables at row `pos` (host int, or device int32 scalar when `pos_dev`
@@ -519,6 +538,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("decode_attn", &decode_attn);
   m.def("decode_attn_dev", &decode_attn_dev);
   m.def("gemv_bf16", &gemv_bf16);
+  m.def("gemv_swiglu_bf16", &gemv_swiglu_bf16);
   m.def("decode_rope_cache", &decode_rope_cache,
         py::arg("qkv"), py::arg("kc"), py::arg("vc"), py::arg("cos"),
         py::arg("sin"), py::arg("pos") = 0,

@@ -142,3 +142,102 @@ extern "C" void gemv_bf16_launch(const void* x, const void* w, void* o,
     default: break;  // caller guards M <= 8
   }
 }
+
+// Fused wgu-GEMV + SwiGLU for the decode MLP: W is the packed
+// [gate; up] = [2I, K] projection; each wave computes one gate row n and
+// its paired up row n + I, then writes silu(g) * u — the [M, 2I]
+// intermediate and the separate swiglu dispatch disappear. Same W-stream
+// properties as the plain kernel (two sequential 8-KB row streams per
+// wave), same dot2 inner loop.
+template <int M>
+__global__ void __launch_bounds__(256)
+gemv_swiglu_bf16_kernel(const unsigned short* __restrict__ x,  // [M, K]
+                        const unsigned short* __restrict__ w,  // [2I, K]
+                        unsigned short* __restrict__ out,      // [M, I]
+                        int I, int K) {
+  const int wave = threadIdx.x / 64;
+  const int lane = threadIdx.x & 63;
+  const int n0 = blockIdx.x * 4 + wave;  // gate row; up row = n0 + I
+  if (n0 >= I) return;
+
+  float acc[M][2];
+  #pragma unroll
+  for (int m = 0; m < M; ++m) acc[m][0] = acc[m][1] = 0.f;
+
+  const int kc0 = lane * 8;
+  int kc = kc0;
+  for (; kc + 512 * 3 < K; kc += 512 * 4) {
+    ushort8 xf[M][4];
+    #pragma unroll
+    for (int u = 0; u < 4; ++u)
+      #pragma unroll
+      for (int m = 0; m < M; ++m)
+        xf[m][u] = *(const ushort8*)(x + (long)m * K + kc + 512 * u);
+    #pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      const unsigned short* wr = w + (long)(n0 + r * I) * K + kc;
+      ushort8 wf[4];
+      #pragma unroll
+      for (int u = 0; u < 4; ++u) wf[u] = *(const ushort8*)(wr + 512 * u);
+      #pragma unroll
+      for (int u = 0; u < 4; ++u)
+        #pragma unroll
+        for (int j2 = 0; j2 < 4; ++j2)
+          #pragma unroll
+          for (int m = 0; m < M; ++m)
+            acc[m][r] = dot2_bf16(xf[m][u], wf[u], acc[m][r], j2);
+    }
+  }
+  for (; kc < K; kc += 512) {
+    ushort8 xf[M];
+    #pragma unroll
+    for (int m = 0; m < M; ++m)
+      xf[m] = *(const ushort8*)(x + (long)m * K + kc);
+    #pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      ushort8 wf = *(const ushort8*)(w + (long)(n0 + r * I) * K + kc);
+      #pragma unroll
+      for (int j2 = 0; j2 < 4; ++j2)
+        #pragma unroll
+        for (int m = 0; m < M; ++m)
+          acc[m][r] = dot2_bf16(xf[m], wf, acc[m][r], j2);
+    }
+  }
+
+  #pragma unroll
+  for (int m = 0; m < M; ++m)
+    #pragma unroll
+    for (int r = 0; r < 2; ++r) {
+      float v = acc[m][r];
+      #pragma unroll
+      for (int off = 32; off >= 1; off >>= 1) v += __shfl_xor(v, off, 64);
+      acc[m][r] = v;
+    }
+  if (lane == 0) {
+    #pragma unroll
+    for (int m = 0; m < M; ++m) {
+      const float g = acc[m][0], u = acc[m][1];
+      const float sig =
+          1.0f / (1.0f + __builtin_amdgcn_exp2f(-1.44269504f * g));
+      out[(long)m * I + n0] = f32_to_bf16(g * sig * u);
+    }
+  }
+}
+
+extern "C" void gemv_swiglu_launch(const void* x, const void* w, void* o,
+                                   int M, int I, int K, hipStream_t stream) {
+  const int blocks = (I + 3) / 4;
+  switch (M) {
+    #define GSW_CASE(MM)                                                    \
+      case MM:                                                              \
+        hipLaunchKernelGGL((gemv_swiglu_bf16_kernel<MM>), dim3(blocks),     \
+                           dim3(256), 0, stream, (const unsigned short*)x,  \
+                           (const unsigned short*)w, (unsigned short*)o, I, \
+                           K);                                              \
+        break;
+    GSW_CASE(1) GSW_CASE(2) GSW_CASE(3) GSW_CASE(4)
+    GSW_CASE(5) GSW_CASE(6) GSW_CASE(7) GSW_CASE(8)
+    #undef GSW_CASE
+    default: break;
+  }
+}
