@@ -57,3 +57,18 @@ def test_builtins_with_defaults_materialize(builtins):
         app = materialize_appdef(comp.fn, [])
         assert isinstance(app, AppDef), name
         assert app.roles, name
+
+
+def test_component_test_case_helper():
+    """The public ComponentTestCase helper validates a builtin module the
+    way a component author would (reference component_test_base.py:60)."""
+    from torchx_amd.components import utils
+    from torchx_amd.components.component_test_base import ComponentTestCase
+
+    class _T(ComponentTestCase):
+        def test_echo(self):
+            self.validate(utils, "echo")
+
+    t = _T("test_echo")
+    result = t.run()
+    assert result.wasSuccessful(), result.errors or result.failures

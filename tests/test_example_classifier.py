@@ -74,3 +74,35 @@ def test_train_ddp_gloo(tmp_path):
     )
     assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
     assert (tmp_path / "out" / "ckpt_epoch0.pt").exists()
+
+
+def test_interpret_app(tmp_path):
+    """interpret.py (reference lightning/interpret.py counterpart): train a
+    checkpoint, attribute a batch, artifacts land on the output path."""
+    import json
+
+    import torch
+
+    from torchx_amd.examples.apps.classifier import interpret, train
+
+    from torchx_amd.examples.apps.classifier.data import create_random_data
+
+    data = create_random_data(str(tmp_path / "data"), n=32)
+    out = str(tmp_path / "out")
+    rc = train.main(["--dataset_path", data, "--output_path", out,
+                     "--epochs", "1", "--batch_size", "8",
+                     "--skip_export"])
+    assert rc == 0
+    ckpts = sorted((tmp_path / "out").glob("ckpt_epoch*.pt"))
+    assert ckpts
+    att_out = str(tmp_path / "attr")
+    rc = interpret.main(["--load-path", str(ckpts[-1]),
+                         "--data-path", data,
+                         "--output-path", att_out, "--batch", "4"])
+    assert rc == 0
+    blob = torch.load(tmp_path / "attr" / "attributions.pt",
+                      weights_only=True)
+    assert blob["saliency"].shape[0] == 4
+    assert blob["occlusion"].shape[0] == 4
+    summary = json.loads((tmp_path / "attr" / "summary.json").read_text())
+    assert summary["n"] == 4
