@@ -16,7 +16,8 @@ import torch
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--model", default="llama3_8b",
-                   choices=["llama3_8b", "gpu_tiny"])
+                   choices=["llama3_8b", "gpu_tiny", "mixtral_8x7b",
+                            "mixtral_gpu_tiny"])
     p.add_argument("--batch", type=int, default=4)
     p.add_argument("--prompt", type=int, default=512)
     p.add_argument("--new", type=int, default=128)
@@ -24,13 +25,34 @@ def main():
                    help="hipGraph decode loop (whole step as one replay)")
     args = p.parse_args()
 
-    from torchx_amd.models.generate import KVCache, decode_step, prefill
+    from torchx_amd.models.generate import KVCache, prefill
+    from torchx_amd.models.generate import decode_step as decode_step_dense
     from torchx_amd.models.llama import LlamaModel, llama3_8b, llama_gpu_tiny
 
     dev = torch.device("cuda:0")
-    cfg = llama3_8b() if args.model == "llama3_8b" else llama_gpu_tiny()
-    torch.manual_seed(0)
-    model = LlamaModel(cfg, device=dev)
+    moe = args.model.startswith("mixtral")
+    if moe:
+        from torchx_amd.models.generate_moe import (
+            decode_step_moe, prefill_moe,
+        )
+        from torchx_amd.models.mixtral import (
+            MixtralModel, mixtral_8x7b, mixtral_gpu_tiny,
+        )
+
+        if args.graph:
+            print("--graph is dense-only (MoE routing is data-dependent)",
+                  file=sys.stderr)
+            return 1
+        cfg = (mixtral_8x7b() if args.model == "mixtral_8x7b"
+               else mixtral_gpu_tiny())
+        torch.manual_seed(0)
+        model = MixtralModel(cfg, device=dev)
+        prefill, decode_step = prefill_moe, decode_step_moe
+    else:
+        cfg = llama3_8b() if args.model == "llama3_8b" else llama_gpu_tiny()
+        torch.manual_seed(0)
+        model = LlamaModel(cfg, device=dev)
+        decode_step = decode_step_dense
     B, S0, N = args.batch, args.prompt, args.new
     tokens = torch.randint(0, cfg.vocab_size, (B, S0), device=dev)
     caches = [KVCache.empty(cfg, B, S0 + N + 8, dev)

@@ -26,7 +26,8 @@ import torch
 def parse_args(argv):
     p = argparse.ArgumentParser(description="torchx_amd serving app")
     p.add_argument("--model", default="llama3_8b",
-                   choices=["llama3_8b", "gpu_tiny", "tiny"])
+                   choices=["llama3_8b", "gpu_tiny", "tiny",
+                            "mixtral_8x7b", "mixtral_tiny"])
     p.add_argument("--batch", type=int, default=4)
     p.add_argument("--prompt-len", type=int, default=512)
     p.add_argument("--new-tokens", type=int, default=64)
@@ -41,26 +42,36 @@ def parse_args(argv):
 def main(argv=None) -> int:
     args = parse_args(argv)
     from torchx_amd.models.generate import generate, generate_graphed
+    from torchx_amd.models.generate_moe import generate_moe
     from torchx_amd.models.llama import (
         LlamaModel, llama3_8b, llama_gpu_tiny, llama_tiny,
     )
+    from torchx_amd.models.mixtral import (
+        MixtralModel, mixtral_8x7b, mixtral_tiny,
+    )
 
+    moe = args.model.startswith("mixtral")
     cfg = {"llama3_8b": llama3_8b, "gpu_tiny": llama_gpu_tiny,
-           "tiny": llama_tiny}[args.model]()
+           "tiny": llama_tiny, "mixtral_8x7b": mixtral_8x7b,
+           "mixtral_tiny": mixtral_tiny}[args.model]()
     use_gpu = torch.cuda.is_available()
     dev = torch.device("cuda:0" if use_gpu else "cpu")
     torch.manual_seed(args.seed)
-    model = LlamaModel(cfg, device=dev if use_gpu else None)
+    cls = MixtralModel if moe else LlamaModel
+    model = cls(cfg, device=dev if use_gpu else None)
     tokens = torch.randint(0, cfg.vocab_size,
                            (args.batch, args.prompt_len), device=dev)
     if use_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     if args.graph:
-        if not use_gpu:
-            print("--graph requires a GPU", file=sys.stderr)
+        if not use_gpu or moe:
+            print("--graph needs a GPU and a dense model", file=sys.stderr)
             return 1
         out = generate_graphed(model, tokens, args.new_tokens)
+    elif moe:
+        out = generate_moe(model, tokens, args.new_tokens,
+                           temperature=args.temperature, top_k=args.top_k)
     else:
         out = generate(model, tokens, args.new_tokens,
                        temperature=args.temperature, top_k=args.top_k)
