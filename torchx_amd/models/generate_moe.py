@@ -37,16 +37,17 @@ def _moe_mlp_decode(moe: MoELayer, x: torch.Tensor) -> torch.Tensor:
     logits = (x @ moe.router.weight.t()).float()          # [B, E]
     w, idx = torch.topk(logits, cfg.top_k, dim=-1)
     w = torch.softmax(w, dim=-1).to(x.dtype)              # [B, k]
+    idx_h = idx.cpu()  # ONE host sync per layer; partitioning is host-side
     out = torch.zeros_like(x)
     for e in range(cfg.num_experts):
-        mask = idx == e                                   # [B, k]
-        rows = mask.any(-1).nonzero(as_tuple=True)[0]
-        if rows.numel() == 0:
+        rows_h = (idx_h == e).any(-1).nonzero(as_tuple=True)[0]
+        if rows_h.numel() == 0:
             continue
+        rows = rows_h.to(x.device, non_blocking=True)
         exp = moe.local_experts[e]
         h = ops.decode_linear_swiglu(x[rows], exp.wgu.weight)
         ye = ops.decode_linear(h, exp.wdown.weight)
-        coef = (w * mask).sum(-1)[rows]                   # [n]
+        coef = (w * (idx == e)).sum(-1)[rows]             # [n], on device
         out[rows] += ye * coef[:, None]
     return out
 
