@@ -72,3 +72,19 @@ def test_generate_moe_gpu_end_to_end():
     err = (dec.float() - full.float()).abs().max().item()
     scale = full.float().abs().max().item() + 1e-6
     assert err < 5e-2 * scale, (err, scale)
+
+
+def test_generate_app_moe_cpu(capsys):
+    from torchx_amd.apps import generate_main
+
+    rc = generate_main.main(["--model", "mixtral_tiny", "--batch", "2",
+                             "--prompt-len", "10", "--new-tokens", "3"])
+    assert rc == 0
+    import json as _json
+
+    rec = _json.loads(capsys.readouterr().out.strip().splitlines()[-1])
+    assert rec["model"] == "mixtral_tiny" and rec["tokens_per_second"] > 0
+    # --graph is refused for MoE models
+    rc = generate_main.main(["--model", "mixtral_tiny", "--graph",
+                             "--prompt-len", "8", "--new-tokens", "2"])
+    assert rc == 1

@@ -50,12 +50,20 @@ def main():
               f"matmul {t_m:7.1f} us ({gb/(t_m*1e-6):6.0f} GB/s)")
         tot_g += t_g
         tot_m += t_m
-        if name != "lm_head":
-            pass
     layer_g = tot_g - t_g  # last shape is lm_head
     layer_m = tot_m - t_m
     print(f"one decode step GEMV total: gemv {(layer_g*32 + t_g)/1e3:.3f} ms"
           f"  matmul {(layer_m*32 + t_m)/1e3:.3f} ms")
+
+    # fused wgu+swiglu vs separate gemv + swiglu kernel
+    M, I, K = 4, 14336, 4096
+    x = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
+    w = torch.randn(2 * I, K, device=dev, dtype=torch.bfloat16)
+    t_f = time_fn(lambda: ops.gemv_swiglu_bf16(x, w))
+    gb = 2 * I * K * 2 / 1e9
+    print(f"wgu+swiglu fused: {t_f:7.1f} us ({gb/(t_f*1e-6):6.0f} GB/s) "
+          f"vs separate gemv {time_fn(lambda: ops.gemv_bf16(x, w)):7.1f} us"
+          f" + swiglu kernel")
     return 0
 
 
