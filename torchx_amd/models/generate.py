@@ -85,7 +85,9 @@ def decode_step(model: LlamaModel, token: torch.Tensor, caches: list,
     GEMV x3 + fused GEMV-SwiGLU x1, rope+cache-append x1, split-K
     flash-decode attention x2, fused residual-add+rmsnorm x2. With ``pos_dev`` (int32 device
     scalar) the step is hipGraph-capturable: the cache position comes off
-    the device and no host state is read."""
+    the device and no host state is read; an int32 [B] ``pos_dev`` runs
+    the step RAGGED — every sequence at its own cache position
+    (continuous batching, see ContinuousBatcher)."""
     cfg = model.cfg
     B = token.shape[0]
     blocks = model.blocks
@@ -100,10 +102,9 @@ def decode_step(model: LlamaModel, token: torch.Tensor, caches: list,
             cache.length = pos + 1
             o = ops.decode_attention(q, cache.k, cache.v, pos + 1)
         else:
-            from torchx_amd.ops import hip_ops
-
-            o = hip_ops().decode_attn_dev(q, cache.k, cache.v, pos_dev,
-                                          1.0 / cfg.head_dim ** 0.5)
+            # scalar pos_dev: hipGraph loop; [B] pos_dev: ragged decode
+            # (continuous batching) — each row attends its own length
+            o = ops.decode_attention_dev(q, cache.k, cache.v, pos_dev)
         a = ops.decode_linear(o.reshape(B, -1), blk.wo.weight)
         x, xn = ops.rmsnorm_res(x, a, blk.mlp_norm, cfg.rms_eps)
         m = ops.decode_linear(ops.decode_linear_swiglu(xn, blk.wgu.weight),
@@ -227,3 +228,78 @@ def generate_graphed(model: LlamaModel, tokens: torch.Tensor,
     for _ in range(remaining):
         out.append(dec.step().clone())
     return torch.cat(out, dim=1)
+
+
+class ContinuousBatcher:
+    """Continuous batching over a fixed pool of cache rows: sequences of
+    DIFFERENT lengths decode together in one ragged step (int32 [B]
+    position vector drives rope, cache append and attention per row), and
+    a finished row can be re-admitted with a new prompt while the others
+    keep decoding — the serving pattern behind vLLM-style engines,
+    without paging (288 GB HBM3E holds the whole [B, T] cache pool).
+
+    Greedy decoding; rows are independent — admit() prefills ONE row's
+    cache slice, step() advances every active row one token.
+    """
+
+    def __init__(self, model: LlamaModel, max_batch: int, max_len: int):
+        self.model = model
+        self.cfg = model.cfg
+        assert max_len <= self.cfg.max_seq_len, "beyond the rope tables"
+        dev = next(model.parameters()).device
+        self.device = dev
+        self.max_len = max_len
+        self.caches = [KVCache.empty(self.cfg, max_batch, max_len, dev)
+                       for _ in range(self.cfg.num_layers)]
+        self.pos = torch.zeros(max_batch, dtype=torch.int32, device=dev)
+        # host mirror of pos (deterministic: admit sets it, step adds 1) —
+        # overflow checks never touch the device
+        self.pos_host = [0] * max_batch
+        self.tok = torch.zeros(max_batch, 1, dtype=torch.long, device=dev)
+        self.active = [False] * max_batch
+
+    def free_rows(self) -> list:
+        return [i for i, a in enumerate(self.active) if not a]
+
+    @torch.no_grad()
+    def admit(self, row: int, prompt: torch.Tensor) -> torch.Tensor:
+        """Prefill ``prompt`` [S0] into cache row ``row``; returns the
+        first generated token (scalar tensor). The row then participates
+        in every subsequent step()."""
+        assert not self.active[row], f"row {row} is occupied"
+        S0 = prompt.numel()
+        assert S0 + 1 < self.max_len
+        row_caches = [
+            KVCache(k=c.k[row:row + 1], v=c.v[row:row + 1], length=0)
+            for c in self.caches
+        ]
+        logits = prefill(self.model, prompt.reshape(1, S0).to(self.device),
+                         row_caches)
+        nxt = logits.argmax(-1)
+        self.tok[row, 0] = nxt[0]
+        self.pos[row] = S0
+        self.pos_host[row] = S0
+        self.active[row] = True
+        return nxt[0]
+
+    def retire(self, row: int) -> None:
+        self.active[row] = False
+
+    @torch.no_grad()
+    def step(self) -> torch.Tensor:
+        """One ragged decode step over ALL rows (inactive rows compute
+        garbage that callers ignore — the batch shape stays static).
+        Returns the new tokens [max_batch]."""
+        assert any(self.active), "no active sequences"
+        logits = decode_step(self.model, self.tok, self.caches,
+                             pos_dev=self.pos)
+        nxt = logits.argmax(-1, keepdim=True)
+        self.tok.copy_(nxt)
+        # bound every row (retired rows keep stepping as ignored garbage;
+        # the wrap keeps their cache writes in range without a host sync)
+        self.pos.add_(1).remainder_(self.max_len)
+        for i in range(len(self.pos_host)):
+            self.pos_host[i] = (self.pos_host[i] + 1) % self.max_len
+            if self.active[i] and self.pos_host[i] + 1 >= self.max_len:
+                self.active[i] = False  # out of cache: auto-retire
+        return nxt.reshape(-1)

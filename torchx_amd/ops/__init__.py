@@ -518,15 +518,25 @@ def decode_rope_cache(qkv: torch.Tensor, kcache: torch.Tensor,
                                                vcache, cos, sin, 0, pos)
         return hip_ops().decode_rope_cache(qkv.contiguous(), kcache, vcache,
                                            cos, sin, int(pos), None)
-    p = int(pos.item()) if isinstance(pos, torch.Tensor) else int(pos)
     q, k, v = qkv.split([num_heads * D, Hkv * D, Hkv * D], dim=-1)
     q = q.reshape(B, 1, num_heads, D)
     k = k.reshape(B, 1, Hkv, D)
+    v = v.reshape(B, 1, Hkv, D)
+    if isinstance(pos, torch.Tensor) and pos.numel() > 1:
+        outs = []  # ragged: each row ropes/appends at its own position
+        for b in range(B):
+            p = int(pos[b].item())
+            cs, sn = cos[p:p + 1], sin[p:p + 1]
+            outs.append(rope(q[b:b + 1].contiguous(), cs, sn))
+            kcache[b, p:p + 1] = rope(k[b:b + 1].contiguous(), cs, sn)[0]
+            vcache[b, p:p + 1] = v[b]
+        return torch.cat(outs, dim=0).reshape(B, num_heads, D)
+    p = int(pos.item()) if isinstance(pos, torch.Tensor) else int(pos)
     cs, sn = cos[p:p + 1], sin[p:p + 1]
     q = rope(q.contiguous(), cs, sn)
     k = rope(k.contiguous(), cs, sn)
     kcache[:, p:p + 1] = k
-    vcache[:, p:p + 1] = v.reshape(B, 1, Hkv, D)
+    vcache[:, p:p + 1] = v
     return q.reshape(B, num_heads, D)
 
 
@@ -544,6 +554,28 @@ def decode_linear_swiglu(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
         g, u = gu.chunk(2, dim=-1)
         out = torch.nn.functional.silu(g.float()).to(u.dtype) * u
     return out.reshape(*x.shape[:-1], w.shape[0] // 2)
+
+
+def decode_attention_dev(q: torch.Tensor, kcache: torch.Tensor,
+                         vcache: torch.Tensor, pos: torch.Tensor,
+                         scale: Optional[float] = None) -> torch.Tensor:
+    """Decode attention with the cache position(s) read from a tensor:
+    ``pos`` is int32, a scalar (all rows at the same position — the
+    hipGraph path) or [B] (ragged / continuous batching — each sequence
+    attends over its own ``pos[b] + 1`` cache rows)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    if _on_gpu(q):
+        return hip_ops().decode_attn_dev(q.contiguous(), kcache, vcache,
+                                         pos.contiguous(), scale)
+    if pos.numel() == 1:
+        return reference.decode_attention(q, kcache, vcache,
+                                          int(pos.item()) + 1, scale)
+    outs = [reference.decode_attention(q[b:b + 1], kcache[b:b + 1],
+                                       vcache[b:b + 1],
+                                       int(pos[b].item()) + 1, scale)
+            for b in range(q.shape[0])]
+    return torch.cat(outs, dim=0)
 
 
 def decode_linear(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:

@@ -39,16 +39,16 @@ void fp8_cast_transpose_launch(const void*, void*, void*, void*,
                                const void*, long, long, int, hipStream_t);
 void transpose_bf16_launch(const void*, void*, long, long, hipStream_t);
 void decode_attn_launch(const void*, const void*, const void*, void*,
-                        const void*, int, int, int, int, int, float,
+                        const void*, int, int, int, int, int, int, float,
                         hipStream_t);
 void gemv_bf16_launch(const void*, const void*, void*, int, int, int,
                       hipStream_t);
 void decode_attn_split_launch(const void*, const void*, const void*, void*,
                               void*, void*, const void*, int, int, int, int,
-                              int, int, float, hipStream_t);
+                              int, int, int, float, hipStream_t);
 void decode_rope_cache_launch(const void*, void*, void*, void*, const void*,
                               const void*, const void*, int, int, int, int,
-                              int, hipStream_t);
+                              int, int, hipStream_t);
 void rmsnorm_res_launch(const void*, const void*, const void*, void*, void*,
                         long, int, float, hipStream_t);
 void gemv_swiglu_launch(const void*, const void*, void*, int, int, int,
@@ -359,11 +359,11 @@ at::Tensor decode_attn(at::Tensor q, at::Tensor kc, at::Tensor vc, long L,
                           q.options().dtype(at::kFloat));
     decode_attn_split_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
                              ml.data_ptr(), oacc.data_ptr(), o.data_ptr(),
-                             nullptr, B, Hq, Hkv, T, (int)L, split,
+                             nullptr, B, Hq, Hkv, T, (int)L, split, 0,
                              (float)scale, cur_stream());
   } else {
     decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
-                       o.data_ptr(), nullptr, B, Hq, Hkv, T, (int)L,
+                       o.data_ptr(), nullptr, B, Hq, Hkv, T, (int)L, 0,
                        (float)scale, cur_stream());
   }
   return o;
@@ -375,9 +375,12 @@ at::Tensor decode_attn(at::Tensor q, at::Tensor kc, at::Tensor vc, long L,
 at::Tensor decode_attn_dev(at::Tensor q, at::Tensor kc, at::Tensor vc,
                            at::Tensor pos, double scale) {
   check_bf16(q, "q");
-  TORCH_CHECK(pos.scalar_type() == at::kInt && pos.numel() == 1,
-              "pos must be an int32 scalar tensor on device");
   const int B = q.size(0), Hq = q.size(1);
+  TORCH_CHECK(pos.scalar_type() == at::kInt &&
+                  (pos.numel() == 1 || pos.numel() == B),
+              "pos must be an int32 scalar or [B] tensor on device");
+  TORCH_CHECK(pos.is_contiguous(), "pos must be contiguous");
+  const int per_row = pos.numel() == B && B > 1 ? 1 : 0;
   const int T = kc.size(1), Hkv = kc.size(2);
   auto o = at::empty_like(q);
   const int split = std::max(1, std::min(16, 512 / (B * Hq)));
@@ -389,11 +392,11 @@ at::Tensor decode_attn_dev(at::Tensor q, at::Tensor kc, at::Tensor vc,
     decode_attn_split_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
                              ml.data_ptr(), oacc.data_ptr(), o.data_ptr(),
                              pos.data_ptr(), B, Hq, Hkv, T, 0, split,
-                             (float)scale, cur_stream());
+                             per_row, (float)scale, cur_stream());
   } else {
     decode_attn_launch(q.data_ptr(), kc.data_ptr(), vc.data_ptr(),
                        o.data_ptr(), pos.data_ptr(), B, Hq, Hkv, T, 0,
-                       (float)scale, cur_stream());
+                       per_row, (float)scale, cur_stream());
   }
   return o;
 }
@@ -450,10 +453,14 @@ at::Tensor decode_rope_cache(at::Tensor qkv, at::Tensor kc, at::Tensor vc,
   const int Hq = (int)nh - 2 * Hkv;
   TORCH_CHECK(Hq >= 1 && qkv.size(1) == nh * 128, "qkv width mismatch");
   const void* pd = nullptr;
+  int per_row = 0;
   if (pos_dev.has_value()) {
-    TORCH_CHECK(pos_dev->scalar_type() == at::kInt && pos_dev->numel() == 1,
-                "pos_dev must be an int32 scalar tensor");
+    TORCH_CHECK(pos_dev->scalar_type() == at::kInt &&
+                    (pos_dev->numel() == 1 || pos_dev->numel() == B),
+                "pos_dev must be an int32 scalar or [B] tensor");
+    TORCH_CHECK(pos_dev->is_contiguous(), "pos_dev must be contiguous");
     pd = pos_dev->data_ptr();
+    per_row = pos_dev->numel() == B && B > 1 ? 1 : 0;
   } else {
     TORCH_CHECK(pos >= 0 && pos < T, "pos out of cache range");
     TORCH_CHECK(pos < cos_t.size(0), "pos beyond rope table");
@@ -461,7 +468,8 @@ at::Tensor decode_rope_cache(at::Tensor qkv, at::Tensor kc, at::Tensor vc,
   auto q = at::empty({B, (long)Hq, 128}, qkv.options());
   decode_rope_cache_launch(qkv.data_ptr(), q.data_ptr(), kc.data_ptr(),
                            vc.data_ptr(), cos_t.data_ptr(), sin_t.data_ptr(),
-                           pd, B, Hq, Hkv, T, (int)pos, cur_stream());
+                           pd, B, Hq, Hkv, T, (int)pos, per_row,
+                           cur_stream());
   return q;
 }
 

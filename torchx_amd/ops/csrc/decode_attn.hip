@@ -19,11 +19,13 @@ decode_attn_kernel(const unsigned short* __restrict__ q,   // [B, Hq, 128]
                    const unsigned short* __restrict__ kc,  // [B, T, Hkv, 128]
                    const unsigned short* __restrict__ vc,  // [B, T, Hkv, 128]
                    unsigned short* __restrict__ o,         // [B, Hq, 128]
-                   const int* __restrict__ len_dev,  // optional: L = *len_dev + 1
-                   int B, int Hq, int Hkv, int T, int L_host, float scale) {
-  const int L = len_dev ? (*len_dev + 1) : L_host;
+                   const int* __restrict__ len_dev,  // optional: L = len+1
+                   int B, int Hq, int Hkv, int T, int L_host, int per_row,
+                   float scale) {
   const int b = blockIdx.x / Hq;
   const int hq = blockIdx.x % Hq;
+  // per_row: len_dev is [B] (ragged / continuous batching); else a scalar
+  const int L = len_dev ? (len_dev[per_row ? b : 0] + 1) : L_host;
   const int hkv = hq / (Hq / Hkv);
 
   const int wave = threadIdx.x / 64;
@@ -116,12 +118,12 @@ decode_attn_kernel(const unsigned short* __restrict__ q,   // [B, Hq, 128]
 extern "C" void decode_attn_launch(const void* q, const void* kc,
                                    const void* vc, void* o,
                                    const void* len_dev, int B, int Hq,
-                                   int Hkv, int T, int L, float scale,
-                                   hipStream_t stream) {
+                                   int Hkv, int T, int L, int per_row,
+                                   float scale, hipStream_t stream) {
   hipLaunchKernelGGL(decode_attn_kernel, dim3(B * Hq), dim3(256), 0, stream,
                      (const unsigned short*)q, (const unsigned short*)kc,
                      (const unsigned short*)vc, (unsigned short*)o,
-                     (const int*)len_dev, B, Hq, Hkv, T, L, scale);
+                     (const int*)len_dev, B, Hq, Hkv, T, L, per_row, scale);
 }
 
 // Fused decode-side rope + cache append: consumes the packed wqkv output
@@ -139,11 +141,12 @@ decode_rope_cache_kernel(const unsigned short* __restrict__ qkv,
                          const float* __restrict__ cos_t,    // [S, 64]
                          const float* __restrict__ sin_t,
                          const int* __restrict__ pos_dev,
-                         int B, int Hq, int Hkv, int T, int pos_host) {
-  const int pos = pos_dev ? *pos_dev : pos_host;
+                         int B, int Hq, int Hkv, int T, int pos_host,
+                         int per_row) {
   const int nh = Hq + 2 * Hkv;
   const int b = blockIdx.x / nh;
   const int h = blockIdx.x % nh;
+  const int pos = pos_dev ? pos_dev[per_row ? b : 0] : pos_host;
   const int lane = threadIdx.x;  // 0..63: one rotation pair (lane, lane+64)
   const unsigned short* src = qkv + ((long)b * nh + h) * HD;
   if (h < Hq + Hkv) {  // q or k head: rotate
@@ -171,13 +174,13 @@ extern "C" void decode_rope_cache_launch(const void* qkv, void* qout,
                                          const void* cos_t, const void* sin_t,
                                          const void* pos_dev, int B, int Hq,
                                          int Hkv, int T, int pos_host,
-                                         hipStream_t stream) {
+                                         int per_row, hipStream_t stream) {
   hipLaunchKernelGGL(decode_rope_cache_kernel, dim3(B * (Hq + 2 * Hkv)),
                      dim3(64), 0, stream, (const unsigned short*)qkv,
                      (unsigned short*)qout, (unsigned short*)kc,
                      (unsigned short*)vc, (const float*)cos_t,
                      (const float*)sin_t, (const int*)pos_dev, B, Hq, Hkv, T,
-                     pos_host);
+                     pos_host, per_row);
 }
 
 // ---- split-K (flash-decode) variant ---------------------------------------
@@ -195,11 +198,11 @@ decode_attn_split_kernel(const unsigned short* __restrict__ q,
                          float* __restrict__ oacc,  // [B*Hq, SPLIT, 128]
                          const int* __restrict__ len_dev,
                          int B, int Hq, int Hkv, int T, int L_host,
-                         float scale) {
-  const int L = len_dev ? (*len_dev + 1) : L_host;
+                         int per_row, float scale) {
   const int bq = blockIdx.x;
   const int b = bq / Hq;
   const int hq = bq % Hq;
+  const int L = len_dev ? (len_dev[per_row ? b : 0] + 1) : L_host;
   const int hkv = hq / (Hq / Hkv);
   const int split = gridDim.y;
   const int chunk = (L + split - 1) / split;
@@ -309,12 +312,13 @@ extern "C" void decode_attn_split_launch(const void* q, const void* kc,
                                          void* oacc, void* o,
                                          const void* len_dev, int B, int Hq,
                                          int Hkv, int T, int L, int split,
-                                         float scale, hipStream_t stream) {
+                                         int per_row, float scale,
+                                         hipStream_t stream) {
   hipLaunchKernelGGL(decode_attn_split_kernel, dim3(B * Hq, split),
                      dim3(256), 0, stream, (const unsigned short*)q,
                      (const unsigned short*)kc, (const unsigned short*)vc,
                      (float*)ml, (float*)oacc, (const int*)len_dev, B, Hq,
-                     Hkv, T, L, scale);
+                     Hkv, T, L, per_row, scale);
   hipLaunchKernelGGL(decode_attn_merge_kernel, dim3(B * Hq), dim3(128), 0,
                      stream, (const float*)ml, (const float*)oacc,
                      (unsigned short*)o, split);
