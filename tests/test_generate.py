@@ -346,3 +346,22 @@ def test_continuous_batcher_gpu_matches_per_sequence():
         got2.append(int(toks[1]))
     assert got1 == solo1, (got1, solo1)
     assert got2 == solo2, (got2, solo2)
+
+
+def test_continuous_batcher_auto_retire():
+    """A row that exhausts its cache pool is retired automatically and
+    its slot becomes re-admittable."""
+    from torchx_amd.models.generate import ContinuousBatcher
+
+    torch.manual_seed(13)
+    cfg = llama_tiny()
+    model = LlamaModel(cfg)
+    cb = ContinuousBatcher(model, max_batch=2, max_len=16)
+    cb.admit(0, torch.randint(0, cfg.vocab_size, (12,)))
+    assert cb.free_rows() == [1]
+    for _ in range(4):
+        cb.step()
+    assert not cb.active[0]          # 12 + 4 past the 16-row pool
+    assert 0 in cb.free_rows()
+    cb.admit(0, torch.randint(0, cfg.vocab_size, (5,)))
+    assert cb.active[0]

@@ -158,10 +158,11 @@ def generate(
 class GraphedDecoder:
     """hipGraph-captured decode loop: the whole per-token step (embed,
     32 layers, lm_head, argmax, cache append, position bump) replays as
-    ONE graph with zero host work (the fused eager step is ~290
-    dispatches; un-fused it was ~450 and launch-bound). The cache length and rope position are driven
-    by a device int32 scalar that the captured step increments itself,
-    so one capture serves every subsequent token.
+    ONE graph with zero host work (the fused eager step is ~260
+    dispatches; un-fused it was ~450 and launch-bound). The cache
+    length and rope position are driven by a device int32 scalar that
+    the captured step increments itself, so one capture serves every
+    subsequent token.
 
     Greedy-only (the argmax feeds back inside the graph).
     """
