@@ -111,3 +111,48 @@ def test_adamw_step_cpu_matches_torch_optim():
     assert torch.allclose(p32, ref_p.detach(), atol=1e-6), \
         (p32 - ref_p.detach()).abs().max()
     assert torch.allclose(p16, p32.to(torch.bfloat16))
+
+
+def test_decode_linear_cpu_unit():
+    import torch
+
+    from torchx_amd import ops
+
+    x = torch.randn(3, 64)
+    w = torch.randn(10, 64)
+    assert torch.allclose(ops.decode_linear(x, w), x @ w.t(), atol=1e-5)
+
+
+def test_decode_linear_swiglu_cpu_unit():
+    import torch
+
+    from torchx_amd import ops
+
+    x = torch.randn(2, 32)
+    w = torch.randn(12, 32)  # [gate(6); up(6)]
+    out = ops.decode_linear_swiglu(x, w)
+    gu = x @ w.t()
+    g, u = gu.chunk(2, -1)
+    ref = torch.nn.functional.silu(g) * u
+    assert out.shape == (2, 6)
+    assert torch.allclose(out, ref, atol=1e-5)
+
+
+def test_decode_attention_dev_cpu_ragged():
+    import torch
+
+    from torchx_amd import ops
+    from torchx_amd.ops import reference
+
+    torch.manual_seed(3)
+    B, Hq, Hkv, T, D = 3, 4, 2, 10, 16
+    q = torch.randn(B, Hq, D)
+    kc = torch.randn(B, T, Hkv, D)
+    vc = torch.randn(B, T, Hkv, D)
+    pos = torch.tensor([2, 7, 4], dtype=torch.int32)
+    out = ops.decode_attention_dev(q, kc, vc, pos)
+    for b in range(B):
+        ref = reference.decode_attention(
+            q[b:b + 1], kc[b:b + 1], vc[b:b + 1], int(pos[b]) + 1,
+            1.0 / D ** 0.5)
+        assert torch.allclose(out[b:b + 1], ref, atol=1e-4)
