@@ -88,3 +88,28 @@ def test_generate_app_moe_cpu(capsys):
     rc = generate_main.main(["--model", "mixtral_tiny", "--graph",
                              "--prompt-len", "8", "--new-tokens", "2"])
     assert rc == 1
+
+
+def test_continuous_batcher_moe_matches_per_sequence():
+    """Ragged MoE decode: two Mixtral prompts of different lengths in one
+    batcher produce exactly their solo greedy streams."""
+    from torchx_amd.models.generate import ContinuousBatcher
+
+    torch.manual_seed(14)
+    cfg = mixtral_tiny()
+    model = MixtralModel(cfg)
+    p1 = torch.randint(0, cfg.vocab_size, (9,))
+    p2 = torch.randint(0, cfg.vocab_size, (14,))
+    solo1 = generate_moe(model, p1.reshape(1, -1), 4)[0, 9:].tolist()
+    solo2 = generate_moe(model, p2.reshape(1, -1), 4)[0, 14:].tolist()
+    cb = ContinuousBatcher(model, max_batch=2, max_len=32,
+                           prefill_fn=prefill_moe,
+                           decode_fn=decode_step_moe)
+    got1 = [int(cb.admit(0, p1))]
+    got2 = [int(cb.admit(1, p2))]
+    for _ in range(3):
+        toks = cb.step()
+        got1.append(int(toks[0]))
+        got2.append(int(toks[1]))
+    assert got1 == solo1, (got1, solo1)
+    assert got2 == solo2, (got2, solo2)

@@ -243,9 +243,15 @@ class ContinuousBatcher:
     cache slice, step() advances every active row one token.
     """
 
-    def __init__(self, model: LlamaModel, max_batch: int, max_len: int):
+    def __init__(self, model, max_batch: int, max_len: int,
+                 prefill_fn=None, decode_fn=None):
+        """``prefill_fn``/``decode_fn`` default to the dense Llama path;
+        pass ``generate_moe.prefill_moe`` / ``decode_step_moe`` to run a
+        Mixtral pool (the MoE step takes the same ``pos_dev`` vector)."""
         self.model = model
         self.cfg = model.cfg
+        self._prefill = prefill_fn or prefill
+        self._decode = decode_fn or decode_step
         assert max_len <= self.cfg.max_seq_len, "beyond the rope tables"
         dev = next(model.parameters()).device
         self.device = dev
@@ -274,8 +280,9 @@ class ContinuousBatcher:
             KVCache(k=c.k[row:row + 1], v=c.v[row:row + 1], length=0)
             for c in self.caches
         ]
-        logits = prefill(self.model, prompt.reshape(1, S0).to(self.device),
-                         row_caches)
+        logits = self._prefill(self.model,
+                               prompt.reshape(1, S0).to(self.device),
+                               row_caches)
         nxt = logits.argmax(-1)
         self.tok[row, 0] = nxt[0]
         self.pos[row] = S0
@@ -292,8 +299,8 @@ class ContinuousBatcher:
         garbage that callers ignore — the batch shape stays static).
         Returns the new tokens [max_batch]."""
         assert any(self.active), "no active sequences"
-        logits = decode_step(self.model, self.tok, self.caches,
-                             pos_dev=self.pos)
+        logits = self._decode(self.model, self.tok, self.caches,
+                              pos_dev=self.pos)
         nxt = logits.argmax(-1, keepdim=True)
         self.tok.copy_(nxt)
         # bound every row (retired rows keep stepping as ignored garbage;

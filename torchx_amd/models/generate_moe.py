@@ -83,23 +83,28 @@ def prefill_moe(model: MixtralModel, tokens: torch.Tensor,
 
 
 @torch.no_grad()
-def decode_step_moe(model: MixtralModel, token: torch.Tensor,
-                    caches: list) -> torch.Tensor:
+def decode_step_moe(model: MixtralModel, token: torch.Tensor, caches: list,
+                    pos_dev=None) -> torch.Tensor:
     """One token [B, 1] -> next-position logits [B, vocab]. Attention
     half = the fused Llama decode kernels; MLP half = active-expert
-    GEMV+SwiGLU."""
+    GEMV+SwiGLU. An int32 [B] ``pos_dev`` runs the step RAGGED (every
+    sequence at its own cache position — continuous batching); the MoE
+    half is position-independent, so it needs no change."""
     cfg = model.cfg
     B = token.shape[0]
     blocks = model.blocks
-    pos = caches[0].length
+    pos = caches[0].length if pos_dev is None else pos_dev
     x = model.embed(token).reshape(B, -1)
     _, xn = ops.rmsnorm_res(x, None, blocks[0].attn_norm, cfg.rms_eps)
     for i, (blk, cache) in enumerate(zip(blocks, caches)):
         qkv = ops.decode_linear(xn, blk.wqkv.weight)
         q = ops.decode_rope_cache(qkv, cache.k, cache.v, model.rope_cos,
                                   model.rope_sin, pos, cfg.num_heads)
-        cache.length = pos + 1
-        o = ops.decode_attention(q, cache.k, cache.v, pos + 1)
+        if pos_dev is None:
+            cache.length = pos + 1
+            o = ops.decode_attention(q, cache.k, cache.v, pos + 1)
+        else:
+            o = ops.decode_attention_dev(q, cache.k, cache.v, pos_dev)
         a = ops.decode_linear(o.reshape(B, -1), blk.wo.weight)
         x, xn = ops.rmsnorm_res(x, a, blk.mlp_norm, cfg.rms_eps)
         m = _moe_mlp_decode(blk.moe, xn)
