@@ -120,3 +120,33 @@ def test_trackers_from_environ_fsspec(tmp_path):
     }
     ts = trackers_from_environ(env)
     assert len(ts) == 1
+
+
+def test_get_device_mounts_count_aware():
+    """reference schedulers/devices.py:43-54 — per-count EFA/Neuron
+    mounts, unknown names warn and skip, AMD GPU nodes are static."""
+    import warnings as _w
+
+    from torchx_amd.schedulers.devices import get_device_mounts
+
+    m = get_device_mounts({"vpc.amazonaws.com/efa": 2})
+    assert [d.src_path for d in m] == ["/dev/infiniband/uverbs0",
+                                       "/dev/infiniband/uverbs1"]
+    m = get_device_mounts({"aws.amazon.com/neurondevice": 1})
+    assert m[0].src_path == "/dev/neuron0"
+    m = get_device_mounts({"amd.com/gpu": 8})
+    assert {d.src_path for d in m} == {"/dev/kfd", "/dev/dri"}
+    with _w.catch_warnings(record=True) as rec:
+        _w.simplefilter("always")
+        assert get_device_mounts({"bogus/dev": 1}) == []
+    assert any("bogus" in str(r.message) for r in rec)
+
+
+def test_scheduler_factories_direct():
+    """Every scheduler module exposes create_scheduler(session_name)."""
+    from torchx_amd.schedulers import docker_scheduler, local_scheduler
+
+    s = local_scheduler.create_scheduler("t")
+    assert s.backend == "local_cwd"
+    d = docker_scheduler.create_scheduler("t")
+    assert d.backend == "local_docker"

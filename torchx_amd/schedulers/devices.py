@@ -11,11 +11,12 @@ DeviceMounts for containerized schedulers.
 from __future__ import annotations
 
 import functools
+import warnings
 import json
 import logging
 import shutil
 import subprocess
-from typing import Dict, List, Optional
+from typing import Any, Dict, List, Optional
 
 from torchx_amd.specs import DeviceMount, Resource
 
@@ -26,18 +27,42 @@ ROCR_VISIBLE_DEVICES = "ROCR_VISIBLE_DEVICES"
 
 # named-device -> container device mounts (KFD + DRI are what ROCm containers
 # need; see the docker scheduler)
-DEVICES: Dict[str, List[DeviceMount]] = {
-    "amd.com/gpu": [
-        DeviceMount(src_path="/dev/kfd", dst_path="/dev/kfd"),
-        DeviceMount(src_path="/dev/dri", dst_path="/dev/dri"),
-    ],
+def _amd_gpu_mounts(n: int) -> List[DeviceMount]:
+    # /dev/kfd + /dev/dri expose every GPU; per-device isolation is done
+    # with HIP_VISIBLE_DEVICES, not per-node mounts
+    return [DeviceMount(src_path="/dev/kfd", dst_path="/dev/kfd"),
+            DeviceMount(src_path="/dev/dri", dst_path="/dev/dri")]
+
+
+def _efa_mounts(n: int) -> List[DeviceMount]:
+    # parity: reference devices.py — one uverbs node per EFA device
+    return [DeviceMount(src_path=f"/dev/infiniband/uverbs{i}",
+                        dst_path=f"/dev/infiniband/uverbs{i}")
+            for i in range(n)]
+
+
+def _neuron_mounts(n: int) -> List[DeviceMount]:
+    return [DeviceMount(src_path=f"/dev/neuron{i}",
+                        dst_path=f"/dev/neuron{i}") for i in range(n)]
+
+
+DEVICES: Dict[str, Any] = {
+    "amd.com/gpu": _amd_gpu_mounts,
+    "vpc.amazonaws.com/efa": _efa_mounts,
+    "aws.amazon.com/neurondevice": _neuron_mounts,
 }
 
 
 def get_device_mounts(devices: Dict[str, int]) -> List[DeviceMount]:
+    """Named device/count map -> DeviceMounts (parity: reference
+    devices.py:43-54 — unknown names warn and are skipped)."""
     mounts: List[DeviceMount] = []
-    for name in devices:
-        mounts.extend(DEVICES.get(name, []))
+    for name, n in devices.items():
+        fn = DEVICES.get(name)
+        if fn is None:
+            warnings.warn(f"could not find named device: {name}")
+            continue
+        mounts.extend(fn(int(n)))
     return mounts
 
 
